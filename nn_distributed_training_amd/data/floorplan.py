@@ -1,0 +1,86 @@
+"""Synthetic floorplans and robot trajectories.
+
+The reference ships a floorplan PNG + hand-drawn waypoint files
+(floorplans/32_data/). This environment has no data files, so the density
+workload runs on procedurally generated floorplans: a bordered box with
+random interior walls, plus smooth random loop trajectories through free
+space. A PNG path can still be used via data.lidar.Lidar2D when present.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+
+def synthetic_floorplan(
+    nx: int = 256,
+    ny: int = 256,
+    num_walls: int = 10,
+    border_width: int = 16,
+    seed: int = 0,
+) -> np.ndarray:
+    """Occupancy image in [0,1], shape [ny, nx]; 1 = wall."""
+    rng = np.random.default_rng(seed)
+    img = np.zeros((ny, nx), dtype=float)
+    img[:border_width, :] = 1.0
+    img[-border_width:, :] = 1.0
+    img[:, :border_width] = 1.0
+    img[:, -border_width:] = 1.0
+    thick = max(2, nx // 64)
+    for _ in range(num_walls):
+        horizontal = rng.random() < 0.5
+        span = rng.integers(nx // 4, nx // 2)
+        x0 = rng.integers(border_width, nx - border_width - span)
+        y0 = rng.integers(border_width + thick, ny - border_width - thick)
+        if horizontal:
+            img[y0 : y0 + thick, x0 : x0 + span] = 1.0
+        else:
+            img[x0 : x0 + span, y0 : y0 + thick] = 1.0
+    return img
+
+
+def synthetic_waypoints(
+    img: np.ndarray,
+    num_nodes: int,
+    points_per_path: int = 8,
+    seed: int = 0,
+):
+    """Random loop waypoints through free space, one array per node.
+
+    Waypoints are in the reference's normalized [-1, 1]-ish convention
+    (TrajectoryLidarDataset multiplies by nx/2, ny/2 —
+    floorplans/lidar/lidar.py:311-318), closed (last == first) so cubic
+    interpolation produces a loop. Each waypoint is rejection-sampled from
+    free space.
+    """
+    rng = np.random.default_rng(seed)
+    ny, nx = img.shape
+    xs = nx * np.linspace(-0.5, 0.5, num=nx)
+    ys = ny * np.linspace(-0.5, 0.5, num=ny)
+
+    def is_free(xn, yn):
+        # xn, yn are normalized in [-1, 1]; map to pixel indices
+        ix = int(np.clip((xn * 0.5 + 0.5) * (nx - 1), 0, nx - 1))
+        iy = int(np.clip((yn * 0.5 + 0.5) * (ny - 1), 0, ny - 1))
+        return img[iy, ix] < 0.5
+
+    del xs, ys
+    paths = []
+    for _ in range(num_nodes):
+        # sample an angular loop around a random free center
+        for _attempt in range(200):
+            cx, cy = rng.uniform(-0.45, 0.45, size=2)
+            r0 = rng.uniform(0.1, 0.3)
+            angles = np.linspace(0, 2 * np.pi, points_per_path, endpoint=False)
+            radii = r0 * rng.uniform(0.6, 1.1, size=points_per_path)
+            px = cx + radii * np.cos(angles)
+            py = cy + radii * np.sin(angles)
+            px = np.clip(px, -0.8, 0.8)
+            py = np.clip(py, -0.8, 0.8)
+            if all(is_free(x, y) for x, y in zip(px, py)):
+                break
+        wp = np.stack(
+            [np.append(px, px[0]), np.append(py, py[0])], axis=1
+        )
+        paths.append(wp)
+    return paths
