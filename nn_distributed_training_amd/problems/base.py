@@ -53,13 +53,21 @@ class ProblemBase:
             for i in self.local_nodes
         }
 
+        # Per-node shuffle generators seeded by (data_seed, node id): node
+        # i's batch stream is identical no matter which rank hosts it, so
+        # a multi-rank run reproduces the single-process run exactly
+        # (tested in tests/test_distributed.py).
+        data_seed = int(self.conf.get("data_seed", 0))
         self.train_loaders = {}
         self.train_iters = {}
         for i in self.local_nodes:
+            g = torch.Generator()
+            g.manual_seed(data_seed * 100003 + i)
             self.train_loaders[i] = torch.utils.data.DataLoader(
                 self.train_sets[i],
                 batch_size=self.conf["train_batch_size"],
                 shuffle=True,
+                generator=g,
             )
             self.train_iters[i] = iter(self.train_loaders[i])
 

@@ -1,0 +1,130 @@
+"""Multi-process (gloo, world_size 2) parity tests.
+
+The algorithms are snapshot-synchronous, per-node data streams are seeded
+by node id, and the neighbor exchange moves full vectors without
+reduction — so a 2-rank run must reproduce the single-process run's
+parameters EXACTLY (no floating-point reduction-order slack).
+"""
+
+import os
+import pickle
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+import networkx as nx
+
+from nn_distributed_training_amd.models import MNISTConvNet
+from nn_distributed_training_amd.data.mnist import SyntheticMNIST, split_train_set
+from nn_distributed_training_amd.problems.dist_mnist_problem import (
+    DistMNISTProblem,
+)
+from nn_distributed_training_amd.optimizers import build_optimizer
+
+N_NODES = 4
+
+
+def _prob_conf(alg):
+    confs = {
+        "dinno": {
+            "alg_name": "dinno",
+            "rho_init": 0.5,
+            "rho_scaling": 1.001,
+            "outer_iterations": 4,
+            "primal_iterations": 2,
+            "primal_optimizer": "adam",
+            "persistant_primal_opt": True,
+            "primal_lr_start": 0.005,
+            "primal_lr_finish": 0.001,
+            "lr_decay_type": "linear",
+            "profile": False,
+        },
+        "dsgd": {
+            "alg_name": "dsgd",
+            "outer_iterations": 4,
+            "alpha0": 0.005,
+            "mu": 0.001,
+            "profile": False,
+        },
+        "dsgt": {
+            "alg_name": "dsgt",
+            "outer_iterations": 4,
+            "alpha": 0.005,
+            "init_grads": True,
+            "profile": False,
+        },
+    }
+    return {
+        "problem_name": alg,
+        "train_batch_size": 16,
+        "val_batch_size": 64,
+        "data_seed": 7,
+        "verbose_evals": False,
+        "metrics": ["consensus_error"],
+        "metrics_config": {"evaluate_frequency": 100},
+        "optimizer_config": confs[alg],
+    }
+
+
+def _run_training(alg):
+    """Build the problem deterministically and train; return [N, n]
+    stacked final parameters for the LOCAL nodes (all nodes if world=1)."""
+    torch.set_default_dtype(torch.float64)
+    torch.manual_seed(42)
+    graph = nx.cycle_graph(N_NODES)
+    train = SyntheticMNIST(400, seed=0)
+    val = SyntheticMNIST(100, seed=1)
+    subsets = split_train_set(train, N_NODES, "hetero")
+    base_model = MNISTConvNet(3, 5, 64)
+    conf = _prob_conf(alg)
+    pr = DistMNISTProblem(
+        graph, base_model, torch.nn.NLLLoss(), subsets, val,
+        torch.device("cpu"), conf,
+    )
+    dopt = build_optimizer(pr, torch.device("cpu"), conf["optimizer_config"])
+    dopt.train()
+    return pr.local_params_stack(), pr.local_nodes
+
+
+def _worker(rank, world, alg, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        stack, nodes = _run_training(alg)
+        with open(os.path.join(out_dir, f"rank{rank}.pkl"), "wb") as f:
+            pickle.dump((list(nodes), stack.numpy()), f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("alg", ["dinno", "dsgd", "dsgt"])
+def test_two_rank_matches_single_process(alg, tmp_path):
+    # single-process golden
+    golden, nodes = _run_training(alg)
+    assert list(nodes) == list(range(N_NODES))
+
+    # two ranks over gloo
+    port = 29510 + hash(alg) % 100
+    mp.start_processes(
+        _worker,
+        args=(2, alg, port, str(tmp_path)),
+        nprocs=2,
+        join=True,
+        start_method="spawn",
+    )
+
+    pieces = {}
+    for r in range(2):
+        with open(tmp_path / f"rank{r}.pkl", "rb") as f:
+            local_nodes, stack = pickle.load(f)
+        for li, i in enumerate(local_nodes):
+            pieces[i] = torch.from_numpy(stack[li])
+    assert sorted(pieces) == list(range(N_NODES))
+    dist_stack = torch.stack([pieces[i] for i in range(N_NODES)])
+
+    torch.testing.assert_close(dist_stack, golden, rtol=0, atol=0)
