@@ -1,0 +1,137 @@
+// MNISTConvNet head: Conv2d(1, F, k) + ReLU + MaxPool2d(2), fused
+// forward and backward, batched over node replicas.
+//
+// The conv is the first layer, so backward only needs dW/db (no dX) —
+// the whole layer costs two kernels per direction. Images are staged in
+// LDS (28*28 doubles = 6.3 KB) so each of the k*k taps reads on-chip.
+
+#include "common.h"
+
+namespace conv {
+
+// X: [L*B, 28*28]; theta: [L, n] flat stack, W at w_off ([F,1,k,k]
+// row-major = [F, k*k]), b at b_off; Y: [L*B, F*P*P] pooled+ReLU output;
+// idx: [L*B, F*P*P] argmax position (0..3) for pool backward.
+// Grid: one block per image, 256 threads.
+template <typename T>
+__global__ void conv_pool_fwd_k(
+    const T* __restrict__ X, const T* __restrict__ theta,
+    T* __restrict__ Y, unsigned char* __restrict__ idx,
+    long n, long w_off, long b_off, int B, int F, int K, int IMG) {
+  extern __shared__ __align__(16) unsigned char smem_raw[];
+  T* img = reinterpret_cast<T*>(smem_raw);             // [IMG*IMG]
+  T* wgt = img + IMG * IMG;                            // [F*K*K + F]
+
+  const long lb = blockIdx.x;           // image index in [0, L*B)
+  const long l = lb / B;
+  const int conv_out = IMG - (K - 1);
+  const int P = conv_out / 2;
+
+  // stage image + this node's conv weights/bias
+  for (int t = threadIdx.x; t < IMG * IMG; t += blockDim.x) {
+    img[t] = X[lb * IMG * IMG + t];
+  }
+  const T* Wg = theta + l * n + w_off;
+  const T* bg = theta + l * n + b_off;
+  for (int t = threadIdx.x; t < F * K * K; t += blockDim.x) {
+    wgt[t] = Wg[t];
+  }
+  for (int t = threadIdx.x; t < F; t += blockDim.x) {
+    wgt[F * K * K + t] = bg[t];
+  }
+  __syncthreads();
+
+  const int npool = F * P * P;
+  for (int t = threadIdx.x; t < npool; t += blockDim.x) {
+    const int f = t / (P * P);
+    const int py = (t / P) % P;
+    const int px = t % P;
+    const T* wf = wgt + f * K * K;
+    const T bias = wgt[F * K * K + f];
+
+    T best = T(0);       // ReLU floor: max(0, .) pooled
+    int best_i = 0;
+    #pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      const int cy = 2 * py + (d >> 1);
+      const int cx = 2 * px + (d & 1);
+      T acc = bias;
+      for (int ky = 0; ky < K; ++ky) {
+        const T* row = img + (cy + ky) * IMG + cx;
+        const T* wr = wf + ky * K;
+        for (int kx = 0; kx < K; ++kx) {
+          acc += row[kx] * wr[kx];
+        }
+      }
+      if (acc > best) { best = acc; best_i = d; }
+    }
+    Y[lb * npool + t] = best;          // relu(max pre-act) == max(relu)
+    idx[lb * npool + t] = (unsigned char)best_i;
+  }
+}
+
+// Backward to weights/bias: route each pooled grad to its argmax conv
+// position, multiply by the image window. dY is the grad AFTER the relu
+// mask (pooled output > 0), applied by the caller via act_grad.
+// Grid: one block per (l, f); each thread strides over (b, py, px) and
+// accumulates a private dW[K*K]+db, then block-reduces.
+template <typename T, int KMAX>
+__global__ void conv_pool_bwd_k(
+    const T* __restrict__ dY, const unsigned char* __restrict__ idx,
+    const T* __restrict__ X, T* __restrict__ gstack,
+    long n, long w_off, long b_off, int B, int F, int K, int IMG) {
+  const int l = blockIdx.x / F;
+  const int f = blockIdx.x % F;
+  const int conv_out = IMG - (K - 1);
+  const int P = conv_out / 2;
+  const int npool = F * P * P;
+
+  T dw[KMAX * KMAX];
+  T db = T(0);
+  #pragma unroll
+  for (int i = 0; i < KMAX * KMAX; ++i) dw[i] = T(0);
+
+  const int work = B * P * P;
+  for (int t = threadIdx.x; t < work; t += blockDim.x) {
+    const int b = t / (P * P);
+    const int py = (t / P) % P;
+    const int px = t % P;
+    const long lb = (long)l * B + b;
+    const long o = lb * npool + f * P * P + py * P + px;
+    const T g = dY[o];
+    if (g == T(0)) continue;
+    const int d = idx[o];
+    const int cy = 2 * py + (d >> 1);
+    const int cx = 2 * px + (d & 1);
+    const T* img = X + lb * IMG * IMG;
+    db += g;
+    for (int ky = 0; ky < K; ++ky) {
+      const T* row = img + (cy + ky) * IMG + cx;
+      for (int kx = 0; kx < K; ++kx) {
+        dw[ky * K + kx] += g * row[kx];
+      }
+    }
+  }
+
+  // block reduction of dw[K*K] and db through LDS
+  __shared__ T red[256];
+  for (int i = 0; i < K * K + 1; ++i) {
+    T v = (i < K * K) ? dw[i] : db;
+    red[threadIdx.x] = v;
+    __syncthreads();
+    for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+      if (threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+      if (i < K * K) {
+        gstack[(long)l * n + w_off + (long)f * K * K + i] = red[0];
+      } else {
+        gstack[(long)l * n + b_off + f] = red[0];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+}  // namespace conv

@@ -1,0 +1,170 @@
+// Decentralized-optimizer elementwise kernels, batched over the L node
+// replicas of one rank (the reference's per-node, per-parameter Python
+// loops — optimizers/dsgd.py:37-58, dinno.py:119-125, dsgt.py:58-105 —
+// each become ONE launch here).
+//
+// Layout: parameter stacks are [L, n] row-major. Neighbor vectors live in
+// a "table" [R, n] (R = L local snapshot rows + received remote rows);
+// per-node neighbor lists are CSR int32 (offsets [L+1], indices into the
+// table). All kernels are memory-bound grid-stride loops — the design
+// goal is one pass over HBM per algorithm step with everything fused.
+
+#include "common.h"
+
+namespace ew {
+
+constexpr int BLOCK = 256;
+
+// ---------------------------------------------------------------------
+// DiNNO round prologue (reference optimizers/dinno.py:119-124, fused):
+//   S_i      = sum_{j in N(i)} th_j
+//   dual_i  += rho * (deg_i * th_i - S_i)
+//   s_i      = (deg_i * th_i + S_i) / 2     (= sum_j th_reg_j, the only
+//                                            reduction the penalty
+//                                            gradient needs)
+template <typename T>
+__global__ void dinno_dual_threg_k(
+    const T* __restrict__ table,   // [R, n]; rows 0..L-1 = local snapshot
+    const int* __restrict__ offs,  // [L+1]
+    const int* __restrict__ idx,   // CSR neighbor rows
+    T* __restrict__ duals,         // [L, n] in/out
+    T* __restrict__ s_out,         // [L, n] out
+    T rho, long n, long L) {
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
+    const int k0 = offs[l], k1 = offs[l + 1];
+    T S = T(0);
+    for (int k = k0; k < k1; ++k) {
+      S += table[(long)idx[k] * n + e];
+    }
+    const T th = table[l * n + e];
+    const T deg = T(k1 - k0);
+    duals[t] += rho * (deg * th - S);
+    s_out[t] = (deg * th + S) * T(0.5);
+  }
+}
+
+// ---------------------------------------------------------------------
+// Generic weighted row-combine (DSGD mixing, reference dsgd.py:37-46,
+// made snapshot-synchronous):  out_l = sum_k w_k * table[idx_k]
+// (the caller includes the self row with weight W_ll in the CSR).
+template <typename T>
+__global__ void mix_rows_k(
+    const T* __restrict__ table, const int* __restrict__ offs,
+    const int* __restrict__ idx, const T* __restrict__ w,
+    T* __restrict__ out, long n, long L) {
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
+    const int k0 = offs[l], k1 = offs[l + 1];
+    T acc = T(0);
+    for (int k = k0; k < k1; ++k) {
+      acc += w[k] * table[(long)idx[k] * n + e];
+    }
+    out[t] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------
+// DSGT mixing (reference dsgt.py:58-75, synchronous):
+//   p_out_l = sum_k w_k * (p_k - alpha * y_k)
+//   y_mix_l = sum_k w_k * y_k
+// The table rows bundle [p | y] as [R, 2n] so params and tracker ride
+// one exchange (2x comm volume).
+template <typename T>
+__global__ void dsgt_mix_k(
+    const T* __restrict__ table,  // [R, 2n]
+    const int* __restrict__ offs, const int* __restrict__ idx,
+    const T* __restrict__ w, T* __restrict__ p_out,
+    T* __restrict__ y_mix, T alpha, long n, long L) {
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
+    const int k0 = offs[l], k1 = offs[l + 1];
+    T accp = T(0), accy = T(0);
+    for (int k = k0; k < k1; ++k) {
+      const T* row = table + (long)idx[k] * (2 * n);
+      accp += w[k] * row[e];
+      accy += w[k] * row[n + e];
+    }
+    p_out[t] = accp - alpha * accy;
+    y_mix[t] = accy;
+  }
+}
+
+// DSGT tracker update (reference dsgt.py:87-105):
+//   y = y_mix + g_new - g_old ;  g_old = g_new
+template <typename T>
+__global__ void dsgt_y_update_k(
+    const T* __restrict__ y_mix, const T* __restrict__ g_new,
+    T* __restrict__ g_old, T* __restrict__ y, long total) {
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const T gn = g_new[t];
+    y[t] = y_mix[t] + gn - g_old[t];
+    g_old[t] = gn;
+  }
+}
+
+// ---------------------------------------------------------------------
+// Fused primal step. The DiNNO penalty gradient is analytic
+// (d/dth [ rho * sum_j ||th - th_reg_j||^2 ] = 2 rho (deg th - s), plus
+// the dual term from <th, dual>), so it folds into the optimizer update
+// and the autograd-visible loss never materializes (reference
+// dinno.py:74-91 builds it through torch.cdist + autograd every primal
+// iteration).  mode: 0=Adam 1=AdamW 2=SGD (matching dinno.py:55-72);
+// with_penalty=false gives the plain local step (DSGD's dsgd.py:49-58).
+template <typename T, int MODE, bool WITH_PENALTY>
+__global__ void fused_step_k(
+    T* __restrict__ theta, const T* __restrict__ grad,
+    const T* __restrict__ dual,   // null unless WITH_PENALTY
+    const T* __restrict__ s,      // null unless WITH_PENALTY
+    const int* __restrict__ deg,  // [L], null unless WITH_PENALTY
+    T* __restrict__ m, T* __restrict__ v,  // Adam state (null for SGD)
+    T rho, T lr, T beta1, T beta2, T eps, T wd, T bc1, T bc2,
+    long n, long L) {
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    T th = theta[t];
+    T g = grad[t];
+    if (WITH_PENALTY) {
+      const long l = t / n;
+      g += dual[t] + T(2) * rho * (T(deg[l]) * th - s[t]);
+    }
+    if (MODE == 2) {  // SGD
+      theta[t] = th - lr * g;
+      continue;
+    }
+    if (MODE == 1) {  // AdamW decoupled weight decay
+      th -= lr * wd * th;
+    } else if (wd != T(0)) {  // Adam L2
+      g += wd * th;
+    }
+    const T mt = beta1 * m[t] + (T(1) - beta1) * g;
+    const T vt = beta2 * v[t] + (T(1) - beta2) * g * g;
+    m[t] = mt;
+    v[t] = vt;
+    // bc1 = 1-beta1^t, bc2 = 1-beta2^t (host-computed per step)
+    theta[t] = th - lr * (mt / bc1) / (::sqrt(vt / bc2) + eps);
+  }
+}
+
+// Plain axpy: theta -= alpha * grad  (DSGD local step)
+template <typename T>
+__global__ void axpy_k(T* __restrict__ x, const T* __restrict__ g,
+                       T alpha, long total) {
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    x[t] += alpha * g[t];
+  }
+}
+
+}  // namespace ew

@@ -1,0 +1,331 @@
+// Torch extension entry: dispatch + launch for the CDNA4 kernel library.
+// Single translation unit (the kernel files are header-style templates).
+
+#include <torch/extension.h>
+
+#include "common.h"
+#include "elementwise.hip"
+#include "gemm.hip"
+#include "conv.hip"
+#include "losses.hip"
+
+#include <ATen/hip/HIPContext.h>
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+inline int grid_1d(long total, int block = 256, int cap = 4096) {
+  long g = (total + block - 1) / block;
+  return (int)std::min<long>(g, cap);
+}
+
+#define CHECK_DEV(t) \
+  TORCH_CHECK((t).is_cuda() && (t).is_contiguous(), #t " must be contiguous on GPU")
+
+#define DISPATCH_FT(scalar_t_tensor, ...)                                  \
+  AT_DISPATCH_FLOATING_TYPES(scalar_t_tensor.scalar_type(), "ndta_ops",    \
+                             [&] { __VA_ARGS__ });
+
+// ---------------------------------------------------------------- ew --
+void dinno_dual_threg(torch::Tensor table, torch::Tensor offs,
+                      torch::Tensor idx, torch::Tensor duals,
+                      torch::Tensor s_out, double rho) {
+  CHECK_DEV(table); CHECK_DEV(duals); CHECK_DEV(s_out);
+  const long L = duals.size(0), n = duals.size(1);
+  DISPATCH_FT(table, {
+    hipLaunchKernelGGL(ew::dinno_dual_threg_k<scalar_t>,
+        dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+        table.data_ptr<scalar_t>(), offs.data_ptr<int>(),
+        idx.data_ptr<int>(), duals.data_ptr<scalar_t>(),
+        s_out.data_ptr<scalar_t>(), (scalar_t)rho, n, L);
+  });
+  HIP_CHECK_LAST();
+}
+
+void mix_rows(torch::Tensor table, torch::Tensor offs, torch::Tensor idx,
+              torch::Tensor w, torch::Tensor out) {
+  CHECK_DEV(table); CHECK_DEV(out);
+  const long L = out.size(0), n = out.size(1);
+  DISPATCH_FT(table, {
+    hipLaunchKernelGGL(ew::mix_rows_k<scalar_t>,
+        dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+        table.data_ptr<scalar_t>(), offs.data_ptr<int>(),
+        idx.data_ptr<int>(), w.data_ptr<scalar_t>(),
+        out.data_ptr<scalar_t>(), n, L);
+  });
+  HIP_CHECK_LAST();
+}
+
+void dsgt_mix(torch::Tensor table, torch::Tensor offs, torch::Tensor idx,
+              torch::Tensor w, torch::Tensor p_out, torch::Tensor y_mix,
+              double alpha) {
+  CHECK_DEV(table); CHECK_DEV(p_out); CHECK_DEV(y_mix);
+  const long L = p_out.size(0), n = p_out.size(1);
+  TORCH_CHECK(table.size(1) == 2 * n, "dsgt table must be [R, 2n]");
+  DISPATCH_FT(table, {
+    hipLaunchKernelGGL(ew::dsgt_mix_k<scalar_t>,
+        dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+        table.data_ptr<scalar_t>(), offs.data_ptr<int>(),
+        idx.data_ptr<int>(), w.data_ptr<scalar_t>(),
+        p_out.data_ptr<scalar_t>(), y_mix.data_ptr<scalar_t>(),
+        (scalar_t)alpha, n, L);
+  });
+  HIP_CHECK_LAST();
+}
+
+void dsgt_y_update(torch::Tensor y_mix, torch::Tensor g_new,
+                   torch::Tensor g_old, torch::Tensor y) {
+  CHECK_DEV(y);
+  const long total = y.numel();
+  DISPATCH_FT(y, {
+    hipLaunchKernelGGL(ew::dsgt_y_update_k<scalar_t>,
+        dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+        y_mix.data_ptr<scalar_t>(), g_new.data_ptr<scalar_t>(),
+        g_old.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(), total);
+  });
+  HIP_CHECK_LAST();
+}
+
+void fused_step(torch::Tensor theta, torch::Tensor grad,
+                c10::optional<torch::Tensor> dual,
+                c10::optional<torch::Tensor> s,
+                c10::optional<torch::Tensor> deg,
+                c10::optional<torch::Tensor> m,
+                c10::optional<torch::Tensor> v,
+                double rho, double lr, double beta1, double beta2,
+                double eps, double wd, long step_t, long mode) {
+  CHECK_DEV(theta); CHECK_DEV(grad);
+  const long L = theta.size(0), n = theta.size(1);
+  const bool pen = dual.has_value();
+  DISPATCH_FT(theta, {
+    const scalar_t bc1 =
+        (scalar_t)(1.0 - std::pow(beta1, (double)step_t));
+    const scalar_t bc2 =
+        (scalar_t)(1.0 - std::pow(beta2, (double)step_t));
+    auto launch = [&](auto mode_c, auto pen_c) {
+      hipLaunchKernelGGL(
+          (ew::fused_step_k<scalar_t, decltype(mode_c)::value,
+                            decltype(pen_c)::value>),
+          dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+          theta.data_ptr<scalar_t>(), grad.data_ptr<scalar_t>(),
+          pen ? dual->data_ptr<scalar_t>() : nullptr,
+          pen ? s->data_ptr<scalar_t>() : nullptr,
+          pen ? deg->data_ptr<int>() : nullptr,
+          m.has_value() ? m->data_ptr<scalar_t>() : nullptr,
+          v.has_value() ? v->data_ptr<scalar_t>() : nullptr,
+          (scalar_t)rho, (scalar_t)lr, (scalar_t)beta1,
+          (scalar_t)beta2, (scalar_t)eps, (scalar_t)wd, bc1, bc2, n, L);
+    };
+    using c0 = std::integral_constant<int, 0>;
+    using c1 = std::integral_constant<int, 1>;
+    using c2 = std::integral_constant<int, 2>;
+    using bt = std::integral_constant<bool, true>;
+    using bf = std::integral_constant<bool, false>;
+    if (mode == 0) { pen ? launch(c0{}, bt{}) : launch(c0{}, bf{}); }
+    else if (mode == 1) { pen ? launch(c1{}, bt{}) : launch(c1{}, bf{}); }
+    else { pen ? launch(c2{}, bt{}) : launch(c2{}, bf{}); }
+  });
+  HIP_CHECK_LAST();
+}
+
+void axpy(torch::Tensor x, torch::Tensor g, double alpha) {
+  CHECK_DEV(x);
+  DISPATCH_FT(x, {
+    hipLaunchKernelGGL(ew::axpy_k<scalar_t>,
+        dim3(grid_1d(x.numel())), dim3(ew::BLOCK), 0, cur_stream(),
+        x.data_ptr<scalar_t>(), g.data_ptr<scalar_t>(),
+        (scalar_t)alpha, x.numel());
+  });
+  HIP_CHECK_LAST();
+}
+
+// -------------------------------------------------------------- gemm --
+void linear_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
+                c10::optional<torch::Tensor> Z, long w_off, long b_off,
+                long M, long I, long O, long act, double scale) {
+  CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
+  const long L = theta.size(0), n = theta.size(1);
+  dim3 grid((O + 15) / 16, (M + 15) / 16, L);
+  DISPATCH_FT(X, {
+    hipLaunchKernelGGL(gemm::linear_fwd_k<scalar_t>,
+        grid, dim3(16, 16), 0, cur_stream(),
+        X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+        Y.data_ptr<scalar_t>(),
+        Z.has_value() ? Z->data_ptr<scalar_t>() : nullptr,
+        n, w_off, b_off, (int)M, (int)I, (int)O, (int)act,
+        (scalar_t)scale);
+  });
+  HIP_CHECK_LAST();
+}
+
+void act_grad(torch::Tensor dY, torch::Tensor Y,
+              c10::optional<torch::Tensor> Z, torch::Tensor dZ,
+              long act, double scale) {
+  CHECK_DEV(dY); CHECK_DEV(dZ);
+  DISPATCH_FT(dY, {
+    hipLaunchKernelGGL(gemm::act_grad_k<scalar_t>,
+        dim3(grid_1d(dY.numel())), dim3(256), 0, cur_stream(),
+        dY.data_ptr<scalar_t>(), Y.data_ptr<scalar_t>(),
+        Z.has_value() ? Z->data_ptr<scalar_t>() : nullptr,
+        dZ.data_ptr<scalar_t>(), dY.numel(), (int)act, (scalar_t)scale);
+  });
+  HIP_CHECK_LAST();
+}
+
+void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
+                   torch::Tensor dX, long w_off, long M, long I, long O) {
+  CHECK_DEV(dZ); CHECK_DEV(dX);
+  const long L = theta.size(0), n = theta.size(1);
+  dim3 grid((I + 15) / 16, (M + 15) / 16, L);
+  DISPATCH_FT(dZ, {
+    hipLaunchKernelGGL(gemm::linear_bwd_dx_k<scalar_t>,
+        grid, dim3(16, 16), 0, cur_stream(),
+        dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+        dX.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+  });
+  HIP_CHECK_LAST();
+}
+
+void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
+                   torch::Tensor gstack, long w_off, long b_off,
+                   long M, long I, long O) {
+  CHECK_DEV(dZ); CHECK_DEV(X); CHECK_DEV(gstack);
+  const long L = gstack.size(0), n = gstack.size(1);
+  dim3 grid((I + 15) / 16, (O + 15) / 16, L);
+  DISPATCH_FT(dZ, {
+    hipLaunchKernelGGL(gemm::linear_bwd_dw_k<scalar_t>,
+        grid, dim3(16, 16), 0, cur_stream(),
+        dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+        gstack.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+    hipLaunchKernelGGL(gemm::bias_grad_k<scalar_t>,
+        dim3((O + 255) / 256, 1, L), dim3(256), 0, cur_stream(),
+        dZ.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+        n, b_off, (int)M, (int)O);
+  });
+  HIP_CHECK_LAST();
+}
+
+// -------------------------------------------------------------- conv --
+void conv_pool_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
+                   torch::Tensor idx, long w_off, long b_off, long B,
+                   long F, long K, long IMG) {
+  CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
+  const long L = theta.size(0), n = theta.size(1);
+  DISPATCH_FT(X, {
+    const size_t shmem =
+        (IMG * IMG + F * K * K + F) * sizeof(scalar_t);
+    hipLaunchKernelGGL(conv::conv_pool_fwd_k<scalar_t>,
+        dim3(L * B), dim3(256), shmem, cur_stream(),
+        X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+        Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+        n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+  });
+  HIP_CHECK_LAST();
+}
+
+void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
+                   torch::Tensor gstack, long w_off, long b_off, long B,
+                   long F, long K, long IMG) {
+  CHECK_DEV(dY); CHECK_DEV(X); CHECK_DEV(gstack);
+  const long L = gstack.size(0), n = gstack.size(1);
+  TORCH_CHECK(K <= 7, "conv_pool_bwd supports kernel size <= 7");
+  DISPATCH_FT(dY, {
+    hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
+        dim3(L * F), dim3(256), 0, cur_stream(),
+        dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+        X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+        n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+  });
+  HIP_CHECK_LAST();
+}
+
+// ------------------------------------------------------------ losses --
+void logsoftmax(torch::Tensor Z, torch::Tensor P, long C) {
+  CHECK_DEV(Z); CHECK_DEV(P);
+  const long M = Z.numel() / C;
+  DISPATCH_FT(Z, {
+    hipLaunchKernelGGL(losses::logsoftmax_k<scalar_t>,
+        dim3(grid_1d(M)), dim3(256), 0, cur_stream(),
+        Z.data_ptr<scalar_t>(), P.data_ptr<scalar_t>(), M, (int)C);
+  });
+  HIP_CHECK_LAST();
+}
+
+void nll_bwd(torch::Tensor logp, torch::Tensor y, torch::Tensor dZ,
+             c10::optional<torch::Tensor> loss, long C, long B,
+             double loss_scale) {
+  CHECK_DEV(logp); CHECK_DEV(dZ);
+  const long M = logp.numel() / C;
+  DISPATCH_FT(logp, {
+    hipLaunchKernelGGL(losses::nll_bwd_k<scalar_t>,
+        dim3(grid_1d(M)), dim3(256), 0, cur_stream(),
+        logp.data_ptr<scalar_t>(), y.data_ptr<long>(),
+        dZ.data_ptr<scalar_t>(),
+        loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr,
+        M, (int)C, (int)B, (scalar_t)loss_scale);
+  });
+  HIP_CHECK_LAST();
+}
+
+void bce_bwd(torch::Tensor p, torch::Tensor tgt, torch::Tensor dZ,
+             c10::optional<torch::Tensor> loss, long B,
+             double loss_scale) {
+  CHECK_DEV(p); CHECK_DEV(dZ);
+  DISPATCH_FT(p, {
+    hipLaunchKernelGGL(losses::bce_bwd_k<scalar_t>,
+        dim3(grid_1d(p.numel())), dim3(256), 0, cur_stream(),
+        p.data_ptr<scalar_t>(), tgt.data_ptr<scalar_t>(),
+        dZ.data_ptr<scalar_t>(),
+        loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr,
+        p.numel(), (int)B, (scalar_t)loss_scale);
+  });
+  HIP_CHECK_LAST();
+}
+
+void regression_bwd(torch::Tensor yhat, torch::Tensor tgt,
+                    torch::Tensor dY, c10::optional<torch::Tensor> loss,
+                    long B, double loss_scale, long mode) {
+  CHECK_DEV(yhat); CHECK_DEV(dY);
+  DISPATCH_FT(yhat, {
+    auto lptr =
+        loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr;
+    if (mode == 0) {
+      hipLaunchKernelGGL((losses::regression_bwd_k<scalar_t, 0>),
+          dim3(grid_1d(yhat.numel())), dim3(256), 0, cur_stream(),
+          yhat.data_ptr<scalar_t>(), tgt.data_ptr<scalar_t>(),
+          dY.data_ptr<scalar_t>(), lptr, yhat.numel(), (int)B,
+          (scalar_t)loss_scale);
+    } else {
+      hipLaunchKernelGGL((losses::regression_bwd_k<scalar_t, 1>),
+          dim3(grid_1d(yhat.numel())), dim3(256), 0, cur_stream(),
+          yhat.data_ptr<scalar_t>(), tgt.data_ptr<scalar_t>(),
+          dY.data_ptr<scalar_t>(), lptr, yhat.numel(), (int)B,
+          (scalar_t)loss_scale);
+    }
+  });
+  HIP_CHECK_LAST();
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("dinno_dual_threg", &dinno_dual_threg);
+  mod.def("mix_rows", &mix_rows);
+  mod.def("dsgt_mix", &dsgt_mix);
+  mod.def("dsgt_y_update", &dsgt_y_update);
+  mod.def("fused_step", &fused_step);
+  mod.def("axpy", &axpy);
+  mod.def("linear_fwd", &linear_fwd);
+  mod.def("act_grad", &act_grad);
+  mod.def("linear_bwd_dx", &linear_bwd_dx);
+  mod.def("linear_bwd_dw", &linear_bwd_dw);
+  mod.def("conv_pool_fwd", &conv_pool_fwd);
+  mod.def("conv_pool_bwd", &conv_pool_bwd);
+  mod.def("logsoftmax", &logsoftmax);
+  mod.def("nll_bwd", &nll_bwd);
+  mod.def("bce_bwd", &bce_bwd);
+  mod.def("regression_bwd", &regression_bwd);
+}

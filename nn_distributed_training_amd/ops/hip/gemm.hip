@@ -1,0 +1,164 @@
+// Stacked linear layers: fwd/bwd GEMMs batched over the L node replicas
+// (blockIdx.z = node). Weights live INSIDE the flat [L, n] parameter
+// stack at per-layer offsets (models/spec.py), torch Linear layout
+// W[O, I] row-major + bias[O] — so the kernels read the training state
+// directly, no per-layer repacking ever happens.
+//
+// v1 kernels are LDS-tiled 16x16 VALU; gemm_mfma.hip provides the MFMA
+// fast path for the GEMM-heavy shapes and the dispatcher in bindings.cpp
+// routes between them.
+
+#include "common.h"
+
+namespace gemm {
+
+constexpr int TILE = 16;
+
+// Y[M,O] = act(X[M,I] @ W^T + b); optionally store pre-activation Z.
+// X rows are node-l's batch: X + l*M*I. W_l = theta + l*n + w_off.
+template <typename T>
+__global__ void linear_fwd_k(
+    const T* __restrict__ X, const T* __restrict__ theta,
+    T* __restrict__ Y, T* __restrict__ Z,  // Z may be null
+    long n, long w_off, long b_off, int M, int I, int O,
+    int act, T scale) {
+  __shared__ T xs[TILE][TILE + 1];
+  __shared__ T ws[TILE][TILE + 1];
+  const long l = blockIdx.z;
+  const T* Xl = X + l * (long)M * I;
+  const T* W = theta + l * n + w_off;
+  const T* b = theta + l * n + b_off;
+
+  const int m0 = blockIdx.y * TILE;
+  const int o0 = blockIdx.x * TILE;
+  const int tm = threadIdx.y, to = threadIdx.x;
+
+  T acc = T(0);
+  for (int k0 = 0; k0 < I; k0 += TILE) {
+    // xs[tm][tk] = X[m0+tm][k0+tk] ; ws[to][tk] = W[o0+to][k0+tk]
+    {
+      const int m = m0 + tm, k = k0 + to;
+      xs[tm][to] = (m < M && k < I) ? Xl[(long)m * I + k] : T(0);
+      const int o = o0 + tm;
+      ws[tm][to] = (o < O && k < I) ? W[(long)o * I + k] : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < TILE; ++k) {
+      acc += xs[tm][k] * ws[to][k];
+    }
+    __syncthreads();
+  }
+  const int m = m0 + tm, o = o0 + to;
+  if (m < M && o < O) {
+    const T z = acc + b[o];
+    if (Z != nullptr) Z[l * (long)M * O + (long)m * O + o] = z;
+    Y[l * (long)M * O + (long)m * O + o] = act_fwd(act, z, scale);
+  }
+}
+
+// dZ = dY * act'(z, y) — fused activation backward, one pass.
+template <typename T>
+__global__ void act_grad_k(
+    const T* __restrict__ dY, const T* __restrict__ Y,
+    const T* __restrict__ Z,  // may be null (derivative from Y only)
+    T* __restrict__ dZ, long total, int act, T scale) {
+  for (long t = blockIdx.x * 256L + threadIdx.x; t < total;
+       t += (long)gridDim.x * 256L) {
+    const T z = Z ? Z[t] : T(0);
+    dZ[t] = dY[t] * act_bwd(act, z, Y[t], scale);
+  }
+}
+
+// dX[M,I] = dZ[M,O] @ W[O,I]
+template <typename T>
+__global__ void linear_bwd_dx_k(
+    const T* __restrict__ dZ, const T* __restrict__ theta,
+    T* __restrict__ dX, long n, long w_off, int M, int I, int O) {
+  __shared__ T gs[TILE][TILE + 1];
+  __shared__ T ws[TILE][TILE + 1];
+  const long l = blockIdx.z;
+  const T* dZl = dZ + l * (long)M * O;
+  const T* W = theta + l * n + w_off;
+
+  const int m0 = blockIdx.y * TILE;
+  const int i0 = blockIdx.x * TILE;
+  const int tm = threadIdx.y, ti = threadIdx.x;
+
+  T acc = T(0);
+  for (int k0 = 0; k0 < O; k0 += TILE) {
+    {
+      const int m = m0 + tm, o = k0 + ti;
+      gs[tm][ti] = (m < M && o < O) ? dZl[(long)m * O + o] : T(0);
+      const int oo = k0 + tm, i = i0 + ti;
+      ws[tm][ti] = (oo < O && i < I) ? W[(long)oo * I + i] : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < TILE; ++k) {
+      acc += gs[tm][k] * ws[k][ti];
+    }
+    __syncthreads();
+  }
+  const int m = m0 + tm, i = i0 + ti;
+  if (m < M && i < I) {
+    dX[l * (long)M * I + (long)m * I + i] = acc;
+  }
+}
+
+// dW[O,I] += dZ_l[M,O]^T @ X_l[M,I], written into the [L, n] grad stack
+// at the layer's offset; db[O] = column sums of dZ (separate kernel).
+template <typename T>
+__global__ void linear_bwd_dw_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, int M, int I, int O) {
+  __shared__ T gs[TILE][TILE + 1];
+  __shared__ T xs[TILE][TILE + 1];
+  const long l = blockIdx.z;
+  const T* dZl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+
+  const int o0 = blockIdx.y * TILE;
+  const int i0 = blockIdx.x * TILE;
+  const int to = threadIdx.y, ti = threadIdx.x;
+
+  T acc = T(0);
+  for (int k0 = 0; k0 < M; k0 += TILE) {
+    {
+      const int m = k0 + to;
+      const int o = o0 + ti;
+      gs[to][ti] = (m < M && o < O) ? dZl[(long)m * O + o] : T(0);
+      const int i = i0 + ti;
+      xs[to][ti] = (m < M && i < I) ? Xl[(long)m * I + i] : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < TILE; ++k) {
+      acc += gs[k][to] * xs[k][ti];
+    }
+    __syncthreads();
+  }
+  const int o = o0 + to, i = i0 + ti;
+  if (o < O && i < I) {
+    gstack[l * n + w_off + (long)o * I + i] = acc;
+  }
+}
+
+// db[O] = sum_m dZ[m, O] into the grad stack at b_off. One block per
+// (l, o-chunk of 256); each thread owns one output column.
+template <typename T>
+__global__ void bias_grad_k(
+    const T* __restrict__ dZ, T* __restrict__ gstack,
+    long n, long b_off, int M, int O) {
+  const long l = blockIdx.z;
+  const int o = blockIdx.x * 256 + threadIdx.x;
+  if (o >= O) return;
+  const T* dZl = dZ + l * (long)M * O;
+  T acc = T(0);
+  for (int m = 0; m < M; ++m) {
+    acc += dZl[(long)m * O + o];
+  }
+  gstack[l * n + b_off + o] = acc;
+}
+
+}  // namespace gemm
