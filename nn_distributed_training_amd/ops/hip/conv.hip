@@ -73,27 +73,37 @@ __global__ void conv_pool_fwd_k(
 // Backward to weights/bias: route each pooled grad to its argmax conv
 // position, multiply by the image window. dY is the grad AFTER the relu
 // mask (pooled output > 0), applied by the caller via act_grad.
-// Grid: one block per (l, f); each thread strides over (b, py, px) and
-// accumulates a private dW[K*K]+db, then block-reduces.
+//
+// Grid: (l, f, batch-chunk) so the chip fills (the v1 one-block-per-
+// (l,f) version was 24 blocks on 256 CUs and 41% of round time);
+// each thread accumulates a private dW[K*K]+db over its strided share
+// of the chunk, waves shuffle-reduce, wave leaders combine in LDS and
+// lane 0 atomically adds into the grad stack (the caller zeroes the
+// conv slice first).
 template <typename T, int KMAX>
 __global__ void conv_pool_bwd_k(
     const T* __restrict__ dY, const unsigned char* __restrict__ idx,
     const T* __restrict__ X, T* __restrict__ gstack,
-    long n, long w_off, long b_off, int B, int F, int K, int IMG) {
-  const int l = blockIdx.x / F;
-  const int f = blockIdx.x % F;
+    long n, long w_off, long b_off, int B, int F, int K, int IMG,
+    int nchunk) {
+  const int chunk = blockIdx.x % nchunk;
+  const int f = (blockIdx.x / nchunk) % F;
+  const int l = blockIdx.x / (nchunk * F);
   const int conv_out = IMG - (K - 1);
   const int P = conv_out / 2;
   const int npool = F * P * P;
+  const int cb = (B + nchunk - 1) / nchunk;       // images per chunk
+  const int b0 = chunk * cb;
+  const int b1 = min(B, b0 + cb);
 
   T dw[KMAX * KMAX];
   T db = T(0);
   #pragma unroll
   for (int i = 0; i < KMAX * KMAX; ++i) dw[i] = T(0);
 
-  const int work = B * P * P;
+  const int work = (b1 - b0) * P * P;
   for (int t = threadIdx.x; t < work; t += blockDim.x) {
-    const int b = t / (P * P);
+    const int b = b0 + t / (P * P);
     const int py = (t / P) % P;
     const int px = t % P;
     const long lb = (long)l * B + b;
@@ -113,21 +123,22 @@ __global__ void conv_pool_bwd_k(
     }
   }
 
-  // block reduction of dw[K*K] and db through LDS
-  __shared__ T red[256];
+  // reduce the K*K+1 partials: shuffle within waves, LDS across waves
+  __shared__ T red[4];  // one slot per wave (256 threads = 4 waves)
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  T* wslice = gstack + (long)l * n + w_off + (long)f * K * K;
   for (int i = 0; i < K * K + 1; ++i) {
     T v = (i < K * K) ? dw[i] : db;
-    red[threadIdx.x] = v;
+    v = wave_reduce_sum(v);
+    if (lane == 0) red[wid] = v;
     __syncthreads();
-    for (int s = blockDim.x / 2; s > 0; s >>= 1) {
-      if (threadIdx.x < s) red[threadIdx.x] += red[threadIdx.x + s];
-      __syncthreads();
-    }
     if (threadIdx.x == 0) {
+      T tot = red[0] + red[1] + red[2] + red[3];
       if (i < K * K) {
-        gstack[(long)l * n + w_off + (long)f * K * K + i] = red[0];
+        atomicAdd(&wslice[i], tot);
       } else {
-        gstack[(long)l * n + b_off + f] = red[0];
+        atomicAdd(&gstack[(long)l * n + b_off + f], tot);
       }
     }
     __syncthreads();

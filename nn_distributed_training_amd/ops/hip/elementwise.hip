@@ -15,6 +15,15 @@ namespace ew {
 
 constexpr int BLOCK = 256;
 
+// Table rows: r < L -> local stack row, else -> received remote row.
+// Avoids ever materializing a concatenated table (the local stack is
+// read in place; remote rows land where irecv wrote them).
+template <typename T>
+DEV_INLINE const T* table_row(const T* local, const T* remote, long L,
+                              long n, int r) {
+  return r < L ? local + (long)r * n : remote + ((long)r - L) * n;
+}
+
 // ---------------------------------------------------------------------
 // DiNNO round prologue (reference optimizers/dinno.py:119-124, fused):
 //   S_i      = sum_{j in N(i)} th_j
@@ -24,7 +33,8 @@ constexpr int BLOCK = 256;
 //                                            gradient needs)
 template <typename T>
 __global__ void dinno_dual_threg_k(
-    const T* __restrict__ table,   // [R, n]; rows 0..L-1 = local snapshot
+    const T* __restrict__ local,   // [L, n] round-start snapshot
+    const T* __restrict__ remote,  // [R-L, n] received rows (may be null)
     const int* __restrict__ offs,  // [L+1]
     const int* __restrict__ idx,   // CSR neighbor rows
     T* __restrict__ duals,         // [L, n] in/out
@@ -38,9 +48,9 @@ __global__ void dinno_dual_threg_k(
     const int k0 = offs[l], k1 = offs[l + 1];
     T S = T(0);
     for (int k = k0; k < k1; ++k) {
-      S += table[(long)idx[k] * n + e];
+      S += table_row(local, remote, L, n, idx[k])[e];
     }
-    const T th = table[l * n + e];
+    const T th = local[l * n + e];
     const T deg = T(k1 - k0);
     duals[t] += rho * (deg * th - S);
     s_out[t] = (deg * th + S) * T(0.5);
@@ -53,9 +63,9 @@ __global__ void dinno_dual_threg_k(
 // (the caller includes the self row with weight W_ll in the CSR).
 template <typename T>
 __global__ void mix_rows_k(
-    const T* __restrict__ table, const int* __restrict__ offs,
-    const int* __restrict__ idx, const T* __restrict__ w,
-    T* __restrict__ out, long n, long L) {
+    const T* __restrict__ local, const T* __restrict__ remote,
+    const int* __restrict__ offs, const int* __restrict__ idx,
+    const T* __restrict__ w, T* __restrict__ out, long n, long L) {
   const long total = L * n;
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
@@ -64,7 +74,7 @@ __global__ void mix_rows_k(
     const int k0 = offs[l], k1 = offs[l + 1];
     T acc = T(0);
     for (int k = k0; k < k1; ++k) {
-      acc += w[k] * table[(long)idx[k] * n + e];
+      acc += w[k] * table_row(local, remote, L, n, idx[k])[e];
     }
     out[t] = acc;
   }
@@ -78,7 +88,9 @@ __global__ void mix_rows_k(
 // one exchange (2x comm volume).
 template <typename T>
 __global__ void dsgt_mix_k(
-    const T* __restrict__ table,  // [R, 2n]
+    const T* __restrict__ p_loc,   // [L, n]
+    const T* __restrict__ y_loc,   // [L, n]
+    const T* __restrict__ remote,  // [R-L, 2n] bundles [p | y]
     const int* __restrict__ offs, const int* __restrict__ idx,
     const T* __restrict__ w, T* __restrict__ p_out,
     T* __restrict__ y_mix, T alpha, long n, long L) {
@@ -90,9 +102,18 @@ __global__ void dsgt_mix_k(
     const int k0 = offs[l], k1 = offs[l + 1];
     T accp = T(0), accy = T(0);
     for (int k = k0; k < k1; ++k) {
-      const T* row = table + (long)idx[k] * (2 * n);
-      accp += w[k] * row[e];
-      accy += w[k] * row[n + e];
+      const int r = idx[k];
+      T pj, yj;
+      if (r < L) {
+        pj = p_loc[(long)r * n + e];
+        yj = y_loc[(long)r * n + e];
+      } else {
+        const T* row = remote + ((long)r - L) * (2 * n);
+        pj = row[e];
+        yj = row[n + e];
+      }
+      accp += w[k] * pj;
+      accy += w[k] * yj;
     }
     p_out[t] = accp - alpha * accy;
     y_mix[t] = accy;
@@ -154,6 +175,40 @@ __global__ void fused_step_k(
     v[t] = vt;
     // bc1 = 1-beta1^t, bc2 = 1-beta2^t (host-computed per step)
     theta[t] = th - lr * (mt / bc1) / (::sqrt(vt / bc2) + eps);
+  }
+}
+
+// Batch assembly: out[l*B + b, :] = X_all[l, idx[l, b], :] in one
+// launch (replaces torch arange + advanced indexing + copies). idx rows
+// may be strided views into a longer per-node index stream.
+template <typename T>
+__global__ void gather_batch_k(
+    const T* __restrict__ X_all,       // [L, maxlen, F]
+    const long* __restrict__ idx,      // [L, B] rows strided by idx_stride
+    T* __restrict__ out,               // [L*B, F]
+    long maxlen, long Fdim, long B, long idx_stride, long total) {
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long f = t % Fdim;
+    const long lb = t / Fdim;
+    const long l = lb / B;
+    const long b = lb - l * B;
+    const long src = idx[l * idx_stride + b];
+    out[t] = X_all[(l * maxlen + src) * Fdim + f];
+  }
+}
+
+// Same for targets (typed independently: long labels or T densities).
+template <typename T>
+__global__ void gather_targets_k(
+    const T* __restrict__ Y_all, const long* __restrict__ idx,
+    T* __restrict__ out, long maxlen, long B, long idx_stride,
+    long total) {
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / B;
+    const long b = t - l * B;
+    out[t] = Y_all[l * maxlen + idx[l * idx_stride + b]];
   }
 }
 

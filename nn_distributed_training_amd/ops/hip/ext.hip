@@ -30,49 +30,98 @@ inline int grid_1d(long total, int block = 256, int cap = 4096) {
                              [&] { __VA_ARGS__ });
 
 // ---------------------------------------------------------------- ew --
-void dinno_dual_threg(torch::Tensor table, torch::Tensor offs,
-                      torch::Tensor idx, torch::Tensor duals,
-                      torch::Tensor s_out, double rho) {
-  CHECK_DEV(table); CHECK_DEV(duals); CHECK_DEV(s_out);
+void dinno_dual_threg(torch::Tensor local,
+                      c10::optional<torch::Tensor> remote,
+                      torch::Tensor offs, torch::Tensor idx,
+                      torch::Tensor duals, torch::Tensor s_out,
+                      double rho) {
+  CHECK_DEV(local); CHECK_DEV(duals); CHECK_DEV(s_out);
   const long L = duals.size(0), n = duals.size(1);
-  DISPATCH_FT(table, {
+  DISPATCH_FT(local, {
     hipLaunchKernelGGL(ew::dinno_dual_threg_k<scalar_t>,
         dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
-        table.data_ptr<scalar_t>(), offs.data_ptr<int>(),
-        idx.data_ptr<int>(), duals.data_ptr<scalar_t>(),
-        s_out.data_ptr<scalar_t>(), (scalar_t)rho, n, L);
+        local.data_ptr<scalar_t>(),
+        remote.has_value() ? remote->data_ptr<scalar_t>() : nullptr,
+        offs.data_ptr<int>(), idx.data_ptr<int>(),
+        duals.data_ptr<scalar_t>(), s_out.data_ptr<scalar_t>(),
+        (scalar_t)rho, n, L);
   });
   HIP_CHECK_LAST();
 }
 
-void mix_rows(torch::Tensor table, torch::Tensor offs, torch::Tensor idx,
-              torch::Tensor w, torch::Tensor out) {
-  CHECK_DEV(table); CHECK_DEV(out);
+void mix_rows(torch::Tensor local, c10::optional<torch::Tensor> remote,
+              torch::Tensor offs, torch::Tensor idx, torch::Tensor w,
+              torch::Tensor out) {
+  CHECK_DEV(local); CHECK_DEV(out);
+  TORCH_CHECK(local.data_ptr() != out.data_ptr(),
+              "mix_rows: out must not alias the local table");
   const long L = out.size(0), n = out.size(1);
-  DISPATCH_FT(table, {
+  DISPATCH_FT(local, {
     hipLaunchKernelGGL(ew::mix_rows_k<scalar_t>,
         dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
-        table.data_ptr<scalar_t>(), offs.data_ptr<int>(),
-        idx.data_ptr<int>(), w.data_ptr<scalar_t>(),
-        out.data_ptr<scalar_t>(), n, L);
+        local.data_ptr<scalar_t>(),
+        remote.has_value() ? remote->data_ptr<scalar_t>() : nullptr,
+        offs.data_ptr<int>(), idx.data_ptr<int>(),
+        w.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(), n, L);
   });
   HIP_CHECK_LAST();
 }
 
-void dsgt_mix(torch::Tensor table, torch::Tensor offs, torch::Tensor idx,
-              torch::Tensor w, torch::Tensor p_out, torch::Tensor y_mix,
-              double alpha) {
-  CHECK_DEV(table); CHECK_DEV(p_out); CHECK_DEV(y_mix);
+void dsgt_mix(torch::Tensor p_loc, torch::Tensor y_loc,
+              c10::optional<torch::Tensor> remote, torch::Tensor offs,
+              torch::Tensor idx, torch::Tensor w, torch::Tensor p_out,
+              torch::Tensor y_mix, double alpha) {
+  CHECK_DEV(p_loc); CHECK_DEV(y_loc); CHECK_DEV(p_out); CHECK_DEV(y_mix);
+  TORCH_CHECK(p_loc.data_ptr() != p_out.data_ptr(),
+              "dsgt_mix: p_out must not alias p_loc");
   const long L = p_out.size(0), n = p_out.size(1);
-  TORCH_CHECK(table.size(1) == 2 * n, "dsgt table must be [R, 2n]");
-  DISPATCH_FT(table, {
+  DISPATCH_FT(p_loc, {
     hipLaunchKernelGGL(ew::dsgt_mix_k<scalar_t>,
         dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
-        table.data_ptr<scalar_t>(), offs.data_ptr<int>(),
-        idx.data_ptr<int>(), w.data_ptr<scalar_t>(),
-        p_out.data_ptr<scalar_t>(), y_mix.data_ptr<scalar_t>(),
-        (scalar_t)alpha, n, L);
+        p_loc.data_ptr<scalar_t>(), y_loc.data_ptr<scalar_t>(),
+        remote.has_value() ? remote->data_ptr<scalar_t>() : nullptr,
+        offs.data_ptr<int>(), idx.data_ptr<int>(),
+        w.data_ptr<scalar_t>(), p_out.data_ptr<scalar_t>(),
+        y_mix.data_ptr<scalar_t>(), (scalar_t)alpha, n, L);
   });
+  HIP_CHECK_LAST();
+}
+
+void gather_batch(torch::Tensor X_all, torch::Tensor idx,
+                  torch::Tensor out, long idx_stride) {
+  CHECK_DEV(X_all); CHECK_DEV(out);
+  const long L = X_all.size(0), maxlen = X_all.size(1),
+             Fdim = X_all.size(2);
+  const long B = out.size(0) / L;
+  const long total = out.numel();
+  DISPATCH_FT(X_all, {
+    hipLaunchKernelGGL(ew::gather_batch_k<scalar_t>,
+        dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+        X_all.data_ptr<scalar_t>(), idx.data_ptr<long>(),
+        out.data_ptr<scalar_t>(), maxlen, Fdim, B, idx_stride, total);
+  });
+  HIP_CHECK_LAST();
+}
+
+void gather_targets(torch::Tensor Y_all, torch::Tensor idx,
+                    torch::Tensor out, long idx_stride) {
+  CHECK_DEV(Y_all); CHECK_DEV(out);
+  const long L = Y_all.size(0), maxlen = Y_all.size(1);
+  const long B = out.numel() / L;
+  const long total = out.numel();
+  if (Y_all.scalar_type() == torch::kLong) {
+    hipLaunchKernelGGL(ew::gather_targets_k<long>,
+        dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+        Y_all.data_ptr<long>(), idx.data_ptr<long>(),
+        out.data_ptr<long>(), maxlen, B, idx_stride, total);
+  } else {
+    DISPATCH_FT(Y_all, {
+      hipLaunchKernelGGL(ew::gather_targets_k<scalar_t>,
+          dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+          Y_all.data_ptr<scalar_t>(), idx.data_ptr<long>(),
+          out.data_ptr<scalar_t>(), maxlen, B, idx_stride, total);
+    });
+  }
   HIP_CHECK_LAST();
 }
 
@@ -199,11 +248,8 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
     hipLaunchKernelGGL(gemm::linear_bwd_dw_k<scalar_t>,
         grid, dim3(16, 16), 0, cur_stream(),
         dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
-        gstack.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
-    hipLaunchKernelGGL(gemm::bias_grad_k<scalar_t>,
-        dim3((O + 255) / 256, 1, L), dim3(256), 0, cur_stream(),
-        dZ.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
-        n, b_off, (int)M, (int)O);
+        gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+        (int)M, (int)I, (int)O);
   });
   HIP_CHECK_LAST();
 }
@@ -232,12 +278,15 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
   CHECK_DEV(dY); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
   TORCH_CHECK(K <= 7, "conv_pool_bwd supports kernel size <= 7");
+  // chunk the batch so the grid fills the 256-CU chip; accumulation is
+  // atomic, so the caller must zero the conv slice of gstack first
+  const int nchunk = (int)std::min<long>(B, 32);
   DISPATCH_FT(dY, {
     hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
-        dim3(L * F), dim3(256), 0, cur_stream(),
+        dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
         dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
         X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
-        n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+        n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
   });
   HIP_CHECK_LAST();
 }
@@ -318,6 +367,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("dsgt_y_update", &dsgt_y_update);
   mod.def("fused_step", &fused_step);
   mod.def("axpy", &axpy);
+  mod.def("gather_batch", &gather_batch);
+  mod.def("gather_targets", &gather_targets);
   mod.def("linear_fwd", &linear_fwd);
   mod.def("act_grad", &act_grad);
   mod.def("linear_bwd_dx", &linear_bwd_dx);

@@ -106,12 +106,15 @@ __global__ void linear_bwd_dx_k(
   }
 }
 
-// dW[O,I] += dZ_l[M,O]^T @ X_l[M,I], written into the [L, n] grad stack
-// at the layer's offset; db[O] = column sums of dZ (separate kernel).
+// dW[O,I] = dZ_l[M,O]^T @ X_l[M,I], written into the [L, n] grad stack
+// at the layer's offset; db[O] (column sums of dZ) is fused: the i0==0
+// block column accumulates bias partials from its staged dZ tiles while
+// it loops over M (the separate bias kernel was an underfilled launch).
 template <typename T>
 __global__ void linear_bwd_dw_k(
     const T* __restrict__ dZ, const T* __restrict__ X,
-    T* __restrict__ gstack, long n, long w_off, int M, int I, int O) {
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O) {
   __shared__ T gs[TILE][TILE + 1];
   __shared__ T xs[TILE][TILE + 1];
   const long l = blockIdx.z;
@@ -121,8 +124,10 @@ __global__ void linear_bwd_dw_k(
   const int o0 = blockIdx.y * TILE;
   const int i0 = blockIdx.x * TILE;
   const int to = threadIdx.y, ti = threadIdx.x;
+  const bool bias_block = (i0 == 0);
 
   T acc = T(0);
+  T bacc = T(0);  // thread (to, ti): partial db[o0+ti] over rows m≡to
   for (int k0 = 0; k0 < M; k0 += TILE) {
     {
       const int m = k0 + to;
@@ -136,29 +141,24 @@ __global__ void linear_bwd_dw_k(
     for (int k = 0; k < TILE; ++k) {
       acc += gs[k][to] * xs[k][ti];
     }
+    if (bias_block) bacc += gs[to][ti];
     __syncthreads();
   }
   const int o = o0 + to, i = i0 + ti;
   if (o < O && i < I) {
     gstack[l * n + w_off + (long)o * I + i] = acc;
   }
-}
-
-// db[O] = sum_m dZ[m, O] into the grad stack at b_off. One block per
-// (l, o-chunk of 256); each thread owns one output column.
-template <typename T>
-__global__ void bias_grad_k(
-    const T* __restrict__ dZ, T* __restrict__ gstack,
-    long n, long b_off, int M, int O) {
-  const long l = blockIdx.z;
-  const int o = blockIdx.x * 256 + threadIdx.x;
-  if (o >= O) return;
-  const T* dZl = dZ + l * (long)M * O;
-  T acc = T(0);
-  for (int m = 0; m < M; ++m) {
-    acc += dZl[(long)m * O + o];
+  if (bias_block) {
+    // column-sum bacc over to (16 rows) through LDS, thread row 0 writes
+    gs[to][ti] = bacc;
+    __syncthreads();
+    if (to == 0 && o0 + ti < O) {
+      T s = T(0);
+#pragma unroll
+      for (int r = 0; r < TILE; ++r) s += gs[r][ti];
+      gstack[l * n + b_off + o0 + ti] = s;
+    }
   }
-  gstack[l * n + b_off + o] = acc;
 }
 
 }  // namespace gemm

@@ -8,17 +8,24 @@ loops over nodes / parameter tensors / autograd (optimizers/dinno.py:119-
 125, dsgd.py:37-58, dsgt.py:58-105); here a full DiNNO round on 8 MNIST
 nodes is ~30 launches regardless of node count.
 
-Data staging: each node's full local dataset is resident in HBM (288 GB
-per GPU dwarfs these workloads); batches are index-gathers. Epoch
-semantics: fixed-size batches drawn from per-node shuffled permutations,
-wrapping into a freshly shuffled permutation at epoch end (the reference's
+Memory/overhead design points (profile-driven, see profiles/):
+* node datasets are resident in HBM; batch assembly is ONE gather kernel
+  reading a strided window of a pre-shuffled per-node index stream — no
+  per-iteration torch indexing chains;
+* neighbor tables are never materialized: kernels take (local stack,
+  packed remote buffer) pointer pairs, irecv lands remote rows directly
+  in the packed buffer, and the local stack is read in place (no clone,
+  no cat in the round path);
+* mixing writes into a second stack buffer and the driver pointer-swaps.
+
+Epoch semantics: fixed-size batches from per-node shuffled permutation
+streams, wrapping into a fresh permutation at epoch end (the reference's
 DataLoader emits one short final batch per epoch instead — documented
 deviation, stream-equivalent otherwise).
 """
 
 from __future__ import annotations
 
-import numpy as np
 import torch
 
 from ..models.spec import ModelSpec, model_spec
@@ -42,10 +49,8 @@ def _dataset_tensors(ds):
         x, y = _dataset_tensors(ds.dataset)
         idx = torch.as_tensor(ds.indices)
         return x[idx], y[idx]
-    # datasets wrapping a TensorDataset under .tds (lidar datasets)
-    if hasattr(ds, "tds"):
+    if hasattr(ds, "tds"):  # lidar datasets wrap a TensorDataset
         return _dataset_tensors(ds.tds)
-    # generic fallback: materialize through a loader
     xs, ys = [], []
     for x, y in torch.utils.data.DataLoader(ds, batch_size=1024):
         xs.append(x)
@@ -53,48 +58,64 @@ def _dataset_tensors(ds):
     return torch.cat(xs), torch.cat(ys)
 
 
-class _PermSampler:
-    """Per-node epoch-permutation batch sampler on device."""
+class _StreamSampler:
+    """Per-node shuffled index streams; batches are strided views.
 
-    def __init__(self, lengths, batch, device, seed, epoch_cb):
+    Each node's stream concatenates whole permutations of its dataset;
+    a batch is stream[:, pos:pos+B] — zero kernels per draw. Streams are
+    rebuilt (host-side, amortized over stream_batches draws) when
+    exhausted. Epoch counters advance by consumed/len.
+    """
+
+    def __init__(self, lengths, batch, device, seed, epoch_cb,
+                 stream_batches=256):
         self.lengths = lengths
         self.B = batch
+        self.S = stream_batches * batch
         self.device = device
-        self.epoch_cb = epoch_cb  # called with node-local index on wrap
+        self.epoch_cb = epoch_cb
         self.gens = [
-            torch.Generator(device="cpu").manual_seed(seed * 100003 + i)
+            torch.Generator().manual_seed(seed * 100003 + i)
             for i in range(len(lengths))
         ]
-        self.perms = [
-            torch.randperm(n, generator=g).to(device)
-            for n, g in zip(lengths, self.gens)
-        ]
-        self.pos = [0] * len(lengths)
+        self.consumed = [0] * len(lengths)
+        self.reported_epochs = [0] * len(lengths)
+        self.stream = torch.empty(
+            len(lengths), self.S, dtype=torch.long, device=device
+        )
+        self.pos = self.S  # force initial fill
 
-    def next_indices(self) -> torch.Tensor:
+    def _refill(self):
         rows = []
         for li, n in enumerate(self.lengths):
-            p = self.pos[li]
-            if p + self.B <= n:
-                rows.append(self.perms[li][p : p + self.B])
-                self.pos[li] = p + self.B
-            else:
-                head = self.perms[li][p:n]
+            parts, tot = [], 0
+            while tot < self.S:
+                parts.append(torch.randperm(n, generator=self.gens[li]))
+                tot += n
+            rows.append(torch.cat(parts)[: self.S])
+        self.stream.copy_(torch.stack(rows).to(self.device))
+        self.pos = 0
+
+    def next_view(self) -> torch.Tensor:
+        """[L, B] strided view into the stream + its row stride."""
+        if self.pos + self.B > self.S:
+            self._refill()
+        view = self.stream[:, self.pos : self.pos + self.B]
+        self.pos += self.B
+        for li, n in enumerate(self.lengths):
+            self.consumed[li] += self.B
+            ep = self.consumed[li] // n
+            while self.reported_epochs[li] < ep:
+                self.reported_epochs[li] += 1
                 self.epoch_cb(li)
-                self.perms[li] = torch.randperm(
-                    n, generator=self.gens[li]
-                ).to(self.device)
-                take = self.B - head.numel()
-                rows.append(torch.cat([head, self.perms[li][:take]]))
-                self.pos[li] = take
-        return torch.stack(rows)  # [L, B]
+        return view
 
 
 class _OnlineWindowSampler:
     """Sliding-window sampler mirroring OnlineTrajectoryLidarDataset:
-    pops fixed batches from the shuffled current window; advances the
-    window (and the dataset's curr_pos, which drives the dynamic graph)
-    when the window empties."""
+    fixed batches from the shuffled current window; advances the window
+    (and the dataset's curr_pos, which drives the dynamic graph) when
+    the window empties."""
 
     def __init__(self, datasets, batch, device, seed, epoch_cb):
         self.dss = datasets
@@ -102,11 +123,14 @@ class _OnlineWindowSampler:
         self.device = device
         self.epoch_cb = epoch_cb
         self.gens = [
-            torch.Generator(device="cpu").manual_seed(seed * 100003 + i)
+            torch.Generator().manual_seed(seed * 100003 + i)
             for i in range(len(datasets))
         ]
         self.pools = [None] * len(datasets)
         self.pos = [0] * len(datasets)
+        self.buf = torch.empty(
+            len(datasets), batch, dtype=torch.long, device=device
+        )
         for li in range(len(datasets)):
             self._refill(li, first=True)
 
@@ -121,21 +145,20 @@ class _OnlineWindowSampler:
         self.pools[li] = idx[perm].to(self.device)
         self.pos[li] = 0
 
-    def next_indices(self) -> torch.Tensor:
-        rows = []
+    def next_view(self) -> torch.Tensor:
         for li in range(len(self.dss)):
             p = self.pos[li]
             pool = self.pools[li]
             if p + self.B <= pool.numel():
-                rows.append(pool[p : p + self.B])
+                self.buf[li] = pool[p : p + self.B]
                 self.pos[li] = p + self.B
             else:
                 head = pool[p:]
                 self._refill(li)
                 take = self.B - head.numel()
-                rows.append(torch.cat([head, self.pools[li][:take]]))
+                self.buf[li] = torch.cat([head, self.pools[li][:take]])
                 self.pos[li] = take
-        return torch.stack(rows)
+        return self.buf
 
 
 class StackedEngine:
@@ -155,7 +178,6 @@ class StackedEngine:
         first = self.local_nodes[0]
         self.spec: ModelSpec = model_spec(problem.models[first])
 
-        # parameter stack from the replicas (they all start identical)
         rows = [
             torch.nn.utils.parameters_to_vector(
                 problem.models[i].parameters()
@@ -215,13 +237,12 @@ class StackedEngine:
                 online, B, self.device, seed, epoch_cb
             )
         else:
-            self.sampler = _PermSampler(
+            self.sampler = _StreamSampler(
                 self.lengths, B, self.device, seed, epoch_cb
             )
 
     # ------------------------------------------------------------------
     def _alloc_bufs(self):
-        """Pre-allocate activation/grad workspaces for batch size B."""
         L, B = self.L, self.B
         mk = lambda el: torch.empty(
             L * B, el, device=self.device, dtype=self.dtype
@@ -229,23 +250,27 @@ class StackedEngine:
         acts, zs, dzs, idxs = [], [], [], []
         for layer in self.spec.layers:
             acts.append(mk(layer.out_elems))
-            need_z = layer.activation in ("sin_relu",) or (
-                layer.activation == "none"
+            zs.append(
+                mk(layer.out_elems)
+                if layer.activation == "sin_relu"
+                else None
             )
-            zs.append(mk(layer.out_elems) if layer.activation == "sin_relu"
-                      else None)
             dzs.append(mk(layer.out_elems))
-            if layer.kind == "conv_pool":
-                idxs.append(
-                    torch.empty(
-                        L * B, layer.out_elems, device=self.device,
-                        dtype=torch.uint8,
-                    )
+            idxs.append(
+                torch.empty(
+                    L * B, layer.out_elems, device=self.device,
+                    dtype=torch.uint8,
                 )
-            else:
-                idxs.append(None)
+                if layer.kind == "conv_pool"
+                else None
+            )
         self._bufs = {
             "acts": acts, "zs": zs, "dzs": dzs, "idxs": idxs,
+            "xb": mk(self.spec.in_elems),
+            "yb": torch.empty(
+                L * B, device=self.device,
+                dtype=torch.long if self.classification else self.dtype,
+            ),
             "logp": mk(self.spec.layers[-1].out_elems)
             if self.classification else None,
             "loss": torch.zeros(L, device=self.device, dtype=self.dtype),
@@ -253,17 +278,20 @@ class StackedEngine:
 
     # ------------------------------------------------------------------
     def next_batch(self):
-        idx = self.sampler.next_indices()  # [L, B]
-        ar = torch.arange(self.L, device=self.device).unsqueeze(1)
-        xb = self.X_all[ar, idx].reshape(self.L * self.B, -1).contiguous()
-        yb = self.Y_all[ar, idx].reshape(-1).contiguous()
+        """Assemble the next per-node batches with two gather kernels."""
+        if self._bufs is None:
+            self._alloc_bufs()
+        view = self.sampler.next_view()  # [L, B] (possibly strided)
+        stride = view.stride(0)
+        xb, yb = self._bufs["xb"], self._bufs["yb"]
+        self.ext.gather_batch(self.X_all, view, xb, stride)
+        self.ext.gather_targets(self.Y_all, view, yb, stride)
         if 0 in self.local_nodes:
             self.pr.forward_cnt += self.pr.conf["train_batch_size"]
         return xb, yb
 
     # ------------------------------------------------------------------
     def forward(self, xb):
-        """Stacked forward through the spec; fills activation buffers."""
         if self._bufs is None:
             self._alloc_bufs()
         bufs = self._bufs
@@ -281,7 +309,6 @@ class StackedEngine:
             else:
                 act = layer.activation
                 if act == "logsoftmax":
-                    # fc produces logits (act none), then row logsoftmax
                     ext.linear_fwd(
                         cur, self.theta, out, None, layer.w_off,
                         layer.b_off, M, layer.in_dim, layer.out_dim,
@@ -299,7 +326,6 @@ class StackedEngine:
 
     # ------------------------------------------------------------------
     def backward(self, xb, yb, loss_scale=1.0, want_loss=False):
-        """Loss grad + full backward; fills self.grad [L, n]."""
         bufs = self._bufs
         ext = self.ext
         layers = self.spec.layers
@@ -335,12 +361,17 @@ class StackedEngine:
                     ACT_IDS[last.activation], last.scale,
                 )
 
-        # walk layers in reverse
         for li in range(nl - 1, -1, -1):
             layer = layers[li]
             below = bufs["acts"][li - 1] if li > 0 else xb
             dz = bufs["dzs"][li]
             if layer.kind == "conv_pool":
+                # atomic accumulation: zero this layer's grad slice
+                w0 = layer.w_off
+                cnt = (
+                    layer.out_dim * layer.kernel_size**2 + layer.out_dim
+                )
+                self.grad[:, w0 : w0 + cnt].zero_()
                 ext.conv_pool_bwd(
                     dz, bufs["idxs"][li], below, self.grad,
                     layer.w_off, layer.b_off, M, layer.out_dim,
@@ -352,7 +383,7 @@ class StackedEngine:
                 layer.in_dim, layer.out_dim,
             )
             if li > 0:
-                dy_below = bufs["dzs"][li - 1]  # reuse as dY scratch
+                dy_below = bufs["dzs"][li - 1]
                 ext.linear_bwd_dx(
                     dz, self.theta, dy_below, layer.w_off, M,
                     layer.in_dim, layer.out_dim,
@@ -368,7 +399,6 @@ class StackedEngine:
 
     # ------------------------------------------------------------------
     def flush_to_models(self):
-        """Write the stack back into the torch replicas (eval/ckpt)."""
         for li, i in enumerate(self.local_nodes):
             torch.nn.utils.vector_to_parameters(
                 self.theta[li].to(torch.get_default_dtype()),
@@ -376,23 +406,26 @@ class StackedEngine:
             )
 
     # ------------------------------------------------------------------
-    def build_table(self, snapshot, remote, width=None):
-        """[R, d] table = local snapshot rows + remote rows (sorted by
-        node id); returns (table, row_of: node -> row)."""
-        width = width if width is not None else snapshot.shape[1]
-        row_of = {
-            i: li for li, i in enumerate(self.local_nodes)
-        }
-        rows = [snapshot]
-        for r, j in enumerate(sorted(remote)):
-            row_of[j] = self.L + r
-            rows.append(remote[j].reshape(1, -1))
-        table = torch.cat(rows, dim=0).contiguous()
-        return table, row_of
+    # neighbor-table plumbing (packed remote buffers, CSR over rows)
+    def remote_plan(self, width_factor=1):
+        """(remote_nodes sorted, remote_buf [R, w*n], dests {node: row})
+        for the current graph; None remote_buf when nothing is remote."""
+        if self.pr.comm.world == 1:
+            return [], None, {}
+        _, recv_nodes = self.pr.comm.edge_transfers(
+            self.pr.layout, list(self.pr.graph.edges())
+        )
+        remote_nodes = sorted(recv_nodes)
+        if not remote_nodes:
+            return [], None, {}
+        buf = torch.empty(
+            len(remote_nodes), width_factor * self.n,
+            device=self.device, dtype=self.dtype,
+        )
+        dests = {j: buf[r] for r, j in enumerate(remote_nodes)}
+        return remote_nodes, buf, dests
 
     def build_csr(self, row_of, include_self=False, W=None):
-        """CSR neighbor lists over table rows for this rank's nodes.
-        With W (mixing matrix) also returns per-entry weights."""
         offs = [0]
         idx = []
         wts = []
@@ -416,6 +449,12 @@ class StackedEngine:
         )
         return offs_t, idx_t, w_t
 
+    def row_map(self, remote_nodes):
+        row_of = {i: li for li, i in enumerate(self.local_nodes)}
+        for r, j in enumerate(remote_nodes):
+            row_of[j] = self.L + r
+        return row_of
+
     def degrees(self) -> torch.Tensor:
         return torch.tensor(
             [self.pr.graph.degree(i) for i in self.local_nodes],
@@ -428,6 +467,12 @@ class StackedEngine:
 # ======================================================================
 
 _OPT_MODE = {"adam": 0, "adamw": 1, "sgd": 2}
+
+
+def _graph_is_static(pr):
+    from ..problems.base import ProblemBase
+
+    return type(pr).update_graph is ProblemBase.update_graph
 
 
 class DiNNOStackedDriver:
@@ -452,37 +497,36 @@ class DiNNOStackedDriver:
         self.v = torch.zeros_like(eng.theta)
         self.rho = conf["rho_init"]
         self.step_t = 0
-        # static-graph fast path: CSR/degrees built once
-        from ..problems.base import ProblemBase
+        self._static = _graph_is_static(self.pr)
+        self._plan = None
 
-        self._static = (
-            type(self.pr).update_graph is ProblemBase.update_graph
-        )
-        self._csr = None
+    def _round_plan(self):
+        if self._static and self._plan is not None:
+            return self._plan
+        eng = self.eng
+        remote_nodes, rbuf, dests = eng.remote_plan()
+        row_of = eng.row_map(remote_nodes)
+        offs, idx, _ = eng.build_csr(row_of, include_self=False)
+        deg = eng.degrees()
+        plan = (rbuf, dests, offs, idx, deg)
+        if self._static:
+            self._plan = plan
+        return plan
 
     def step_round(self, k):
-        """One full DiNNO communication round (no evaluation)."""
         opt, pr, eng = self.opt, self.pr, self.eng
         ext = eng.ext
 
-        snapshot = eng.theta.clone()
         self.rho *= opt.rho_scaling
         pr.update_graph()
-
-        remote = pr.comm.exchange_node_vectors(
-            pr.layout, list(pr.graph.edges()), snapshot
+        rbuf, dests, offs, idx, deg = self._round_plan()
+        if pr.comm.world > 1:
+            pr.comm.exchange_rows(
+                pr.layout, list(pr.graph.edges()), [eng.theta], [dests]
+            )
+        ext.dinno_dual_threg(
+            eng.theta, rbuf, offs, idx, self.duals, self.s, self.rho
         )
-        if self._static and self._csr is not None:
-            offs, idx, deg, row_of = self._csr
-            table, _ = eng.build_table(snapshot, remote)
-        else:
-            table, row_of = eng.build_table(snapshot, remote)
-            offs, idx, _ = eng.build_csr(row_of, include_self=False)
-            deg = eng.degrees()
-            if self._static:
-                self._csr = (offs, idx, deg, row_of)
-        ext.dinno_dual_threg(table, offs, idx, self.duals, self.s,
-                             self.rho)
 
         if not self.persistent:
             self.m.zero_()
@@ -526,86 +570,135 @@ class DSGDStackedDriver:
         self.pr = pr
         self.eng: StackedEngine = pr.stacked
 
-    def run(self, profiler=None):
+    def prepare(self):
+        self.alph = self.opt.alph0
+        self.theta_next = torch.empty_like(self.eng.theta)
+        self._static = _graph_is_static(self.pr)
+        self._plan = None
+
+    def _round_plan(self):
         from ..utils import graph_generation
 
+        if self._static and self._plan is not None:
+            return self._plan
+        eng = self.eng
+        W = graph_generation.get_metropolis(self.pr.graph)
+        remote_nodes, rbuf, dests = eng.remote_plan()
+        row_of = eng.row_map(remote_nodes)
+        offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
+        plan = (rbuf, dests, offs, idx, w)
+        if self._static:
+            self._plan = plan
+        return plan
+
+    def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
         ext = eng.ext
-        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
-        oits = opt.conf["outer_iterations"]
-        alph = opt.alph0
+        pr.update_graph()
+        rbuf, dests, offs, idx, w = self._round_plan()
+        self.alph = self.alph * (1 - opt.mu * self.alph)
+        if pr.comm.world > 1:
+            pr.comm.exchange_rows(
+                pr.layout, list(pr.graph.edges()), [eng.theta], [dests]
+            )
+        ext.mix_rows(eng.theta, rbuf, offs, idx, w, self.theta_next)
+        eng.theta, self.theta_next = self.theta_next, eng.theta
 
+        xb, yb = eng.next_batch()
+        eng.forward(xb)
+        eng.backward(xb, yb)
+        ext.axpy(eng.theta, eng.grad, -self.alph)
+
+    def run(self, profiler=None):
+        pr = self.pr
+        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
+        oits = self.opt.conf["outer_iterations"]
+        self.prepare()
         for k in range(oits):
             if k % eval_every == 0 or k == oits - 1:
                 pr.evaluate_metrics(at_end=(k == oits - 1))
-
-            pr.update_graph()
-            W = graph_generation.get_metropolis(pr.graph)
-            alph = alph * (1 - opt.mu * alph)
-
-            snapshot = eng.theta.clone()
-            remote = pr.comm.exchange_node_vectors(
-                pr.layout, list(pr.graph.edges()), snapshot
-            )
-            table, row_of = eng.build_table(snapshot, remote)
-            offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
-            ext.mix_rows(table, offs, idx, w, eng.theta)
-
-            xb, yb = eng.next_batch()
-            eng.forward(xb)
-            eng.backward(xb, yb)
-            ext.axpy(eng.theta, eng.grad, -alph)
-
+            self.step_round(k)
             if profiler is not None:
                 profiler.step()
 
 
 class DSGTStackedDriver:
-    """DSGT round with fused (p, y) mixing kernels."""
+    """DSGT round with fused (p, y) mixing kernels.
+
+    Params and tracker ride one batched exchange (two rows per edge =
+    the algorithm's 2x comm volume)."""
 
     def __init__(self, dsgt, pr):
         self.opt = dsgt
         self.pr = pr
         self.eng: StackedEngine = pr.stacked
 
-    def run(self, profiler=None):
-        from ..utils import graph_generation
-
-        opt, pr, eng = self.opt, self.pr, self.eng
-        ext = eng.ext
-        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
-        oits = opt.conf["outer_iterations"]
-        alpha = opt.alpha
-
-        y = torch.zeros_like(eng.theta)
-        g = torch.zeros_like(eng.theta)
-        if opt.conf["init_grads"]:
+    def prepare(self):
+        eng = self.eng
+        self.alpha = self.opt.alpha
+        self.y = torch.zeros_like(eng.theta)
+        self.g = torch.zeros_like(eng.theta)
+        self.y_mix = torch.zeros_like(eng.theta)
+        self.theta_next = torch.empty_like(eng.theta)
+        self._static = _graph_is_static(self.pr)
+        self._plan = None
+        if self.opt.conf["init_grads"]:
             xb, yb = eng.next_batch()
             eng.forward(xb)
             eng.backward(xb, yb)
-            y.copy_(eng.grad)
-            g.copy_(eng.grad)
+            self.y.copy_(eng.grad)
+            self.g.copy_(eng.grad)
 
-        y_mix = torch.zeros_like(eng.theta)
+    def _round_plan(self):
+        from ..utils import graph_generation
+
+        if self._static and self._plan is not None:
+            return self._plan
+        eng = self.eng
+        W = graph_generation.get_metropolis(self.pr.graph)
+        remote_nodes, rbuf, _ = eng.remote_plan(width_factor=2)
+        # destination views: [p | y] halves of each remote bundle row
+        n = eng.n
+        dests_p = {j: rbuf[r, :n] for r, j in enumerate(remote_nodes)} \
+            if rbuf is not None else {}
+        dests_y = {j: rbuf[r, n:] for r, j in enumerate(remote_nodes)} \
+            if rbuf is not None else {}
+        row_of = eng.row_map(remote_nodes)
+        offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
+        plan = (rbuf, dests_p, dests_y, offs, idx, w)
+        if self._static:
+            self._plan = plan
+        return plan
+
+    def step_round(self, k):
+        opt, pr, eng = self.opt, self.pr, self.eng
+        ext = eng.ext
+        pr.update_graph()
+        rbuf, dests_p, dests_y, offs, idx, w = self._round_plan()
+        if pr.comm.world > 1:
+            pr.comm.exchange_rows(
+                pr.layout, list(pr.graph.edges()),
+                [eng.theta, self.y], [dests_p, dests_y],
+            )
+        ext.dsgt_mix(
+            eng.theta, self.y, rbuf, offs, idx, w, self.theta_next,
+            self.y_mix, self.alpha,
+        )
+        eng.theta, self.theta_next = self.theta_next, eng.theta
+
+        xb, yb = eng.next_batch()
+        eng.forward(xb)
+        eng.backward(xb, yb)
+        ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
+
+    def run(self, profiler=None):
+        pr = self.pr
+        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
+        oits = self.opt.conf["outer_iterations"]
+        self.prepare()
         for k in range(oits):
             if k % eval_every == 0 or k == oits - 1:
                 pr.evaluate_metrics(at_end=(k == oits - 1))
-
-            pr.update_graph()
-            W = graph_generation.get_metropolis(pr.graph)
-
-            bundle = torch.cat([eng.theta, y], dim=1)
-            remote = pr.comm.exchange_node_vectors(
-                pr.layout, list(pr.graph.edges()), bundle
-            )
-            table, row_of = eng.build_table(bundle, remote)
-            offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
-            ext.dsgt_mix(table, offs, idx, w, eng.theta, y_mix, alpha)
-
-            xb, yb = eng.next_batch()
-            eng.forward(xb)
-            eng.backward(xb, yb)
-            ext.dsgt_y_update(y_mix, eng.grad, g, y)
-
+            self.step_round(k)
             if profiler is not None:
                 profiler.step()

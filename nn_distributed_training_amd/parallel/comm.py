@@ -79,6 +79,63 @@ class Communicator:
             self.world = 1
 
     # ------------------------------------------------------------------
+    def edge_transfers(self, layout: NodeLayout, edges: Sequence[tuple]):
+        """(send_pairs, recv_nodes) for this rank given the edge list.
+
+        send_pairs: {(local_node, peer_rank)}; recv_nodes: {remote_node}.
+        Both endpoints of an edge derive the identical sorted transfer
+        list, so batched P2P ops pair up deterministically.
+        """
+        r = self.rank
+        recv_nodes = set()
+        send_pairs = set()
+        for a, b in edges:
+            ra, rb = layout.rank_of(a), layout.rank_of(b)
+            if ra == rb:
+                continue
+            if ra == r:
+                recv_nodes.add(b)
+                send_pairs.add((a, rb))
+            elif rb == r:
+                recv_nodes.add(a)
+                send_pairs.add((b, ra))
+        return send_pairs, recv_nodes
+
+    def exchange_rows(
+        self,
+        layout: NodeLayout,
+        edges: Sequence[tuple],
+        stacks: Sequence[torch.Tensor],
+        dests: Sequence[Dict[int, torch.Tensor]],
+    ) -> None:
+        """Batched P2P exchange of per-node rows for several stacks.
+
+        ``stacks[k]`` is this rank's [L, d_k] tensor; ``dests[k]`` maps
+        each remote neighbor node id to the (contiguous) destination row
+        irecv writes into.  All sends/recvs of the round go in ONE
+        batch_isend_irecv group so RCCL can spread them across xGMI
+        links. No-op when world == 1.
+        """
+        if self.world == 1:
+            return
+        send_pairs, recv_nodes = self.edge_transfers(layout, edges)
+        r = self.rank
+        ops: List[dist.P2POp] = []
+        for node, peer in sorted(send_pairs, key=lambda t: (t[1], t[0])):
+            li = layout.local_index(node, r)
+            for st in stacks:
+                ops.append(
+                    dist.P2POp(dist.isend, st[li].contiguous(), peer)
+                )
+        for j in sorted(recv_nodes, key=lambda j: (layout.rank_of(j), j)):
+            peer = layout.rank_of(j)
+            for d in dests:
+                ops.append(dist.P2POp(dist.irecv, d[j], peer))
+        if ops:
+            for work in dist.batch_isend_irecv(ops):
+                work.wait()
+
+    # ------------------------------------------------------------------
     def exchange_node_vectors(
         self,
         layout: NodeLayout,

@@ -224,6 +224,8 @@ def test_dinno_dual_threg(ext, dtype):
     L, n, R = 3, 101, 5
     dev = _dev()
     table = torch.randn(R, n, dtype=dtype, device=dev)
+    local = table[:L].contiguous()
+    remote = table[L:].contiguous()
     # node 0 -> rows {1, 3}; node 1 -> rows {0, 4}; node 2 -> {}
     offs = torch.tensor([0, 2, 4, 4], dtype=torch.int32, device=dev)
     idx = torch.tensor([1, 3, 0, 4], dtype=torch.int32, device=dev)
@@ -231,7 +233,7 @@ def test_dinno_dual_threg(ext, dtype):
     duals0 = duals.clone()
     s = torch.empty_like(duals)
     rho = 0.37
-    ext.dinno_dual_threg(table, offs, idx, duals, s, rho)
+    ext.dinno_dual_threg(local, remote, offs, idx, duals, s, rho)
 
     nbrs = [[1, 3], [0, 4], []]
     for l in range(L):
@@ -251,29 +253,39 @@ def test_mix_and_dsgt_kernels(ext, dtype):
     L, n, R = 2, 77, 4
     dev = _dev()
     table = torch.randn(R, n, dtype=dtype, device=dev)
+    local = table[:L].contiguous()
+    remote = table[L:].contiguous()
     offs = torch.tensor([0, 3, 5], dtype=torch.int32, device=dev)
     idx = torch.tensor([0, 2, 3, 1, 2], dtype=torch.int32, device=dev)
     w = torch.randn(5, dtype=dtype, device=dev)
     out = torch.empty(L, n, dtype=dtype, device=dev)
-    ext.mix_rows(table, offs, idx, w, out)
+    ext.mix_rows(local, remote, offs, idx, w, out)
     ref0 = w[0] * table[0] + w[1] * table[2] + w[2] * table[3]
     ref1 = w[3] * table[1] + w[4] * table[2]
     torch.testing.assert_close(out[0], ref0, **TOL[dtype])
     torch.testing.assert_close(out[1], ref1, **TOL[dtype])
 
-    # dsgt: bundle table [R, 2n]
-    tab2 = torch.randn(R, 2 * n, dtype=dtype, device=dev)
+    # dsgt: local (p, y) stacks + remote [p | y] bundles
+    p_loc = torch.randn(L, n, dtype=dtype, device=dev)
+    y_loc = torch.randn(L, n, dtype=dtype, device=dev)
+    rem2 = torch.randn(R - L, 2 * n, dtype=dtype, device=dev)
     alpha = 0.05
     p_out = torch.empty(L, n, dtype=dtype, device=dev)
     y_mix = torch.empty(L, n, dtype=dtype, device=dev)
-    ext.dsgt_mix(tab2, offs, idx, w, p_out, y_mix, alpha)
+    ext.dsgt_mix(p_loc, y_loc, rem2, offs, idx, w, p_out, y_mix, alpha)
+
+    def row_py(r):
+        if r < L:
+            return p_loc[r], y_loc[r]
+        return rem2[r - L, :n], rem2[r - L, n:]
+
     for l, ks in enumerate([[0, 1, 2], [3, 4]]):
         accp = torch.zeros(n, dtype=dtype, device=dev)
         accy = torch.zeros(n, dtype=dtype, device=dev)
         for k in ks:
-            row = tab2[idx[k].item()]
-            accp += w[k] * row[:n]
-            accy += w[k] * row[n:]
+            pj, yj = row_py(idx[k].item())
+            accp += w[k] * pj
+            accy += w[k] * yj
         torch.testing.assert_close(p_out[l], accp - alpha * accy,
                                    **TOL[dtype])
         torch.testing.assert_close(y_mix[l], accy, **TOL[dtype])
@@ -286,6 +298,29 @@ def test_mix_and_dsgt_kernels(ext, dtype):
     ext.dsgt_y_update(y_mix2, g_new, g_old, y)
     torch.testing.assert_close(y, y_mix2 + g_new - g_old0, **TOL[dtype])
     torch.testing.assert_close(g_old, g_new, **TOL[dtype])
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+def test_gather_batch(ext, dtype):
+    torch.manual_seed(9)
+    L, maxlen, Fdim, B, S = 3, 50, 7, 8, 40
+    dev = _dev()
+    X = torch.randn(L, maxlen, Fdim, dtype=dtype, device=dev)
+    Yt = torch.randint(0, 10, (L, maxlen), device=dev)
+    stream = torch.randint(0, maxlen, (L, S), device=dev)
+    view = stream[:, 5 : 5 + B]  # strided view (row stride S)
+    xb = torch.empty(L * B, Fdim, dtype=dtype, device=dev)
+    yb = torch.empty(L * B, dtype=torch.long, device=dev)
+    ext.gather_batch(X, view, xb, view.stride(0))
+    ext.gather_targets(Yt, view, yb, view.stride(0))
+    ar = torch.arange(L, device=dev).unsqueeze(1)
+    torch.testing.assert_close(
+        xb, X[ar, view].reshape(L * B, Fdim), rtol=0, atol=0
+    )
+    torch.testing.assert_close(
+        yb, Yt[ar, view].reshape(-1), rtol=0, atol=0
+    )
 
 
 @requires_gpu
