@@ -65,22 +65,37 @@ def synthetic_waypoints(
         return img[iy, ix] < 0.5
 
     del xs, ys
+
+    def path_is_free(wp):
+        """Whole interpolated trajectory must stay in free space (the
+        spline can cut through walls between free waypoints)."""
+        from .lidar import interpolate_waypoints
+
+        traj = interpolate_waypoints(wp[:, 0], wp[:, 1], 8)
+        return all(is_free(x, y) for x, y in traj)
+
     paths = []
     for _ in range(num_nodes):
-        # sample an angular loop around a random free center
-        for _attempt in range(200):
+        wp = None
+        for _attempt in range(500):
             cx, cy = rng.uniform(-0.45, 0.45, size=2)
-            r0 = rng.uniform(0.1, 0.3)
-            angles = np.linspace(0, 2 * np.pi, points_per_path, endpoint=False)
-            radii = r0 * rng.uniform(0.6, 1.1, size=points_per_path)
-            px = cx + radii * np.cos(angles)
-            py = cy + radii * np.sin(angles)
-            px = np.clip(px, -0.8, 0.8)
-            py = np.clip(py, -0.8, 0.8)
-            if all(is_free(x, y) for x, y in zip(px, py)):
+            r0 = rng.uniform(0.08, 0.25)
+            angles = np.linspace(
+                0, 2 * np.pi, points_per_path, endpoint=False
+            )
+            radii = r0 * rng.uniform(0.8, 1.05, size=points_per_path)
+            px = np.clip(cx + radii * np.cos(angles), -0.8, 0.8)
+            py = np.clip(cy + radii * np.sin(angles), -0.8, 0.8)
+            cand = np.stack(
+                [np.append(px, px[0]), np.append(py, py[0])], axis=1
+            )
+            if path_is_free(cand):
+                wp = cand
                 break
-        wp = np.stack(
-            [np.append(px, px[0]), np.append(py, py[0])], axis=1
-        )
+        if wp is None:
+            raise RuntimeError(
+                "could not place a wall-free trajectory; reduce "
+                "num_walls or wall density"
+            )
         paths.append(wp)
     return paths
