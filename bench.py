@@ -42,10 +42,16 @@ def parse_args():
     p.add_argument("--graph", default="ring",
                    choices=["ring", "complete", "random"])
     p.add_argument("--dtype", default="fp64", choices=["fp64", "fp32"])
-    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--batch", type=int, default=0,
+                   help="0 = workload default (64 mnist / 20000 density)")
     p.add_argument("--engine", default="auto",
                    choices=["auto", "torch", "hip"])
     p.add_argument("--samples-per-node", type=int, default=2048)
+    p.add_argument(
+        "--workload", default="mnist", choices=["mnist", "density"],
+        help="mnist = DiNNO MNIST 8-node (headline, BASELINE cfg 3/5); "
+             "density = DSGT online implicit density (BASELINE cfg 4)",
+    )
     return p.parse_args()
 
 
@@ -70,76 +76,13 @@ def main():
         device = torch.device("cpu")
 
     torch.manual_seed(0)
-    N = args.nodes
-    if args.graph == "ring":
-        graph = nx.cycle_graph(N)
-    elif args.graph == "complete":
-        graph = nx.complete_graph(N)
-    else:
-        graph = nx.erdos_renyi_graph(N, 0.5, seed=1)
-
-    from nn_distributed_training_amd.data.mnist import (
-        SyntheticMNIST,
-        split_train_set,
-    )
-    from nn_distributed_training_amd.models import MNISTConvNet
-    from nn_distributed_training_amd.optimizers.dinno import DiNNO
-    from nn_distributed_training_amd.problems.dist_mnist_problem import (
-        DistMNISTProblem,
-    )
-
-    train = SyntheticMNIST(args.samples_per_node * N, seed=0)
-    val = SyntheticMNIST(1024, seed=1)
-    subsets = split_train_set(train, N, "hetero" if N <= 10 else
-                              "hetero_sorted")
-    base_model = MNISTConvNet(3, 5, 64)
-
-    opt_conf = {
-        "alg_name": "dinno",
-        "rho_init": 0.5,
-        "rho_scaling": 1.0003,
-        "outer_iterations": args.warmup + args.steps,
-        "primal_iterations": 2,
-        "primal_optimizer": "adam",
-        "persistant_primal_opt": False,
-        "primal_lr_start": 0.005,
-        "primal_lr_finish": 0.0005,
-        "lr_decay_type": "log",
-        "profile": False,
-    }
-    prob_conf = {
-        "problem_name": "bench_dinno",
-        "train_batch_size": args.batch,
-        "val_batch_size": 256,
-        "data_seed": 0,
-        "verbose_evals": False,
-        "metrics": ["consensus_error", "validation_loss",
-                    "top1_accuracy"],
-        "metrics_config": {"evaluate_frequency": 10**9},
-        "optimizer_config": opt_conf,
-    }
-
-    pr = DistMNISTProblem(
-        graph, base_model, torch.nn.NLLLoss(), subsets, val, device,
-        prob_conf,
-    )
     use_hip = args.engine == "hip" or (
         args.engine == "auto" and device.type == "cuda"
     )
-    if use_hip:
-        from nn_distributed_training_amd.ops.stacked import (
-            DiNNOStackedDriver,
-            StackedEngine,
-        )
-
-        pr.stacked = StackedEngine(pr)
-        opt = DiNNO(pr, device, opt_conf)
-        driver = DiNNOStackedDriver(opt, pr)
-        driver.prepare()
-        step_fn = driver.step_round
+    if args.workload == "mnist":
+        pr, step_fn, wl_cfg = _build_mnist(args, device, use_hip)
     else:
-        opt = DiNNO(pr, device, opt_conf)
-        step_fn = _golden_round_fn(opt, pr)
+        pr, step_fn, wl_cfg = _build_density(args, device, use_hip)
 
     def sync():
         if device.type == "cuda":
@@ -165,10 +108,17 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    # post-timing quality metrics (val acc + consensus error)
+    # post-timing quality metrics (val acc/loss + consensus error)
     pr.evaluate_metrics(at_end=True)
-    accs = pr.metrics["top1_accuracy"][-1]
     cons = pr.metrics["consensus_error"][-1][1]
+    quality = {"consensus_err_max": float(cons.amax())}
+    if "top1_accuracy" in pr.metrics:
+        accs = pr.metrics["top1_accuracy"][-1]
+        quality["val_acc_min"] = round(float(accs.amin()), 4)
+        quality["val_acc_max"] = round(float(accs.amax()), 4)
+    if "validation_loss" in pr.metrics and pr.metrics["validation_loss"]:
+        vl = pr.metrics["validation_loss"][-1]
+        quality["val_loss_max"] = round(float(vl.amax()), 4)
 
     if rank == 0:
         rounds_per_sec = args.steps / elapsed
@@ -186,24 +136,224 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "model": "MNISTConvNet(3,5,64) n=28440",
-                "alg": "dinno",
-                "nodes": N,
-                "graph": args.graph,
-                "global_batch": args.batch * N,
-                "primal_iterations": 2,
-                "parallelism": f"graph-decentralized dp, {N} nodes on "
+                **wl_cfg,
+                "parallelism": f"graph-decentralized dp, "
+                               f"{wl_cfg['nodes']} nodes on "
                                f"{world} rank(s)",
                 "engine": "hip-stacked" if use_hip else "torch-golden",
-                "val_acc_min": round(float(accs.amin()), 4),
-                "val_acc_max": round(float(accs.amax()), 4),
-                "consensus_err_max": float(cons.amax()),
+                **quality,
             },
         }
         print(json.dumps(out), flush=True)
 
     if world > 1:
         dist.destroy_process_group()
+
+
+def _make_graph(args, N):
+    if args.graph == "ring":
+        return nx.cycle_graph(N)
+    if args.graph == "complete":
+        return nx.complete_graph(N)
+    return nx.erdos_renyi_graph(N, 0.5, seed=1)
+
+
+def _build_mnist(args, device, use_hip):
+    from nn_distributed_training_amd.data.mnist import (
+        SyntheticMNIST,
+        split_train_set,
+    )
+    from nn_distributed_training_amd.models import MNISTConvNet
+    from nn_distributed_training_amd.optimizers.dinno import DiNNO
+    from nn_distributed_training_amd.problems.dist_mnist_problem import (
+        DistMNISTProblem,
+    )
+
+    N = args.nodes
+    B = args.batch or 64
+    graph = _make_graph(args, N)
+    train = SyntheticMNIST(args.samples_per_node * N, seed=0)
+    val = SyntheticMNIST(1024, seed=1)
+    subsets = split_train_set(train, N, "hetero" if N <= 10 else
+                              "hetero_sorted")
+    base_model = MNISTConvNet(3, 5, 64)
+
+    opt_conf = {
+        "alg_name": "dinno",
+        "rho_init": 0.5,
+        "rho_scaling": 1.0003,
+        "outer_iterations": args.warmup + args.steps,
+        "primal_iterations": 2,
+        "primal_optimizer": "adam",
+        "persistant_primal_opt": False,
+        "primal_lr_start": 0.005,
+        "primal_lr_finish": 0.0005,
+        "lr_decay_type": "log",
+        "profile": False,
+    }
+    prob_conf = {
+        "problem_name": "bench_dinno",
+        "train_batch_size": B,
+        "val_batch_size": 256,
+        "data_seed": 0,
+        "verbose_evals": False,
+        "metrics": ["consensus_error", "validation_loss",
+                    "top1_accuracy"],
+        "metrics_config": {"evaluate_frequency": 10**9},
+        "optimizer_config": opt_conf,
+    }
+    pr = DistMNISTProblem(
+        graph, base_model, torch.nn.NLLLoss(), subsets, val, device,
+        prob_conf,
+    )
+    if use_hip:
+        from nn_distributed_training_amd.ops.stacked import (
+            DiNNOStackedDriver,
+            StackedEngine,
+        )
+
+        pr.stacked = StackedEngine(pr)
+        driver = DiNNOStackedDriver(DiNNO(pr, device, opt_conf), pr)
+        driver.prepare()
+        step_fn = driver.step_round
+    else:
+        step_fn = _golden_round_fn(DiNNO(pr, device, opt_conf), pr)
+    cfg = {
+        "model": "MNISTConvNet(3,5,64) n=28440",
+        "alg": "dinno",
+        "nodes": N,
+        "graph": args.graph,
+        "global_batch": B * N,
+        "primal_iterations": 2,
+    }
+    return pr, step_fn, cfg
+
+
+def _build_density(args, device, use_hip):
+    """BASELINE config 4: DSGT online implicit density (FourierNet,
+    synthetic lidar sliding windows, dynamic disk graph)."""
+    import numpy as np
+
+    from nn_distributed_training_amd.data.floorplan import (
+        synthetic_floorplan,
+        synthetic_waypoints,
+    )
+    from nn_distributed_training_amd.data.lidar import (
+        Lidar2D,
+        OnlineTrajectoryLidarDataset,
+        RandomPoseLidarDataset,
+    )
+    from nn_distributed_training_amd.optimizers.dsgt import DSGT
+    from nn_distributed_training_amd.problems.dist_online_dense_problem \
+        import DistOnlineDensityProblem
+
+    np.random.seed(0)
+    N = args.nodes
+    B = args.batch or 20000  # paper DSGT batch (dist_online_dense_PAPER)
+    img = synthetic_floorplan(nx=256, ny=256, num_walls=10,
+                              border_width=24, seed=0)
+    lidar = Lidar2D(img, 20, 0.2, 25, 1.0, 50, 3)
+    waypoints = synthetic_waypoints(img, N, seed=0)
+    subsets = [
+        OnlineTrajectoryLidarDataset(lidar, wp, 30, 100)
+        for wp in waypoints
+    ]
+    val = RandomPoseLidarDataset(lidar, 100)
+
+    from nn_distributed_training_amd.models import FourierNet
+
+    base_model = FourierNet([2, 256, 64, 64, 64, 1], scale=0.05)
+    opt_conf = {
+        "alg_name": "dsgt",
+        "alpha": 0.001,
+        "outer_iterations": args.warmup + args.steps,
+        "init_grads": True,
+        "profile": False,
+    }
+    prob_conf = {
+        "problem_name": "bench_dsgt_density",
+        "train_batch_size": B,
+        "val_batch_size": 10000,
+        "comm_radius": 120.0,
+        "dynamic_graph": True,
+        "save_models": False,
+        "data_seed": 0,
+        "verbose_evals": False,
+        "metrics": ["consensus_error", "validation_loss"],
+        "metrics_config": {"evaluate_frequency": 10**9,
+                           "tloss_decay": 0.2,
+                           "mesh_only_at_end": True},
+        "optimizer_config": opt_conf,
+    }
+    pr = DistOnlineDensityProblem(
+        base_model, torch.nn.BCELoss(), subsets, val, device, prob_conf
+    )
+    if use_hip:
+        from nn_distributed_training_amd.ops.stacked import (
+            DSGTStackedDriver,
+            StackedEngine,
+        )
+
+        pr.stacked = StackedEngine(pr)
+        driver = DSGTStackedDriver(DSGT(pr, device, opt_conf), pr)
+        driver.prepare()
+        step_fn = driver.step_round
+    else:
+        dopt = DSGT(pr, device, opt_conf)
+        # golden per-round closure (init_grads bootstrap first)
+        state = {"ready": False}
+
+        def step_fn(k, _dopt=dopt):
+            if not state["ready"]:
+                if _dopt.conf["init_grads"]:
+                    for i in pr.local_nodes:
+                        g = _dopt._local_grad_vector(i)
+                        _dopt.y[i] = g.clone()
+                        _dopt.g[i] = g.clone()
+                state["ready"] = True
+            _golden_dsgt_round(_dopt, pr)
+
+    cfg = {
+        "model": "FourierNet[2,256,64,64,64,1] n=25601",
+        "alg": "dsgt",
+        "nodes": N,
+        "graph": "dynamic-disk(r=120)",
+        "global_batch": B * N,
+        "window": 100,
+    }
+    return pr, step_fn, cfg
+
+
+def _golden_dsgt_round(opt, pr):
+    import torch as _t
+
+    from nn_distributed_training_amd.optimizers.neighbors import (
+        gather_neighbor_stacks,
+    )
+    from nn_distributed_training_amd.utils import graph_generation
+
+    pr.update_graph()
+    W = graph_generation.get_metropolis(pr.graph).to(pr.device)
+    ths = pr.local_params_stack()
+    ys = _t.stack([opt.y[i] for i in pr.local_nodes])
+    bundle = _t.cat([ths, ys], dim=1)
+    neigh = gather_neighbor_stacks(pr, bundle)
+    n = pr.n
+    y_new = {}
+    for li, i in enumerate(pr.local_nodes):
+        p_mix = W[i, i] * (ths[li] - opt.alpha * opt.y[i])
+        y_mix = W[i, i] * opt.y[i]
+        for row, j in zip(neigh[i], pr.graph.neighbors(i)):
+            p_mix = p_mix + W[i, j] * (row[:n] - opt.alpha * row[n:])
+            y_mix = y_mix + W[i, j] * row[n:]
+        _t.nn.utils.vector_to_parameters(
+            p_mix, pr.models[i].parameters()
+        )
+        y_new[i] = y_mix
+    for i in pr.local_nodes:
+        g_next = opt._local_grad_vector(i)
+        opt.y[i] = y_new[i] + g_next - opt.g[i]
+        opt.g[i] = g_next
 
 
 def _golden_round_fn(opt, pr):

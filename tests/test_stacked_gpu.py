@@ -112,6 +112,60 @@ def test_stacked_matches_golden(alg):
 
 
 @requires_gpu
+@pytest.mark.parametrize("alg", ["dinno", "dsgt"])
+def test_stacked_density_matches_golden(alg):
+    """FourierNet + BCE density problem: stacked engine vs golden torch
+    over 3 rounds (equal-size random-pose node datasets, full batch)."""
+    import numpy as np
+
+    from nn_distributed_training_amd.data.floorplan import (
+        synthetic_floorplan,
+    )
+    from nn_distributed_training_amd.data.lidar import (
+        Lidar2D,
+        RandomPoseLidarDataset,
+    )
+    from nn_distributed_training_amd.problems.dist_dense_problem import (
+        DistDensityProblem,
+    )
+
+    torch.set_default_dtype(torch.float64)
+    conf = _conf(copy.deepcopy(ALG_CONFS[alg]))
+
+    def build():
+        torch.manual_seed(7)
+        np.random.seed(7)
+        img = synthetic_floorplan(nx=96, ny=96, num_walls=3,
+                                  border_width=10, seed=0)
+        lidar = Lidar2D(img, 6, 0.25, 8, 1.0, 20, 3)
+        sets = [
+            RandomPoseLidarDataset(lidar, 4) for _ in range(N_NODES)
+        ]
+        val = RandomPoseLidarDataset(lidar, 4)
+        per = len(sets[0])
+        c = dict(conf, train_batch_size=per)
+        graph = nx.cycle_graph(N_NODES)
+        model = FourierNet([2, 16, 8, 1], scale=0.05)
+        return DistDensityProblem(
+            graph, model, torch.nn.BCELoss(), sets, val,
+            torch.device("cuda"), c,
+        ), c
+
+    pr_g, c = build()
+    opt_g = build_optimizer(pr_g, pr_g.device, c["optimizer_config"])
+    opt_g.train()
+    golden = pr_g.local_params_stack()
+
+    pr_s, c = build()
+    pr_s.stacked = StackedEngine(pr_s)
+    opt_s = build_optimizer(pr_s, pr_s.device, c["optimizer_config"])
+    opt_s.train()
+    torch.testing.assert_close(
+        pr_s.stacked.theta, golden, rtol=1e-8, atol=1e-8
+    )
+
+
+@requires_gpu
 def test_stacked_fourier_forward_matches_module():
     """Stacked FourierNet forward == eager module forward (fp64)."""
     torch.set_default_dtype(torch.float64)
