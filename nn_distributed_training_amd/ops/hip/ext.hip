@@ -6,6 +6,7 @@
 #include "common.h"
 #include "elementwise.hip"
 #include "gemm.hip"
+#include "gemm_mfma.hip"
 #include "conv.hip"
 #include "losses.hip"
 
@@ -197,15 +198,24 @@ void linear_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
                 long M, long I, long O, long act, double scale) {
   CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
   const long L = theta.size(0), n = theta.size(1);
-  dim3 grid((O + 15) / 16, (M + 15) / 16, L);
+  const bool use_mfma = (M >= 32 && O >= 16 && I >= 8);
   DISPATCH_FT(X, {
-    hipLaunchKernelGGL(gemm::linear_fwd_k<scalar_t>,
-        grid, dim3(16, 16), 0, cur_stream(),
-        X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
-        Y.data_ptr<scalar_t>(),
-        Z.has_value() ? Z->data_ptr<scalar_t>() : nullptr,
-        n, w_off, b_off, (int)M, (int)I, (int)O, (int)act,
-        (scalar_t)scale);
+    auto zp = Z.has_value() ? Z->data_ptr<scalar_t>() : nullptr;
+    if (use_mfma) {
+      dim3 grid((O + 63) / 64, (M + 63) / 64, L);
+      hipLaunchKernelGGL(gmfma::mfma_fwd_k<scalar_t>,
+          grid, dim3(256), 0, cur_stream(),
+          X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), zp, n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)act, (scalar_t)scale);
+    } else {
+      dim3 grid((O + 15) / 16, (M + 15) / 16, L);
+      hipLaunchKernelGGL(gemm::linear_fwd_k<scalar_t>,
+          grid, dim3(16, 16), 0, cur_stream(),
+          X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), zp, n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)act, (scalar_t)scale);
+    }
   });
   HIP_CHECK_LAST();
 }
@@ -228,28 +238,70 @@ void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
                    torch::Tensor dX, long w_off, long M, long I, long O) {
   CHECK_DEV(dZ); CHECK_DEV(dX);
   const long L = theta.size(0), n = theta.size(1);
-  dim3 grid((I + 15) / 16, (M + 15) / 16, L);
+  const bool use_mfma = (M >= 32 && I >= 16 && O >= 8);
   DISPATCH_FT(dZ, {
-    hipLaunchKernelGGL(gemm::linear_bwd_dx_k<scalar_t>,
-        grid, dim3(16, 16), 0, cur_stream(),
-        dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
-        dX.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+    if (use_mfma) {
+      dim3 grid((I + 63) / 64, (M + 63) / 64, L);
+      hipLaunchKernelGGL(gmfma::mfma_dx_k<scalar_t>,
+          grid, dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          dX.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+    } else {
+      dim3 grid((I + 15) / 16, (M + 15) / 16, L);
+      hipLaunchKernelGGL(gemm::linear_bwd_dx_k<scalar_t>,
+          grid, dim3(16, 16), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          dX.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+    }
   });
   HIP_CHECK_LAST();
 }
 
+// dW/db. PRECONDITION for the atomic paths (MFMA and small-chunked):
+// the layer's slice of gstack is zeroed (the stacked engine zeroes the
+// whole grad stack at backward start). The plain VALU path overwrites.
 void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
                    torch::Tensor gstack, long w_off, long b_off,
                    long M, long I, long O) {
   CHECK_DEV(dZ); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
-  dim3 grid((I + 15) / 16, (O + 15) / 16, L);
   DISPATCH_FT(dZ, {
-    hipLaunchKernelGGL(gemm::linear_bwd_dw_k<scalar_t>,
-        grid, dim3(16, 16), 0, cur_stream(),
-        dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
-        gstack.data_ptr<scalar_t>(), n, w_off, b_off,
-        (int)M, (int)I, (int)O);
+    if (M >= 64 && I >= 16 && O >= 16) {
+      // fill the chip: tiles * L * nchunk ≈ 2048 blocks
+      const long tiles = ((I + 63) / 64) * ((O + 63) / 64);
+      long nchunk = std::max<long>(1, 2048 / std::max<long>(1, tiles * L));
+      nchunk = std::min<long>(nchunk, (M + 63) / 64);
+      dim3 grid((I + 63) / 64, (O + 63) / 64, L * nchunk);
+      hipLaunchKernelGGL(gmfma::mfma_dw_k<scalar_t>,
+          grid, dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+          gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)nchunk);
+      long nb = std::max<long>(1, 1024 / std::max<long>(1, L));
+      nb = std::min<long>(nb, (M + 255) / 256);
+      hipLaunchKernelGGL(gmfma::bias_grad_chunked_k<scalar_t>,
+          dim3((O + 255) / 256, 1, L * nb), dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+          n, b_off, (int)M, (int)O, (int)nb);
+    } else if (M > 2048) {
+      const long total = O * I + O;
+      long nchunk = std::max<long>(
+          1, 2048 / std::max<long>(1, ((total + 255) / 256) * L));
+      nchunk = std::min<long>(nchunk, (M + 255) / 256);
+      dim3 grid((total + 255) / 256, 1, L * nchunk);
+      hipLaunchKernelGGL(gemm::dw_small_chunked_k<scalar_t>,
+          grid, dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+          gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)nchunk);
+    } else {
+      dim3 grid((I + 15) / 16, (O + 15) / 16, L);
+      hipLaunchKernelGGL(gemm::linear_bwd_dw_k<scalar_t>,
+          grid, dim3(16, 16), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+          gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+          (int)M, (int)I, (int)O);
+    }
   });
   HIP_CHECK_LAST();
 }

@@ -161,4 +161,38 @@ __global__ void linear_bwd_dw_k(
   }
 }
 
+// M-chunked dW+db for skinny layers (I or O < 16, e.g. the FourierNet
+// encode's dW[256, 2] with M = 20k): each thread owns one (o, i) weight
+// entry (or one bias entry), reduces its M chunk, atomically adds.
+// Requires the layer's grad slice zeroed by the caller.
+template <typename T>
+__global__ void dw_small_chunked_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O, int nchunk) {
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const int mc = (M + nchunk - 1) / nchunk;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+  const T* dZl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+
+  const int t = blockIdx.x * blockDim.x + threadIdx.x;
+  const int total = O * I + O;  // weights then biases
+  if (t >= total) return;
+  T acc = T(0);
+  if (t < O * I) {
+    const int o = t / I, i = t % I;
+    for (int m = mlo; m < mhi; ++m) {
+      acc += dZl[(long)m * O + o] * Xl[(long)m * I + i];
+    }
+    atomicAdd(&gstack[l * n + w_off + t], acc);
+  } else {
+    const int o = t - O * I;
+    for (int m = mlo; m < mhi; ++m) acc += dZl[(long)m * O + o];
+    atomicAdd(&gstack[l * n + b_off + o], acc);
+  }
+}
+
 }  // namespace gemm

@@ -1,0 +1,306 @@
+// MFMA-tiled stacked linear kernels (CDNA4 gfx950).
+//
+// The density workload's GEMMs (M up to 20k rows per node, I/O 64-256)
+// run on the matrix cores via v_mfma_f64_16x16x4_f64 (fp64 parity path)
+// / v_mfma_f32_16x16x4_f32 (fp32 path — exact f32, §3 of the CDNA guide;
+// there is no TF32 on gfx950). Geometry: 256-thread blocks = 4 waves in
+// a 2x2 grid, each wave owns a 32x32 output sub-tile as 2x2 16x16 MFMA
+// fragments; K advances 16 per LDS stage (4 MFMA k-steps).
+//
+// Fragment maps for the 16x16x4 shapes (CDNA ISA §10; verified against
+// torch references in tests/test_ops_gpu.py):
+//   A operand: lane l holds A[i = l&15][k = l>>4]   (one element)
+//   B operand: lane l holds B[k = l>>4][j = l&15]
+//   C/D:       4 elements, elem r at row (l>>4)*4 + r, col l&15
+//
+// Three contraction variants cover fwd/bwd:
+//   NT (fwd):  Y[M,O] = act(X[M,I] @ W[O,I]^T + b)   contraction I
+//   NN (dx):   dX[M,I] = dZ[M,O] @ W[O,I]            contraction O
+//   TN (dw):   dW[O,I] = dZ[M,O]^T @ X[M,I]          contraction M,
+//              M-chunked across blocks with atomic accumulation
+//              (+ fused db), so 20k-row reductions spread over the
+//              256-CU chip instead of serializing in one block's K loop.
+
+#include "common.h"
+
+namespace gmfma {
+
+typedef double f64x4 __attribute__((ext_vector_type(4)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+template <typename T>
+struct mfma_t;
+
+template <>
+struct mfma_t<double> {
+  using acc_t = f64x4;
+  static DEV_INLINE acc_t mma(double a, double b, acc_t c) {
+    return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
+  }
+};
+
+template <>
+struct mfma_t<float> {
+  using acc_t = f32x4;
+  static DEV_INLINE acc_t mma(float a, float b, acc_t c) {
+    return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+};
+
+constexpr int BM = 64;   // block tile rows
+constexpr int BN = 64;   // block tile cols
+constexpr int BK = 16;   // K per LDS stage
+
+// ---------------------------------------------------------------------
+// NT: Y[M,O] = act(X[M,I] @ W_l[O,I]^T + b_l)  [+ optional Z store]
+template <typename T>
+__global__ __launch_bounds__(256) void mfma_fwd_k(
+    const T* __restrict__ X, const T* __restrict__ theta,
+    T* __restrict__ Y, T* __restrict__ Z,
+    long n, long w_off, long b_off, int M, int I, int O,
+    int act, T scale) {
+  using MF = mfma_t<T>;
+  using acc_t = typename MF::acc_t;
+  __shared__ T As[BK][BM + 1];   // A^T image: As[k][m]
+  __shared__ T Bs[BK][BN + 1];   // Bs[k][o] = W[o][k]
+
+  const long l = blockIdx.z;
+  const T* Xl = X + l * (long)M * I;
+  const T* W = theta + l * n + w_off;
+  const T* bias = theta + l * n + b_off;
+
+  const int m0 = blockIdx.y * BM;
+  const int o0 = blockIdx.x * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;          // 4 waves: (wm, wn) = (wid>>1, wid&1)
+  const int wm = (wid >> 1) * 32;
+  const int wn = (wid & 1) * 32;
+
+  acc_t acc[2][2] = {};
+
+  for (int k0 = 0; k0 < I; k0 += BK) {
+    // stage A^T: 64 rows x 16 k  (1024 elems, 4 per thread)
+    for (int t = tid; t < BM * BK; t += 256) {
+      const int m = t / BK, k = t % BK;
+      As[k][m] = (m0 + m < M && k0 + k < I)
+                     ? Xl[(long)(m0 + m) * I + (k0 + k)]
+                     : T(0);
+    }
+    // stage B: Bs[k][o] = W[o0+o][k0+k]
+    for (int t = tid; t < BN * BK; t += 256) {
+      const int o = t / BK, k = t % BK;
+      Bs[k][o] = (o0 + o < O && k0 + k < I)
+                     ? W[(long)(o0 + o) * I + (k0 + k)]
+                     : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int ka = kk + (lane >> 4);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        const T a = As[ka][wm + fm * 16 + (lane & 15)];
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
+          acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // epilogue: bias + activation, coalesced per-fragment store
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        const int o = o0 + wn + fn * 16 + (lane & 15);
+        if (m < M && o < O) {
+          const T z = acc[fm][fn][r] + bias[o];
+          const long off = l * (long)M * O + (long)m * O + o;
+          if (Z != nullptr) Z[off] = z;
+          Y[off] = act_fwd(act, z, scale);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// NN: dX[M,I] = dZ[M,O] @ W_l[O,I]   (contraction over O)
+template <typename T>
+__global__ __launch_bounds__(256) void mfma_dx_k(
+    const T* __restrict__ dZ, const T* __restrict__ theta,
+    T* __restrict__ dX, long n, long w_off, int M, int I, int O) {
+  using MF = mfma_t<T>;
+  using acc_t = typename MF::acc_t;
+  __shared__ T As[BK][BM + 1];   // As[k=o][m] = dZ[m][o]
+  __shared__ T Bs[BK][BN + 1];   // Bs[k=o][i] = W[o][i]
+
+  const long l = blockIdx.z;
+  const T* Gl = dZ + l * (long)M * O;
+  const T* W = theta + l * n + w_off;
+
+  const int m0 = blockIdx.y * BM;
+  const int i0 = blockIdx.x * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 32;
+  const int wn = (wid & 1) * 32;
+
+  acc_t acc[2][2] = {};
+
+  for (int k0 = 0; k0 < O; k0 += BK) {
+    for (int t = tid; t < BM * BK; t += 256) {
+      const int m = t / BK, k = t % BK;
+      As[k][m] = (m0 + m < M && k0 + k < O)
+                     ? Gl[(long)(m0 + m) * O + (k0 + k)]
+                     : T(0);
+    }
+    for (int t = tid; t < BN * BK; t += 256) {
+      const int i = t % BN, k = t / BN;
+      Bs[k][i] = (k0 + k < O && i0 + i < I)
+                     ? W[(long)(k0 + k) * I + (i0 + i)]
+                     : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int ka = kk + (lane >> 4);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        const T a = As[ka][wm + fm * 16 + (lane & 15)];
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
+          acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        const int i = i0 + wn + fn * 16 + (lane & 15);
+        if (m < M && i < I) {
+          dX[l * (long)M * I + (long)m * I + i] = acc[fm][fn][r];
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// TN: dW[O,I] partial = dZ_chunk[Mc,O]^T @ X_chunk[Mc,I], atomically
+// accumulated into the grad stack; db fused (i-tile 0 blocks).
+// Grid: (I/BN, O/BM-rows?, L * nchunk). We tile dW as [O rows][I cols]
+// with the same 64x64 block: A[k=m][o] = dZ[m][o], B[k=m][i] = X[m][i].
+template <typename T>
+__global__ __launch_bounds__(256) void mfma_dw_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O, int nchunk) {
+  using MF = mfma_t<T>;
+  using acc_t = typename MF::acc_t;
+  __shared__ T As[BK][BM + 1];   // As[k=m][o]
+  __shared__ T Bs[BK][BN + 1];   // Bs[k=m][i]
+
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const T* Gl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+
+  const int mc = (M + nchunk - 1) / nchunk;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+
+  const int o0 = blockIdx.y * BM;
+  const int i0 = blockIdx.x * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 32;   // o-direction
+  const int wn = (wid & 1) * 32;    // i-direction
+
+  acc_t acc[2][2] = {};
+
+  for (int k0 = mlo; k0 < mhi; k0 += BK) {
+    for (int t = tid; t < BM * BK; t += 256) {
+      const int o = t / BK, k = t % BK;
+      As[k][o] = (o0 + o < O && k0 + k < mhi)
+                     ? Gl[(long)(k0 + k) * O + (o0 + o)]
+                     : T(0);
+    }
+    for (int t = tid; t < BN * BK; t += 256) {
+      const int i = t % BN, k = t / BN;
+      Bs[k][i] = (k0 + k < mhi && i0 + i < I)
+                     ? Xl[(long)(k0 + k) * I + (i0 + i)]
+                     : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int ka = kk + (lane >> 4);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        const T a = As[ka][wm + fm * 16 + (lane & 15)];
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
+          acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int o = o0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        const int i = i0 + wn + fn * 16 + (lane & 15);
+        if (o < O && i < I) {
+          atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
+                    acc[fm][fn][r]);
+        }
+      }
+    }
+  }
+}
+
+// db[O] = sum_m dZ[m, o], M-chunked over blocks, atomic accumulate.
+// (kept separate from mfma_dw_k: the bias reduction is over the A image
+// only and one light kernel beats complicating the MFMA epilogue)
+template <typename T>
+__global__ void bias_grad_chunked_k(
+    const T* __restrict__ dZ, T* __restrict__ gstack,
+    long n, long b_off, int M, int O, int nchunk) {
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const int mc = (M + nchunk - 1) / nchunk;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+  const int o = blockIdx.x * blockDim.x + threadIdx.x;
+  if (o >= O) return;
+  const T* Gl = dZ + l * (long)M * O;
+  T acc = T(0);
+  for (int m = mlo; m < mhi; ++m) acc += Gl[(long)m * O + o];
+  atomicAdd(&gstack[l * n + b_off + o], acc);
+}
+
+}  // namespace gmfma

@@ -331,6 +331,8 @@ class StackedEngine:
         layers = self.spec.layers
         nl = len(layers)
         M = self.B
+        # conv bwd and the chunked/MFMA dW paths accumulate atomically
+        self.grad.zero_()
         loss_buf = None
         if want_loss:
             bufs["loss"].zero_()
@@ -366,12 +368,6 @@ class StackedEngine:
             below = bufs["acts"][li - 1] if li > 0 else xb
             dz = bufs["dzs"][li]
             if layer.kind == "conv_pool":
-                # atomic accumulation: zero this layer's grad slice
-                w0 = layer.w_off
-                cnt = (
-                    layer.out_dim * layer.kernel_size**2 + layer.out_dim
-                )
-                self.grad[:, w0 : w0 + cnt].zero_()
                 ext.conv_pool_bwd(
                     dz, bufs["idxs"][li], below, self.grad,
                     layer.w_off, layer.b_off, M, layer.out_dim,
