@@ -7,11 +7,13 @@
 // a 2x2 grid, each wave owns a 32x32 output sub-tile as 2x2 16x16 MFMA
 // fragments; K advances 16 per LDS stage (4 MFMA k-steps).
 //
-// Fragment maps for the 16x16x4 shapes (CDNA ISA §10; verified against
-// torch references in tests/test_ops_gpu.py):
+// Fragment maps for the 16x16x4 shapes (measured with
+// tools/mfma_probe.hip on gfx950; verified against torch references in
+// tests/test_ops_gpu.py):
 //   A operand: lane l holds A[i = l&15][k = l>>4]   (one element)
 //   B operand: lane l holds B[k = l>>4][j = l&15]
-//   C/D:       4 elements, elem r at row (l>>4)*4 + r, col l&15
+//   C/D:       4 elements, col l&15; row differs BY DTYPE:
+//              f32: (l>>4)*4 + r      f64: 4*r + (l>>4)
 //
 // Three contraction variants cover fwd/bwd:
 //   NT (fwd):  Y[M,O] = act(X[M,I] @ W[O,I]^T + b)   contraction I
@@ -37,6 +39,11 @@ struct mfma_t<double> {
   static DEV_INLINE acc_t mma(double a, double b, acc_t c) {
     return __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, c, 0, 0, 0);
   }
+  // measured on gfx950 (tools/mfma_probe.hip): f64 acc rows interleave
+  // as 4*reg + (lane>>4) — DIFFERENT from the f32 form
+  static DEV_INLINE int acc_row(int lane, int r) {
+    return 4 * r + (lane >> 4);
+  }
 };
 
 template <>
@@ -44,6 +51,9 @@ struct mfma_t<float> {
   using acc_t = f32x4;
   static DEV_INLINE acc_t mma(float a, float b, acc_t c) {
     return __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, c, 0, 0, 0);
+  }
+  static DEV_INLINE int acc_row(int lane, int r) {
+    return (lane >> 4) * 4 + r;
   }
 };
 
@@ -118,7 +128,7 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
     for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int m = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
         const int o = o0 + wn + fn * 16 + (lane & 15);
         if (m < M && o < O) {
           const T z = acc[fm][fn][r] + bias[o];
@@ -192,7 +202,7 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
     for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int m = m0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
         const int i = i0 + wn + fn * 16 + (lane & 15);
         if (m < M && i < I) {
           dX[l * (long)M * I + (long)m * I + i] = acc[fm][fn][r];
@@ -272,7 +282,7 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
     for (int fn = 0; fn < 2; ++fn) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int o = o0 + wm + fm * 16 + (lane >> 4) * 4 + r;
+        const int o = o0 + wm + fm * 16 + MF::acc_row(lane, r);
         const int i = i0 + wn + fn * 16 + (lane & 15);
         if (o < O && i < I) {
           atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
