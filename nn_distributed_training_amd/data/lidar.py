@@ -279,23 +279,31 @@ class OnlineTrajectoryLidarDataset(torch.utils.data.Dataset):
         self._advance_window()
 
     def _advance_window(self):
-        """Move the window forward one step and refill the index pool."""
+        """Move the window forward one step; index pool refills lazily
+        (the stacked engine samples the window's [lb, ub) bounds with a
+        device-side randperm instead of a host list shuffle)."""
         w = self.num_scans_in_window
         start = self.curr_scan_idx
         end = min(start + w, self.num_scans)
         self.curr_scan_idx = end % self.num_scans if end >= self.num_scans \
             else end
-        lb, ub = self.scan_size * start, self.scan_size * end
+        self.window_bounds = (self.scan_size * start, self.scan_size * end)
         self.curr_pos = self.scan_locs[min(end, self.num_scans - 1)]
-        self.curr_idx_list = list(range(lb, ub))
-        random.shuffle(self.curr_idx_list)
+        self.curr_idx_list = None  # lazy; see curr_index_pool()
+
+    def curr_index_pool(self):
+        if self.curr_idx_list is None:
+            lb, ub = self.window_bounds
+            self.curr_idx_list = list(range(lb, ub))
+            random.shuffle(self.curr_idx_list)
+        return self.curr_idx_list
 
     # the reference pops shuffled indices until the window empties, then
     # slides the window (lidar.py:383-424); same contract here
     def __getitem__(self, index):
-        if not self.curr_idx_list:
+        if not self.curr_index_pool():
             self._advance_window()
-        return self.tds[self.curr_idx_list.pop()]
+        return self.tds[self.curr_index_pool().pop()]
 
     def __len__(self):
         return len(self.tds)

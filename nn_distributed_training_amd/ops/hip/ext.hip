@@ -198,7 +198,9 @@ void linear_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
                 long M, long I, long O, long act, double scale) {
   CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
   const long L = theta.size(0), n = theta.size(1);
-  const bool use_mfma = (M >= 32 && O >= 16 && I >= 8);
+  // MFMA pays only when M fills the 64-row tiles across the chip;
+  // small-M layers (MNIST fc, per-node B=64) stay on the VALU kernels
+  const bool use_mfma = (M >= 128 && O >= 16 && I >= 8);
   DISPATCH_FT(X, {
     auto zp = Z.has_value() ? Z->data_ptr<scalar_t>() : nullptr;
     if (use_mfma) {
@@ -238,7 +240,7 @@ void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
                    torch::Tensor dX, long w_off, long M, long I, long O) {
   CHECK_DEV(dZ); CHECK_DEV(dX);
   const long L = theta.size(0), n = theta.size(1);
-  const bool use_mfma = (M >= 32 && I >= 16 && O >= 8);
+  const bool use_mfma = (M >= 128 && I >= 16 && O >= 8);
   DISPATCH_FT(dZ, {
     if (use_mfma) {
       dim3 grid((I + 63) / 64, (M + 63) / 64, L);
@@ -266,7 +268,7 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
   CHECK_DEV(dZ); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
   DISPATCH_FT(dZ, {
-    if (M >= 64 && I >= 16 && O >= 16) {
+    if (M >= 256 && I >= 16 && O >= 16) {
       // fill the chip: tiles * L * nchunk ≈ 2048 blocks
       const long tiles = ((I + 63) / 64) * ((O + 63) / 64);
       long nchunk = std::max<long>(1, 2048 / std::max<long>(1, tiles * L));

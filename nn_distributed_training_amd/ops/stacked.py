@@ -140,9 +140,14 @@ class _OnlineWindowSampler:
             ds._advance_window()
             if ds.curr_scan_idx <= ds.num_scans_in_window:
                 self.epoch_cb(li)  # wrapped around the trajectory
-        idx = torch.as_tensor(ds.curr_idx_list, dtype=torch.long)
-        perm = torch.randperm(idx.numel(), generator=self.gens[li])
-        self.pools[li] = idx[perm].to(self.device)
+        # device-side shuffle of the window's [lb, ub) index range — the
+        # golden path's host list shuffle was ~30 ms per 50k-sample
+        # window and dominated the round (profiles/ density pass 2)
+        lb, ub = ds.window_bounds
+        perm = torch.randperm(ub - lb, generator=self.gens[li]).to(
+            self.device
+        )
+        self.pools[li] = perm + lb
         self.pos[li] = 0
 
     def next_view(self) -> torch.Tensor:
