@@ -79,7 +79,7 @@ def test_online_dataset_window_advances(lidar):
         lidar, wps, spline_res=2, num_scans_in_window=3
     )
     pos0 = ds.curr_pos.copy()
-    window = len(ds.curr_idx_list)
+    window = len(ds.curr_index_pool())
     assert window == 3 * 8 * 10
     # drain the window; the robot must advance
     for _ in range(window + 1):
