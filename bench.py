@@ -145,6 +145,14 @@ def main():
             },
         }
         print(json.dumps(out), flush=True)
+        try:
+            from nn_distributed_training_amd.ops.stacked import (
+                timing_report,
+            )
+
+            timing_report()
+        except ImportError:
+            pass
 
     if world > 1:
         dist.destroy_process_group()

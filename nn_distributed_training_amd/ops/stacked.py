@@ -26,7 +26,43 @@ deviation, stream-equivalent otherwise).
 
 from __future__ import annotations
 
+import os
+import time
+from collections import defaultdict
+
 import torch
+
+# NDTA_TIMING=1 prints a host-side wall-time breakdown per section when
+# a driver finishes (bench/diagnostic aid; no cost when off)
+_TIMING = os.environ.get("NDTA_TIMING", "0") == "1"
+_tacc = defaultdict(float)
+_tcnt = defaultdict(int)
+
+
+class _timer:
+    __slots__ = ("key", "t0")
+
+    def __init__(self, key):
+        self.key = key
+
+    def __enter__(self):
+        if _TIMING:
+            self.t0 = time.perf_counter()
+
+    def __exit__(self, *a):
+        if _TIMING:
+            _tacc[self.key] += time.perf_counter() - self.t0
+            _tcnt[self.key] += 1
+
+
+def timing_report():
+    if _TIMING:
+        tot = sum(_tacc.values())
+        print(f"[ndta timing] total {tot*1e3:.1f} ms")
+        for k in sorted(_tacc, key=lambda k: -_tacc[k]):
+            print(
+                f"  {k:20s} {_tacc[k]*1e3:9.1f} ms  {_tcnt[k]:6d} calls"
+            )
 
 from ..models.spec import ModelSpec, model_spec
 from . import get_ext
@@ -674,23 +710,31 @@ class DSGTStackedDriver:
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
         ext = eng.ext
-        pr.update_graph()
-        rbuf, dests_p, dests_y, offs, idx, w = self._round_plan()
-        if pr.comm.world > 1:
-            pr.comm.exchange_rows(
-                pr.layout, list(pr.graph.edges()),
-                [eng.theta, self.y], [dests_p, dests_y],
+        with _timer("update_graph"):
+            pr.update_graph()
+        with _timer("round_plan"):
+            rbuf, dests_p, dests_y, offs, idx, w = self._round_plan()
+        with _timer("exchange"):
+            if pr.comm.world > 1:
+                pr.comm.exchange_rows(
+                    pr.layout, list(pr.graph.edges()),
+                    [eng.theta, self.y], [dests_p, dests_y],
+                )
+        with _timer("mix"):
+            ext.dsgt_mix(
+                eng.theta, self.y, rbuf, offs, idx, w, self.theta_next,
+                self.y_mix, self.alpha,
             )
-        ext.dsgt_mix(
-            eng.theta, self.y, rbuf, offs, idx, w, self.theta_next,
-            self.y_mix, self.alpha,
-        )
-        eng.theta, self.theta_next = self.theta_next, eng.theta
+            eng.theta, self.theta_next = self.theta_next, eng.theta
 
-        xb, yb = eng.next_batch()
-        eng.forward(xb)
-        eng.backward(xb, yb)
-        ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
+        with _timer("next_batch"):
+            xb, yb = eng.next_batch()
+        with _timer("forward"):
+            eng.forward(xb)
+        with _timer("backward"):
+            eng.backward(xb, yb)
+        with _timer("y_update"):
+            ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
 
     def run(self, profiler=None):
         pr = self.pr
