@@ -158,8 +158,9 @@ class _OnlineWindowSampler:
         self.B = batch
         self.device = device
         self.epoch_cb = epoch_cb
+        # device-resident generators: window shuffles never touch host
         self.gens = [
-            torch.Generator().manual_seed(seed * 100003 + i)
+            torch.Generator(device=device).manual_seed(seed * 100003 + i)
             for i in range(len(datasets))
         ]
         self.pools = [None] * len(datasets)
@@ -180,8 +181,8 @@ class _OnlineWindowSampler:
         # golden path's host list shuffle was ~30 ms per 50k-sample
         # window and dominated the round (profiles/ density pass 2)
         lb, ub = ds.window_bounds
-        perm = torch.randperm(ub - lb, generator=self.gens[li]).to(
-            self.device
+        perm = torch.randperm(
+            ub - lb, generator=self.gens[li], device=self.device
         )
         self.pools[li] = perm + lb
         self.pos[li] = 0
