@@ -513,6 +513,13 @@ def _graph_is_static(pr):
     return type(pr).update_graph is ProblemBase.update_graph
 
 
+def _edge_key(pr):
+    """Hashable identity of the current communication graph; dynamic
+    graphs change edges only when robots cross the comm radius, so
+    plans/CSRs are cached on this key instead of rebuilt every round."""
+    return tuple(sorted(map(tuple, map(sorted, pr.graph.edges()))))
+
+
 class DiNNOStackedDriver:
     """DiNNO outer loop with the fused kernels (same math as
     optimizers/dinno.py)."""
@@ -535,21 +542,21 @@ class DiNNOStackedDriver:
         self.v = torch.zeros_like(eng.theta)
         self.rho = conf["rho_init"]
         self.step_t = 0
-        self._static = _graph_is_static(self.pr)
+        self._plan_key = None
         self._plan = None
 
     def _round_plan(self):
-        if self._static and self._plan is not None:
+        key = _edge_key(self.pr)
+        if key == self._plan_key:
             return self._plan
         eng = self.eng
         remote_nodes, rbuf, dests = eng.remote_plan()
         row_of = eng.row_map(remote_nodes)
         offs, idx, _ = eng.build_csr(row_of, include_self=False)
         deg = eng.degrees()
-        plan = (rbuf, dests, offs, idx, deg)
-        if self._static:
-            self._plan = plan
-        return plan
+        self._plan = (rbuf, dests, offs, idx, deg)
+        self._plan_key = key
+        return self._plan
 
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
@@ -611,23 +618,23 @@ class DSGDStackedDriver:
     def prepare(self):
         self.alph = self.opt.alph0
         self.theta_next = torch.empty_like(self.eng.theta)
-        self._static = _graph_is_static(self.pr)
+        self._plan_key = None
         self._plan = None
 
     def _round_plan(self):
         from ..utils import graph_generation
 
-        if self._static and self._plan is not None:
+        key = _edge_key(self.pr)
+        if key == self._plan_key:
             return self._plan
         eng = self.eng
         W = graph_generation.get_metropolis(self.pr.graph)
         remote_nodes, rbuf, dests = eng.remote_plan()
         row_of = eng.row_map(remote_nodes)
         offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
-        plan = (rbuf, dests, offs, idx, w)
-        if self._static:
-            self._plan = plan
-        return plan
+        self._plan = (rbuf, dests, offs, idx, w)
+        self._plan_key = key
+        return self._plan
 
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
@@ -678,7 +685,7 @@ class DSGTStackedDriver:
         self.g = torch.zeros_like(eng.theta)
         self.y_mix = torch.zeros_like(eng.theta)
         self.theta_next = torch.empty_like(eng.theta)
-        self._static = _graph_is_static(self.pr)
+        self._plan_key = None
         self._plan = None
         if self.opt.conf["init_grads"]:
             xb, yb = eng.next_batch()
@@ -690,7 +697,8 @@ class DSGTStackedDriver:
     def _round_plan(self):
         from ..utils import graph_generation
 
-        if self._static and self._plan is not None:
+        key = _edge_key(self.pr)
+        if key == self._plan_key:
             return self._plan
         eng = self.eng
         W = graph_generation.get_metropolis(self.pr.graph)
@@ -703,10 +711,9 @@ class DSGTStackedDriver:
             if rbuf is not None else {}
         row_of = eng.row_map(remote_nodes)
         offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
-        plan = (rbuf, dests_p, dests_y, offs, idx, w)
-        if self._static:
-            self._plan = plan
-        return plan
+        self._plan = (rbuf, dests_p, dests_y, offs, idx, w)
+        self._plan_key = key
+        return self._plan
 
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
