@@ -1,0 +1,165 @@
+"""RL tree tests: env physics/API, rollout bookkeeping, PPO loss math,
+and short end-to-end runs of all three decentralized PPO optimizers."""
+
+import networkx as nx
+import numpy as np
+import pytest
+import torch
+
+from nn_distributed_training_amd.rl.dist_ppo import DistPPOProblem
+from nn_distributed_training_amd.rl.envs import SimpleTagEnv
+from nn_distributed_training_amd.rl.ppo import PPO
+from nn_distributed_training_amd.rl.ppo_optimizers import (
+    build_ppo_optimizer,
+)
+from nn_distributed_training_amd.rl.train_multi import default_conf
+
+
+@pytest.fixture(autouse=True)
+def _fp32():
+    # RL runs in fp32 (reference RL tree never sets DoubleTensor)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(torch.float32)
+    yield
+    torch.set_default_dtype(prev)
+
+
+def test_env_api():
+    env = SimpleTagEnv(num_predators=3, num_obstacles=2, seed=0)
+    obs = env.reset()
+    assert obs.shape == (3, env.obs_dim)
+    assert env.obs_dim == 2 + 2 + 4 + 4 + 2 + 2
+    a = np.zeros((3, 5))
+    obs2, rews, done, info = env.step(a)
+    assert obs2.shape == obs.shape
+    assert rews.shape == (3,)
+    assert not done
+    # episode terminates at max_steps
+    for _ in range(env.max_steps):
+        _, _, done, _ = env.step(a)
+    assert done
+
+
+def test_env_forces_move_agents():
+    env = SimpleTagEnv(num_predators=2, seed=1)
+    env.reset()
+    p0 = env.pred_pos.copy()
+    a = np.zeros((2, 5))
+    a[:, 1] = 1.0  # push +x
+    for _ in range(5):
+        env.step(a)
+    assert (env.pred_pos[:, 0] > p0[:, 0]).all()
+
+
+def test_prey_evades():
+    env = SimpleTagEnv(num_predators=1, num_obstacles=0, seed=2)
+    env.reset()
+    env.pred_pos[0] = np.array([0.0, 0.0])
+    env.prey_pos = np.array([0.2, 0.0])
+    env.prey_vel = np.zeros(2)
+    a = np.zeros((1, 5))
+    env.step(a)
+    # prey accelerates away from the predator (+x)
+    assert env.prey_vel[0] > 0
+
+
+def _tiny_problem(seed=0):
+    torch.manual_seed(seed)
+    env = SimpleTagEnv(num_predators=3, num_obstacles=1, seed=seed,
+                       max_steps=20)
+    conf = {
+        "timesteps_per_batch": 60,
+        "max_timesteps_per_episode": 20,
+        "hidden": (16, 16),
+        "verbose": False,
+    }
+    graph = nx.wheel_graph(3)
+    return DistPPOProblem(graph, env, torch.device("cpu"), conf)
+
+
+def test_rollout_shapes_and_rtgs():
+    pr = _tiny_problem()
+    pr.rollout()
+    assert pr.total_timesteps >= 60
+    for i in range(3):
+        b = pr.buf[i]
+        T = b["obs"].shape[0]
+        assert b["acts"].shape == (T, 5)
+        assert b["logp"].shape == (T,)
+        assert b["rtgs"].shape == (T,)
+        # normalized advantages
+        assert abs(b["adv"].mean().item()) < 1e-5
+    # loss is differentiable through both actor and critic
+    loss = pr.local_batch_loss(0)
+    loss.backward()
+    for p in pr.node_parameters(0):
+        assert p.grad is not None
+
+
+def test_vector_roundtrip():
+    pr = _tiny_problem()
+    v = pr.node_vector(1)
+    assert v.numel() == pr.n == pr.n_actor + pr.n_critic
+    v2 = v * 2.0
+    pr.set_node_vector(1, v2)
+    torch.testing.assert_close(pr.node_vector(1), v2)
+
+
+@pytest.mark.parametrize("alg", ["dinno", "dsgd", "dsgt"])
+def test_ppo_optimizers_run(alg, tmp_path):
+    pr = _tiny_problem(seed=1)
+    conf = default_conf(alg)
+    conf.update(
+        max_rl_timesteps=130, save_freq=1, verbose=False,
+        output_dir=str(tmp_path), primal_iterations=2,
+        timesteps_per_batch=60,
+    )
+    opt = build_ppo_optimizer(alg, pr, conf)
+    v0 = pr.node_vector(0).clone()
+    opt.train()
+    assert not torch.allclose(pr.node_vector(0), v0)  # learned something
+    assert torch.isfinite(pr.node_vector(0)).all()
+    # checkpoint layout parity
+    files = {p.name for p in tmp_path.iterdir()}
+    tag = {"dinno": "cadmm", "dsgd": "dsgd", "dsgt": "dsgt"}[alg]
+    assert any(f.startswith(f"ppo_actors_tag_{tag}_") for f in files)
+    assert any(f.startswith(f"ppo_critics_tag_{tag}_") for f in files)
+    assert f"avg_ep_rews_tag_{tag}_0.npy" in files
+    assert f"agreements_tag_{tag}_0.npz" in files
+    ag = np.load(tmp_path / f"agreements_tag_{tag}_0.npz")
+    assert ag["actor"].shape[1:] == (3, 3)
+
+
+def test_single_agent_ppo_learns():
+    torch.manual_seed(0)
+    env = SimpleTagEnv(num_predators=1, num_obstacles=0, seed=0,
+                       max_steps=30)
+    agent = PPO(env, timesteps_per_batch=90,
+                max_timesteps_per_episode=30, hidden=(16, 16),
+                verbose=False)
+    rews = agent.learn(total_timesteps=400)
+    assert len(rews) >= 2
+    assert all(np.isfinite(rews))
+
+
+def test_eval_policy_roundtrip(tmp_path):
+    from nn_distributed_training_amd.rl.eval_policy import (
+        eval_episodes,
+        load_actors,
+    )
+
+    pr = _tiny_problem()
+    opt = build_ppo_optimizer("dsgd", pr, {
+        **default_conf("dsgd"), "max_rl_timesteps": 60,
+        "save_freq": 1, "verbose": False,
+        "output_dir": str(tmp_path), "timesteps_per_batch": 60,
+    })
+    opt.train()
+    path = next(
+        p for p in tmp_path.iterdir()
+        if p.name.startswith("ppo_actors_tag_dsgd_0")
+    )
+    env = SimpleTagEnv(num_predators=3, num_obstacles=1, max_steps=20)
+    actors = load_actors(str(path), env, hidden=(16, 16))
+    rews, _ = eval_episodes(actors, env, episodes=2, max_steps=20)
+    assert len(rews) == 2 and all(np.isfinite(rews))
