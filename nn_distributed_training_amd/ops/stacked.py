@@ -235,6 +235,22 @@ class StackedEngine:
         self.B = problem.conf["train_batch_size"]
         self._bufs = None
         self._loss_kind = type(problem.base_loss).__name__  # NLLLoss etc.
+        # device-resident train-loss EMA (the reference's tloss tracker,
+        # problems/dist_online_dense_problem.py:129-137, without a
+        # host sync per round)
+        self.tloss_dev = None
+
+    # ------------------------------------------------------------------
+    def update_tloss(self, loss_l: torch.Tensor):
+        d = float(self.pr.tloss_decay)
+        if self.tloss_dev is None:
+            self.tloss_dev = loss_l.clone()
+        else:
+            self.tloss_dev = torch.where(
+                self.tloss_dev == 0,
+                loss_l,
+                (1 - d) * self.tloss_dev + d * loss_l,
+            )
 
     # ------------------------------------------------------------------
     def _stage_data(self):
@@ -284,45 +300,49 @@ class StackedEngine:
             )
 
     # ------------------------------------------------------------------
-    def _alloc_bufs(self):
-        L, B = self.L, self.B
+    def _alloc_bufs(self, M=None, train=True):
+        """Workspace pack for per-node batch size M (cached by M)."""
+        L = self.L
+        M = M if M is not None else self.B
         mk = lambda el: torch.empty(
-            L * B, el, device=self.device, dtype=self.dtype
+            L * M, el, device=self.device, dtype=self.dtype
         )
         acts, zs, dzs, idxs = [], [], [], []
         for layer in self.spec.layers:
             acts.append(mk(layer.out_elems))
             zs.append(
                 mk(layer.out_elems)
-                if layer.activation == "sin_relu"
+                if (train and layer.activation == "sin_relu")
                 else None
             )
-            dzs.append(mk(layer.out_elems))
+            dzs.append(mk(layer.out_elems) if train else None)
             idxs.append(
                 torch.empty(
-                    L * B, layer.out_elems, device=self.device,
+                    L * M, layer.out_elems, device=self.device,
                     dtype=torch.uint8,
                 )
                 if layer.kind == "conv_pool"
                 else None
             )
-        self._bufs = {
+        pack = {
+            "M": M,
             "acts": acts, "zs": zs, "dzs": dzs, "idxs": idxs,
             "xb": mk(self.spec.in_elems),
             "yb": torch.empty(
-                L * B, device=self.device,
+                L * M, device=self.device,
                 dtype=torch.long if self.classification else self.dtype,
             ),
             "logp": mk(self.spec.layers[-1].out_elems)
             if self.classification else None,
             "loss": torch.zeros(L, device=self.device, dtype=self.dtype),
         }
+        return pack
 
     # ------------------------------------------------------------------
     def next_batch(self):
         """Assemble the next per-node batches with two gather kernels."""
         if self._bufs is None:
-            self._alloc_bufs()
+            self._bufs = self._alloc_bufs()
         view = self.sampler.next_view()  # [L, B] (possibly strided)
         stride = view.stride(0)
         xb, yb = self._bufs["xb"], self._bufs["yb"]
@@ -333,13 +353,13 @@ class StackedEngine:
         return xb, yb
 
     # ------------------------------------------------------------------
-    def forward(self, xb):
+    def forward(self, xb, bufs=None):
         if self._bufs is None:
-            self._alloc_bufs()
-        bufs = self._bufs
+            self._bufs = self._alloc_bufs()
+        bufs = bufs if bufs is not None else self._bufs
         ext = self.ext
         cur = xb
-        M = self.B
+        M = bufs["M"]
         for li, layer in enumerate(self.spec.layers):
             out = bufs["acts"][li]
             if layer.kind == "conv_pool":
@@ -434,6 +454,89 @@ class StackedEngine:
                         ACT_IDS[lb.activation], lb.scale,
                     )
         return bufs["loss"] if want_loss else None
+
+    # ------------------------------------------------------------------
+    # batched validation: all local nodes evaluate the shared val set in
+    # one stacked forward per chunk (the eager per-node torch loop cost
+    # ~130 ms per evaluation at the paper config — 400x a training round)
+    def _stage_val(self):
+        if getattr(self, "_val", None) is not None:
+            return self._val
+        pr = self.pr
+        x, y = _dataset_tensors(pr.val_set)
+        xd = x.reshape(x.shape[0], -1).to(self.device, self.dtype)
+        if self.classification:
+            yd = y.long().to(self.device)
+        else:
+            yd = y.to(self.device, self.dtype)
+        self._val = (xd, yd)
+        self._val_pack = None
+        return self._val
+
+    def validate_all(self):
+        """Per-node validation on the shared val set.
+
+        Classification: (loss [L], top1 acc [L], correct [L, V] bool) —
+        replicating the reference's metric convention of summing batch
+        means then dividing by the dataset size
+        (problems/dist_mnist_problem.py:111-132).
+        Regression: summed batch-mean losses [L]
+        (dist_dense_problem.py:119-133).
+        """
+        pr = self.pr
+        xd, yd = self._stage_val()
+        V = xd.shape[0]
+        Bv = min(int(pr.conf.get("val_batch_size", 1024)), V)
+        if self._val_pack is None or self._val_pack["M"] != Bv:
+            self._val_pack = self._alloc_bufs(M=Bv, train=False)
+        pack = self._val_pack
+        L = self.L
+        loss_sum = torch.zeros(L, device=self.device, dtype=self.dtype)
+        if self.classification:
+            correct = torch.zeros(
+                L, V, device=self.device, dtype=torch.bool
+            )
+        nchunks = 0
+        for v0 in range(0, V, Bv):
+            v1 = min(V, v0 + Bv)
+            c = v1 - v0
+            xb = pack["xb"]
+            chunk = xd[v0:v1]
+            if c < Bv:  # ragged tail: pad with the first row
+                chunk = torch.cat(
+                    [chunk, xd[:1].expand(Bv - c, -1)], dim=0
+                )
+            xb.copy_(chunk.repeat(L, 1))
+            out = self.forward(xb, bufs=pack)
+            nchunks += 1
+            if self.classification:
+                logp = out.reshape(L, Bv, -1)[:, :c]
+                yv = yd[v0:v1]
+                loss_sum += torch.nn.functional.nll_loss(
+                    logp.reshape(L * c, -1),
+                    yv.repeat(L),
+                    reduction="none",
+                ).reshape(L, c).mean(dim=1)
+                correct[:, v0:v1] = logp.argmax(dim=2) == yv
+            else:
+                yhat = out.reshape(L, Bv)[:, :c]
+                yv = yd[v0:v1]
+                kind = self._loss_kind
+                if kind == "BCELoss":
+                    eps = 1e-12
+                    pcl = yhat.clamp(eps, 1 - eps)
+                    bl = -(
+                        yv * pcl.log() + (1 - yv) * (1 - pcl).log()
+                    ).mean(dim=1)
+                elif kind == "MSELoss":
+                    bl = ((yhat - yv) ** 2).mean(dim=1)
+                else:
+                    bl = (yhat - yv).abs().mean(dim=1)
+                loss_sum += bl
+        if self.classification:
+            acc = correct.sum(dim=1).to(self.dtype) / V
+            return loss_sum.cpu() / V, acc.cpu(), correct.cpu()
+        return loss_sum.cpu()
 
     # ------------------------------------------------------------------
     def flush_to_models(self):
@@ -581,10 +684,14 @@ class DiNNOStackedDriver:
             opt.primal_lr[0] if self.persistent else opt.primal_lr[k]
         )
 
-        for _ in range(self.pits):
+        want_tl = bool(getattr(pr, "track_tloss", False))
+        for pi in range(self.pits):
             xb, yb = eng.next_batch()
             eng.forward(xb)
-            eng.backward(xb, yb)
+            wl = want_tl and pi == self.pits - 1
+            lb = eng.backward(xb, yb, want_loss=wl)
+            if wl:
+                eng.update_tloss(lb)
             self.step_t += 1
             ext.fused_step(
                 eng.theta, eng.grad, self.duals, self.s, deg,
@@ -651,7 +758,10 @@ class DSGDStackedDriver:
 
         xb, yb = eng.next_batch()
         eng.forward(xb)
-        eng.backward(xb, yb)
+        want_tl = bool(getattr(pr, "track_tloss", False))
+        lb = eng.backward(xb, yb, want_loss=want_tl)
+        if want_tl:
+            eng.update_tloss(lb)
         ext.axpy(eng.theta, eng.grad, -self.alph)
 
     def run(self, profiler=None):
@@ -740,7 +850,10 @@ class DSGTStackedDriver:
         with _timer("forward"):
             eng.forward(xb)
         with _timer("backward"):
-            eng.backward(xb, yb)
+            want_tl = bool(getattr(pr, "track_tloss", False))
+            lb = eng.backward(xb, yb, want_loss=want_tl)
+            if want_tl:
+                eng.update_tloss(lb)
         with _timer("y_update"):
             ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
 

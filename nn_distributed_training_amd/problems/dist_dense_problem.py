@@ -66,9 +66,12 @@ class DistDensityProblem(ProblemBase):
                     distances_mean.amax().item(),
                 )
             elif met_name == "validation_loss":
-                vl = torch.tensor(
-                    [self.validate(i) for i in self.local_nodes]
-                )
+                if self.stacked is not None:
+                    vl = self.stacked.validate_all()
+                else:
+                    vl = torch.tensor(
+                        [self.validate(i) for i in self.local_nodes]
+                    )
                 val_losses = self.gather_per_node(vl)
                 self.metrics[met_name].append(val_losses)
                 evalprint += "Val Loss: {:.4f} - {:.4f} | ".format(

@@ -58,13 +58,23 @@ class DistMNISTProblem(ProblemBase):
         )
         if want_val:
             if self.stacked is not None:
+                # batched stacked-kernel validation: all local nodes in
+                # one forward per val chunk
+                vl, va, vc = self.stacked.validate_all()
+                loc_losses = vl.tolist()
+                loc_accs = va.tolist()
+                valid_vecs = {
+                    i: vc[li].reshape(-1, 1)
+                    for li, i in enumerate(self.local_nodes)
+                }
                 self.stacked.flush_to_models()
-            loc_losses, loc_accs, valid_vecs = [], [], {}
-            for i in self.local_nodes:
-                l, a, v = self.validate(i)
-                loc_losses.append(l)
-                loc_accs.append(a)
-                valid_vecs[i] = v
+            else:
+                loc_losses, loc_accs, valid_vecs = [], [], {}
+                for i in self.local_nodes:
+                    l, a, v = self.validate(i)
+                    loc_losses.append(l)
+                    loc_accs.append(a)
+                    valid_vecs[i] = v
             avg_losses = self.gather_per_node(torch.tensor(loc_losses))
             accs = self.gather_per_node(torch.tensor(loc_accs))
 

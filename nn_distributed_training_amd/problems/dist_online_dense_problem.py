@@ -168,9 +168,12 @@ class DistOnlineDensityProblem(ProblemBase):
                     distances_mean.amax().item(),
                 )
             elif met_name == "validation_loss":
-                vl = torch.tensor(
-                    [self.validate(i) for i in self.local_nodes]
-                )
+                if self.stacked is not None:
+                    vl = self.stacked.validate_all()
+                else:
+                    vl = torch.tensor(
+                        [self.validate(i) for i in self.local_nodes]
+                    )
                 val_losses = self.gather_per_node(vl)
                 self.metrics[met_name].append(val_losses)
                 evalprint += "Val Loss: {:.4f} - {:.4} - {:.4f} | ".format(
@@ -179,6 +182,12 @@ class DistOnlineDensityProblem(ProblemBase):
                     val_losses.amax().item(),
                 )
             elif met_name == "train_loss_moving_average":
+                if self.stacked is not None and \
+                        self.stacked.tloss_dev is not None:
+                    self.tloss_tracker[self.local_nodes] = \
+                        self.stacked.tloss_dev.cpu().to(
+                            self.tloss_tracker.dtype
+                        )
                 tl = self.gather_per_node(
                     self.tloss_tracker[self.local_nodes]
                 )

@@ -166,6 +166,27 @@ def test_stacked_density_matches_golden(alg):
 
 
 @requires_gpu
+def test_stacked_validation_matches_eager():
+    """Stacked batched validation == eager per-node torch validation."""
+    torch.set_default_dtype(torch.float64)
+    conf = _conf(copy.deepcopy(ALG_CONFS["dsgd"]))
+    pr = _build_problem(conf)
+    # desynchronize the replicas so per-node metrics differ
+    for i in pr.local_nodes:
+        with torch.no_grad():
+            for p in pr.models[i].parameters():
+                p.add_(0.01 * (i + 1) * torch.randn_like(p))
+    eager = [pr.validate(i) for i in pr.local_nodes]
+
+    pr.stacked = StackedEngine(pr)
+    vl, va, vc = pr.stacked.validate_all()
+    for li, (l, a, v) in enumerate(eager):
+        assert abs(vl[li].item() - l) < 1e-9
+        assert abs(va[li].item() - a) < 1e-12
+        assert vc[li].sum().item() == v.sum().item()
+
+
+@requires_gpu
 def test_stacked_fourier_forward_matches_module():
     """Stacked FourierNet forward == eager module forward (fp64)."""
     torch.set_default_dtype(torch.float64)
