@@ -99,6 +99,10 @@ def run_problem(problem, dopt, prob_conf, exp_conf, output_dir):
     """Optionally profile (parity with reference dist_mnist_ex.py:207-220),
     train, save metrics."""
     opt_conf = prob_conf["optimizer_config"]
+    # checkpoints land in the run dir unless the config says otherwise
+    opt_conf.setdefault("checkpoint_dir", output_dir)
+    if getattr(dopt, "checkpoint_dir", None) is None:
+        dopt.checkpoint_dir = output_dir
     if opt_conf.get("profile", False):
         with torch.profiler.profile(
             schedule=torch.profiler.schedule(

@@ -701,17 +701,22 @@ class DiNNOStackedDriver:
                 self.mode,
             )
 
+    def _opt_state(self):
+        return {
+            "duals": self.duals, "s": self.s, "m": self.m, "v": self.v,
+            "rho": self.rho, "step_t": self.step_t,
+        }
+
+    def _load_opt_state(self, st):
+        self.duals.copy_(st["duals"])
+        self.s.copy_(st["s"])
+        self.m.copy_(st["m"])
+        self.v.copy_(st["v"])
+        self.rho = st["rho"]
+        self.step_t = st["step_t"]
+
     def run(self, profiler=None):
-        pr = self.pr
-        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
-        oits = self.opt.conf["outer_iterations"]
-        self.prepare()
-        for k in range(oits):
-            if k % eval_every == 0 or k == oits - 1:
-                pr.evaluate_metrics(at_end=(k == oits - 1))
-            self.step_round(k)
-            if profiler is not None:
-                profiler.step()
+        _run_rounds(self, profiler)
 
 
 class DSGDStackedDriver:
@@ -764,17 +769,14 @@ class DSGDStackedDriver:
             eng.update_tloss(lb)
         ext.axpy(eng.theta, eng.grad, -self.alph)
 
+    def _opt_state(self):
+        return {"alph": self.alph}
+
+    def _load_opt_state(self, st):
+        self.alph = st["alph"]
+
     def run(self, profiler=None):
-        pr = self.pr
-        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
-        oits = self.opt.conf["outer_iterations"]
-        self.prepare()
-        for k in range(oits):
-            if k % eval_every == 0 or k == oits - 1:
-                pr.evaluate_metrics(at_end=(k == oits - 1))
-            self.step_round(k)
-            if profiler is not None:
-                profiler.step()
+        _run_rounds(self, profiler)
 
 
 class DSGTStackedDriver:
@@ -857,14 +859,51 @@ class DSGTStackedDriver:
         with _timer("y_update"):
             ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
 
+    def _opt_state(self):
+        return {"y": self.y, "g": self.g, "y_mix": self.y_mix}
+
+    def _load_opt_state(self, st):
+        self.y.copy_(st["y"])
+        self.g.copy_(st["g"])
+        self.y_mix.copy_(st["y_mix"])
+
     def run(self, profiler=None):
-        pr = self.pr
-        eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
-        oits = self.opt.conf["outer_iterations"]
-        self.prepare()
-        for k in range(oits):
-            if k % eval_every == 0 or k == oits - 1:
-                pr.evaluate_metrics(at_end=(k == oits - 1))
-            self.step_round(k)
-            if profiler is not None:
-                profiler.step()
+        _run_rounds(self, profiler, skip_bootstrap_on_resume=True)
+
+
+def _run_rounds(driver, profiler=None, skip_bootstrap_on_resume=False):
+    """Shared outer loop for the stacked drivers: eval cadence,
+    checkpoint/resume (optimizers/checkpointing.py), profiler hook."""
+    from ..optimizers.checkpointing import load_checkpoint, save_checkpoint
+
+    pr = driver.pr
+    conf = driver.opt.conf
+    eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
+    oits = conf["outer_iterations"]
+    ck_every = conf.get("checkpoint_every", 0)
+    ck_dir = conf.get("checkpoint_dir") or getattr(
+        driver.opt, "checkpoint_dir", None
+    )
+
+    resume = conf.get("resume_from")
+    if resume and skip_bootstrap_on_resume:
+        # prepare() runs the DSGT init_grads bootstrap; suppress it,
+        # the tracker state comes from the checkpoint
+        saved = conf.get("init_grads")
+        conf["init_grads"] = False
+        driver.prepare()
+        conf["init_grads"] = saved
+    else:
+        driver.prepare()
+    k0 = 0
+    if resume:
+        k0, st = load_checkpoint(resume, pr)
+        driver._load_opt_state(st)
+    for k in range(k0, oits):
+        if ck_every and k > k0 and k % ck_every == 0:
+            save_checkpoint(ck_dir, pr, k - 1, driver._opt_state())
+        if k % eval_every == 0 or k == oits - 1:
+            pr.evaluate_metrics(at_end=(k == oits - 1))
+        driver.step_round(k)
+        if profiler is not None:
+            profiler.step()

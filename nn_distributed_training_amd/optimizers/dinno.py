@@ -51,6 +51,7 @@ class DiNNO:
         if conf["persistant_primal_opt"]:
             for i in self.pr.local_nodes:
                 self.opts[i] = self._make_opt(i, float(self.primal_lr[0]))
+        self.checkpoint_dir = conf.get("checkpoint_dir")
 
     # ------------------------------------------------------------------
     @staticmethod
@@ -101,13 +102,39 @@ class DiNNO:
             opt.step()
 
     # ------------------------------------------------------------------
+    # -- checkpoint/resume (optimizers/checkpointing.py) -----------------
+    def _opt_state(self):
+        st = {"duals": self.duals, "rho": self.rho}
+        if self.conf["persistant_primal_opt"]:
+            st["opts"] = {i: o.state_dict() for i, o in self.opts.items()}
+        return st
+
+    def _load_opt_state(self, st):
+        self.duals = {
+            i: v.to(self.device) for i, v in st["duals"].items()
+        }
+        self.rho = st["rho"]
+        for i, sd in st.get("opts", {}).items():
+            self.opts[i].load_state_dict(sd)
+
     def train(self, profiler=None):
         if self.pr.stacked is not None:
             return self._train_stacked(profiler)
+        from .checkpointing import load_checkpoint, save_checkpoint
+
         pr = self.pr
         eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
         oits = self.conf["outer_iterations"]
-        for k in range(oits):
+        ck_every = self.conf.get("checkpoint_every", 0)
+        k0 = 0
+        if self.conf.get("resume_from"):
+            k0, st = load_checkpoint(self.conf["resume_from"], pr)
+            self._load_opt_state(st)
+        for k in range(k0, oits):
+            if ck_every and k > k0 and k % ck_every == 0:
+                save_checkpoint(
+                    self.checkpoint_dir, pr, k - 1, self._opt_state()
+                )
             if k % eval_every == 0 or k == oits - 1:
                 pr.evaluate_metrics(at_end=(k == oits - 1))
 

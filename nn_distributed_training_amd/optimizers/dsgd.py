@@ -31,16 +31,28 @@ class DSGD:
         self.device = torch.device(device)
         self.alph0 = conf["alpha0"]
         self.mu = conf["mu"]
+        self.checkpoint_dir = conf.get("checkpoint_dir")
 
     def train(self, profiler=None):
         if self.pr.stacked is not None:
             return self._train_stacked(profiler)
+        from .checkpointing import load_checkpoint, save_checkpoint
+
         pr = self.pr
         eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
         oits = self.conf["outer_iterations"]
+        ck_every = self.conf.get("checkpoint_every", 0)
 
         alph = self.alph0
-        for k in range(oits):
+        k0 = 0
+        if self.conf.get("resume_from"):
+            k0, st = load_checkpoint(self.conf["resume_from"], pr)
+            alph = st["alph"]
+        for k in range(k0, oits):
+            if ck_every and k > k0 and k % ck_every == 0:
+                save_checkpoint(
+                    self.checkpoint_dir, pr, k - 1, {"alph": alph}
+                )
             if k % eval_every == 0 or k == oits - 1:
                 pr.evaluate_metrics(at_end=(k == oits - 1))
 

@@ -39,6 +39,7 @@ class DSGT:
             i: torch.zeros(n, device=self.device)
             for i in self.pr.local_nodes
         }
+        self.checkpoint_dir = conf.get("checkpoint_dir")
 
     # ------------------------------------------------------------------
     def _local_grad_vector(self, i) -> torch.Tensor:
@@ -61,13 +62,26 @@ class DSGT:
         eval_every = pr.conf["metrics_config"]["evaluate_frequency"]
         oits = self.conf["outer_iterations"]
 
-        if self.conf["init_grads"]:
+        from .checkpointing import load_checkpoint, save_checkpoint
+
+        ck_every = self.conf.get("checkpoint_every", 0)
+        k0 = 0
+        if self.conf.get("resume_from"):
+            k0, st = load_checkpoint(self.conf["resume_from"], pr)
+            self.y = {i: v.to(self.device) for i, v in st["y"].items()}
+            self.g = {i: v.to(self.device) for i, v in st["g"].items()}
+        elif self.conf["init_grads"]:
             for i in pr.local_nodes:
                 g = self._local_grad_vector(i)
                 self.y[i] = g.clone()
                 self.g[i] = g.clone()
 
-        for k in range(oits):
+        for k in range(k0, oits):
+            if ck_every and k > k0 and k % ck_every == 0:
+                save_checkpoint(
+                    self.checkpoint_dir, pr, k - 1,
+                    {"y": self.y, "g": self.g},
+                )
             if k % eval_every == 0 or k == oits - 1:
                 pr.evaluate_metrics(at_end=(k == oits - 1))
 

@@ -166,6 +166,38 @@ def test_stacked_density_matches_golden(alg):
 
 
 @requires_gpu
+@pytest.mark.parametrize("alg", ["dinno", "dsgt"])
+def test_stacked_checkpoint_resume(alg, tmp_path):
+    torch.set_default_dtype(torch.float64)
+    conf = _conf(copy.deepcopy(ALG_CONFS[alg]))
+    conf["optimizer_config"]["outer_iterations"] = 6
+    conf["optimizer_config"]["checkpoint_every"] = 4
+    conf["optimizer_config"]["checkpoint_dir"] = str(tmp_path)
+
+    pr = _build_problem(conf)
+    pr.stacked = StackedEngine(pr)
+    opt = build_optimizer(pr, pr.device, conf["optimizer_config"])
+    opt.train()
+    ckpt = tmp_path / "parity_ckpt_rank0.pt"
+    assert ckpt.exists()
+
+    conf2 = _conf(copy.deepcopy(ALG_CONFS[alg]))
+    conf2["optimizer_config"]["outer_iterations"] = 6
+    conf2["optimizer_config"]["resume_from"] = str(tmp_path)
+    pr2 = _build_problem(conf2)
+    pr2.stacked = StackedEngine(pr2)
+    opt2 = build_optimizer(pr2, pr2.device, conf2["optimizer_config"])
+    init_theta = pr2.stacked.theta.clone()  # pre-train (fresh init)
+    opt2.train()
+    assert torch.isfinite(pr2.stacked.theta).all()
+    payload = torch.load(ckpt, weights_only=False)
+    assert payload["round"] == 3
+    # the resumed run trained (moved past the restored checkpoint) and
+    # did not restart from the fresh initialization
+    assert not torch.allclose(pr2.stacked.theta, init_theta)
+
+
+@requires_gpu
 def test_stacked_validation_matches_eager():
     """Stacked batched validation == eager per-node torch validation."""
     torch.set_default_dtype(torch.float64)
