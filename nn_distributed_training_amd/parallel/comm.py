@@ -26,10 +26,33 @@ exchange degenerates to local reads, which is the single-GPU packed mode
 
 from __future__ import annotations
 
+import os
+from datetime import timedelta
 from typing import Dict, List, Sequence
 
 import torch
 import torch.distributed as dist
+
+#: deadlock watchdog: P2P waits that exceed this raise instead of
+#: hanging the job (a disconnected dynamic graph or a desynchronized
+#: schedule shows up as a stuck recv). Override: NDTA_COMM_TIMEOUT_S.
+COMM_TIMEOUT_S = float(os.environ.get("NDTA_COMM_TIMEOUT_S", "300"))
+
+
+def _wait_all(works, what: str, rank: int):
+    """Wait for a batch of P2P works with the watchdog timeout."""
+    for w in works:
+        try:
+            ok = w.wait(timedelta(seconds=COMM_TIMEOUT_S))
+        except TypeError:  # backend without timeout support
+            w.wait()
+            ok = True
+        if ok is False:
+            raise RuntimeError(
+                f"[rank {rank}] neighbor-exchange watchdog: {what} did "
+                f"not complete within {COMM_TIMEOUT_S}s — check graph "
+                "connectivity / schedule symmetry"
+            )
 
 
 class NodeLayout:
@@ -132,8 +155,8 @@ class Communicator:
             for d in dests:
                 ops.append(dist.P2POp(dist.irecv, d[j], peer))
         if ops:
-            for work in dist.batch_isend_irecv(ops):
-                work.wait()
+            _wait_all(dist.batch_isend_irecv(ops),
+                      f"{len(ops)} P2P ops", self.rank)
 
     # ------------------------------------------------------------------
     def exchange_node_vectors(
@@ -188,8 +211,8 @@ class Communicator:
         for j in sorted(recv_nodes, key=lambda j: (layout.rank_of(j), j)):
             ops.append(dist.P2POp(dist.irecv, recv_bufs[j], layout.rank_of(j)))
         if ops:
-            for work in dist.batch_isend_irecv(ops):
-                work.wait()
+            _wait_all(dist.batch_isend_irecv(ops),
+                      f"{len(ops)} P2P ops", self.rank)
         return recv_bufs
 
     # ------------------------------------------------------------------
@@ -245,7 +268,10 @@ def init_from_env(backend: str | None = None) -> tuple:
     if not dist.is_initialized():
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
-        dist.init_process_group(backend=backend)
+        dist.init_process_group(
+            backend=backend,
+            timeout=timedelta(seconds=max(COMM_TIMEOUT_S, 60.0)),
+        )
     rank = dist.get_rank()
     world = dist.get_world_size()
     local_rank = int(os.environ.get("LOCAL_RANK", rank))

@@ -1,0 +1,169 @@
+"""Figure generation from experiment result files.
+
+The reference produces its paper figures from Jupyter notebooks
+(visualization/*.ipynb) reading ``<problem>_results.pt``; this script
+covers the same plots headlessly:
+
+  * top-1 accuracy band (min-max over nodes) vs communication rounds
+  * validation-loss band
+  * consensus error (log scale) vs rounds
+  * scaling summary (rounds/sec and convergence vs N) from
+    dist_mnist_scaling's ``scaling_summary.pt``
+  * RL reward / agreement curves from the trained/ npy+npz artifacts
+
+Usage:
+  python visualization/plot_results.py <run_dir> [--out figs/]
+  python visualization/plot_results.py --rl trained/ [--out figs/]
+"""
+
+from __future__ import annotations
+
+import argparse
+import glob
+import os
+
+import matplotlib
+
+matplotlib.use("Agg")
+import matplotlib.pyplot as plt  # noqa: E402
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+
+def _band(ax, xs, series, label):
+    arr = torch.stack(
+        [torch.as_tensor(s, dtype=torch.float64) for s in series]
+    )
+    lo = arr.amin(dim=1)
+    hi = arr.amax(dim=1)
+    mid = arr.mean(dim=1)
+    ax.plot(xs, mid, label=label)
+    ax.fill_between(xs, lo, hi, alpha=0.25)
+
+
+def plot_run_dir(run_dir: str, out_dir: str, eval_every: int = 20):
+    os.makedirs(out_dir, exist_ok=True)
+    results = sorted(glob.glob(os.path.join(run_dir, "*_results.pt")))
+    if not results:
+        print(f"no *_results.pt under {run_dir}")
+        return
+
+    for metric, fname, ylabel, logy in (
+        ("top1_accuracy", "accuracy.png", "top-1 accuracy", False),
+        ("validation_loss", "val_loss.png", "validation loss", False),
+    ):
+        fig, ax = plt.subplots(figsize=(6, 4))
+        drew = False
+        for rp in results:
+            name = os.path.basename(rp).replace("_results.pt", "")
+            res = torch.load(rp, weights_only=False)
+            if metric not in res or not res[metric]:
+                continue
+            xs = np.arange(len(res[metric])) * eval_every
+            _band(ax, xs, res[metric], name)
+            drew = True
+        if drew:
+            ax.set_xlabel("communication rounds")
+            ax.set_ylabel(ylabel)
+            if logy:
+                ax.set_yscale("log")
+            ax.legend()
+            fig.tight_layout()
+            fig.savefig(os.path.join(out_dir, fname), dpi=130)
+        plt.close(fig)
+
+    # consensus error (max pairwise distance, log scale)
+    fig, ax = plt.subplots(figsize=(6, 4))
+    drew = False
+    for rp in results:
+        name = os.path.basename(rp).replace("_results.pt", "")
+        res = torch.load(rp, weights_only=False)
+        if "consensus_error" not in res or not res["consensus_error"]:
+            continue
+        vals = [
+            float(torch.as_tensor(c[1]).amax())
+            for c in res["consensus_error"]
+        ]
+        xs = np.arange(len(vals)) * eval_every
+        ax.plot(xs, np.maximum(vals, 1e-12), label=name)
+        drew = True
+    if drew:
+        ax.set_xlabel("communication rounds")
+        ax.set_ylabel("max consensus error")
+        ax.set_yscale("log")
+        ax.legend()
+        fig.tight_layout()
+        fig.savefig(os.path.join(out_dir, "consensus.png"), dpi=130)
+    plt.close(fig)
+
+    summary = os.path.join(run_dir, "scaling_summary.pt")
+    if os.path.exists(summary):
+        s = torch.load(summary, weights_only=False)
+        fig, ax = plt.subplots(figsize=(6, 4))
+        by_alg = {}
+        for (t, alg), d in s.items():
+            by_alg.setdefault(alg, []).append(
+                (d["N"], d["rounds_per_sec"])
+            )
+        for alg, pts in by_alg.items():
+            pts.sort()
+            ax.plot([p[0] for p in pts], [p[1] for p in pts], "o-",
+                    label=alg)
+        ax.set_xlabel("graph nodes N")
+        ax.set_ylabel("comm rounds / sec")
+        ax.legend()
+        fig.tight_layout()
+        fig.savefig(os.path.join(out_dir, "scaling.png"), dpi=130)
+        plt.close(fig)
+    print(f"figures -> {out_dir}")
+
+
+def plot_rl_dir(rl_dir: str, out_dir: str):
+    os.makedirs(out_dir, exist_ok=True)
+    fig, ax = plt.subplots(figsize=(6, 4))
+    for rews_f in sorted(glob.glob(
+        os.path.join(rl_dir, "avg_ep_rews_*.npy")
+    )):
+        tag = os.path.basename(rews_f)[len("avg_ep_rews_"):-4]
+        rews = np.load(rews_f)
+        ts_f = os.path.join(rl_dir, f"timesteps_{tag}.npy")
+        xs = np.load(ts_f) if os.path.exists(ts_f) \
+            else np.arange(len(rews))
+        ax.plot(xs, rews, label=tag)
+    ax.set_xlabel("environment timesteps")
+    ax.set_ylabel("avg episodic reward")
+    ax.legend(fontsize=7)
+    fig.tight_layout()
+    fig.savefig(os.path.join(out_dir, "rl_rewards.png"), dpi=130)
+    plt.close(fig)
+
+    fig, ax = plt.subplots(figsize=(6, 4))
+    for ag_f in sorted(glob.glob(
+        os.path.join(rl_dir, "agreements_*.npz")
+    )):
+        tag = os.path.basename(ag_f)[len("agreements_"):-4]
+        ag = np.load(ag_f)
+        ax.plot(ag["actor"].max(axis=(1, 2)), label=f"{tag} actor")
+        ax.plot(ag["critic"].max(axis=(1, 2)), "--",
+                label=f"{tag} critic")
+    ax.set_xlabel("iteration")
+    ax.set_ylabel("max agreement distance")
+    ax.set_yscale("log")
+    ax.legend(fontsize=7)
+    fig.tight_layout()
+    fig.savefig(os.path.join(out_dir, "rl_agreements.png"), dpi=130)
+    plt.close(fig)
+    print(f"RL figures -> {out_dir}")
+
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("run_dir", nargs="?", default=None)
+    p.add_argument("--rl", default=None)
+    p.add_argument("--out", default="./figs")
+    p.add_argument("--eval-every", type=int, default=20)
+    args = p.parse_args()
+    if args.rl:
+        plot_rl_dir(args.rl, args.out)
+    if args.run_dir:
+        plot_run_dir(args.run_dir, args.out, args.eval_every)
