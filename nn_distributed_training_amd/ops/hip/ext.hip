@@ -236,24 +236,37 @@ void act_grad(torch::Tensor dY, torch::Tensor Y,
   HIP_CHECK_LAST();
 }
 
+// dX = dZ @ W with the BELOW layer's activation backward fused when
+// act_below != 0 (Yb = below activation output, Zb = its pre-activation
+// where needed, i.e. sin_relu).
 void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
-                   torch::Tensor dX, long w_off, long M, long I, long O) {
+                   torch::Tensor dX,
+                   c10::optional<torch::Tensor> Yb,
+                   c10::optional<torch::Tensor> Zb,
+                   long act_below, double scale_below,
+                   long w_off, long M, long I, long O) {
   CHECK_DEV(dZ); CHECK_DEV(dX);
   const long L = theta.size(0), n = theta.size(1);
   const bool use_mfma = (M >= 128 && I >= 16 && O >= 8);
   DISPATCH_FT(dZ, {
+    auto ybp = Yb.has_value() ? Yb->data_ptr<scalar_t>() : nullptr;
+    auto zbp = Zb.has_value() ? Zb->data_ptr<scalar_t>() : nullptr;
+    TORCH_CHECK(act_below == 0 || ybp != nullptr,
+                "act_below needs Yb");
     if (use_mfma) {
       dim3 grid((I + 63) / 64, (M + 63) / 64, L);
       hipLaunchKernelGGL(gmfma::mfma_dx_k<scalar_t>,
           grid, dim3(256), 0, cur_stream(),
           dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
-          dX.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+          dX.data_ptr<scalar_t>(), ybp, zbp, (int)act_below,
+          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O);
     } else {
       dim3 grid((I + 15) / 16, (M + 15) / 16, L);
       hipLaunchKernelGGL(gemm::linear_bwd_dx_k<scalar_t>,
           grid, dim3(16, 16), 0, cur_stream(),
           dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
-          dX.data_ptr<scalar_t>(), n, w_off, (int)M, (int)I, (int)O);
+          dX.data_ptr<scalar_t>(), ybp, zbp, (int)act_below,
+          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O);
     }
   });
   HIP_CHECK_LAST();
@@ -332,9 +345,11 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
   CHECK_DEV(dY); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
   TORCH_CHECK(K <= 7, "conv_pool_bwd supports kernel size <= 7");
-  // chunk the batch so the grid fills the 256-CU chip; accumulation is
-  // atomic, so the caller must zero the conv slice of gstack first
-  const int nchunk = (int)std::min<long>(B, 32);
+  // chunk the batch so the grid fills the chip; accumulation is atomic
+  // (caller zeroes the grad stack). nchunk trades occupancy against
+  // f64-atomic contention on the 26 dw/db words per (l, f): 32 chunks
+  // serialized ~100 us/call on the MNIST bench, 8 chunks ~4x less.
+  const int nchunk = (int)std::min<long>(B, 8);
   DISPATCH_FT(dY, {
     hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
         dim3(L * F * nchunk), dim3(256), 0, cur_stream(),

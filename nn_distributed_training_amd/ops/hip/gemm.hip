@@ -70,11 +70,14 @@ __global__ void act_grad_k(
   }
 }
 
-// dX[M,I] = dZ[M,O] @ W[O,I]
+// dX[M,I] = dZ[M,O] @ W[O,I], below-layer activation bwd fused
 template <typename T>
 __global__ void linear_bwd_dx_k(
     const T* __restrict__ dZ, const T* __restrict__ theta,
-    T* __restrict__ dX, long n, long w_off, int M, int I, int O) {
+    T* __restrict__ dX,
+    const T* __restrict__ Yb, const T* __restrict__ Zb,
+    int act_below, T scale_below,
+    long n, long w_off, int M, int I, int O) {
   __shared__ T gs[TILE][TILE + 1];
   __shared__ T ws[TILE][TILE + 1];
   const long l = blockIdx.z;
@@ -102,7 +105,12 @@ __global__ void linear_bwd_dx_k(
   }
   const int m = m0 + tm, i = i0 + ti;
   if (m < M && i < I) {
-    dX[l * (long)M * I + (long)m * I + i] = acc;
+    const long off = l * (long)M * I + (long)m * I + i;
+    if (act_below != ACT_NONE) {
+      acc *= act_bwd(act_below, Zb ? Zb[off] : T(0), Yb[off],
+                     scale_below);
+    }
+    dX[off] = acc;
   }
 }
 

@@ -142,11 +142,17 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
 }
 
 // ---------------------------------------------------------------------
-// NN: dX[M,I] = dZ[M,O] @ W_l[O,I]   (contraction over O)
+// NN: dX[M,I] = dZ[M,O] @ W_l[O,I]   (contraction over O), with the
+// BELOW layer's activation backward fused into the epilogue
+// (dX *= act'(z_below, y_below)): the separate memory-bound act_grad
+// pass was ~14% of the density round.
 template <typename T>
 __global__ __launch_bounds__(256) void mfma_dx_k(
     const T* __restrict__ dZ, const T* __restrict__ theta,
-    T* __restrict__ dX, long n, long w_off, int M, int I, int O) {
+    T* __restrict__ dX,
+    const T* __restrict__ Yb, const T* __restrict__ Zb,  // may be null
+    int act_below, T scale_below,
+    long n, long w_off, int M, int I, int O) {
   using MF = mfma_t<T>;
   using acc_t = typename MF::acc_t;
   __shared__ T As[BK][BM + 1];   // As[k=o][m] = dZ[m][o]
@@ -205,7 +211,13 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
         const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
         const int i = i0 + wn + fn * 16 + (lane & 15);
         if (m < M && i < I) {
-          dX[l * (long)M * I + (long)m * I + i] = acc[fm][fn][r];
+          const long off = l * (long)M * I + (long)m * I + i;
+          T v = acc[fm][fn][r];
+          if (act_below != ACT_NONE) {
+            v *= act_bwd(act_below, Zb ? Zb[off] : T(0), Yb[off],
+                         scale_below);
+          }
+          dX[off] = v;
         }
       }
     }

@@ -441,18 +441,21 @@ class StackedEngine:
                 layer.in_dim, layer.out_dim,
             )
             if li > 0:
-                dy_below = bufs["dzs"][li - 1]
-                ext.linear_bwd_dx(
-                    dz, self.theta, dy_below, layer.w_off, M,
-                    layer.in_dim, layer.out_dim,
-                )
+                # dX with the below layer's activation bwd fused into
+                # the epilogue (no separate act_grad pass)
+                dz_below = bufs["dzs"][li - 1]
                 lb = layers[li - 1]
-                if lb.activation not in ("none", "logsoftmax"):
-                    ext.act_grad(
-                        dy_below, bufs["acts"][li - 1],
-                        bufs["zs"][li - 1], dy_below,
-                        ACT_IDS[lb.activation], lb.scale,
-                    )
+                act_b = (
+                    ACT_IDS[lb.activation]
+                    if lb.activation not in ("none", "logsoftmax")
+                    else 0
+                )
+                ext.linear_bwd_dx(
+                    dz, self.theta, dz_below,
+                    bufs["acts"][li - 1] if act_b else None,
+                    bufs["zs"][li - 1], act_b, lb.scale,
+                    layer.w_off, M, layer.in_dim, layer.out_dim,
+                )
         return bufs["loss"] if want_loss else None
 
     # ------------------------------------------------------------------

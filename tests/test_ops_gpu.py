@@ -116,7 +116,7 @@ def test_linear_bwd(ext, dtype, L, M, I, O):
     dZ = torch.empty_like(dY)
     ext.act_grad(dY, Y, None, dZ, 1, 1.0)
     dX = torch.empty(L * M, I, dtype=dtype, device=dev)
-    ext.linear_bwd_dx(dZ, theta, dX, 0, M, I, O)
+    ext.linear_bwd_dx(dZ, theta, dX, None, None, 0, 1.0, 0, M, I, O)
     gstack = torch.zeros_like(theta)
     ext.linear_bwd_dw(dZ, X.detach(), gstack, 0, I * O, M, I, O)
 
