@@ -669,40 +669,51 @@ class DiNNOStackedDriver:
         ext = eng.ext
 
         self.rho *= opt.rho_scaling
-        pr.update_graph()
-        rbuf, dests, offs, idx, deg = self._round_plan()
-        if pr.comm.world > 1:
-            pr.comm.exchange_rows(
-                pr.layout, list(pr.graph.edges()), [eng.theta], [dests]
+        with _timer("update_graph"):
+            pr.update_graph()
+        with _timer("round_plan"):
+            rbuf, dests, offs, idx, deg = self._round_plan()
+        with _timer("exchange"):
+            if pr.comm.world > 1:
+                pr.comm.exchange_rows(
+                    pr.layout, list(pr.graph.edges()), [eng.theta],
+                    [dests],
+                )
+        with _timer("dual_threg"):
+            ext.dinno_dual_threg(
+                eng.theta, rbuf, offs, idx, self.duals, self.s, self.rho
             )
-        ext.dinno_dual_threg(
-            eng.theta, rbuf, offs, idx, self.duals, self.s, self.rho
-        )
 
-        if not self.persistent:
-            self.m.zero_()
-            self.v.zero_()
-            self.step_t = 0
-        lr = float(
-            opt.primal_lr[0] if self.persistent else opt.primal_lr[k]
-        )
+        with _timer("state_reset"):
+            if not self.persistent:
+                self.m.zero_()
+                self.v.zero_()
+                self.step_t = 0
+            lr = float(
+                opt.primal_lr[0] if self.persistent
+                else opt.primal_lr[k]
+            )
 
         want_tl = bool(getattr(pr, "track_tloss", False))
         for pi in range(self.pits):
-            xb, yb = eng.next_batch()
-            eng.forward(xb)
-            wl = want_tl and pi == self.pits - 1
-            lb = eng.backward(xb, yb, want_loss=wl)
-            if wl:
-                eng.update_tloss(lb)
-            self.step_t += 1
-            ext.fused_step(
-                eng.theta, eng.grad, self.duals, self.s, deg,
-                None if self.mode == 2 else self.m,
-                None if self.mode == 2 else self.v,
-                self.rho, lr, 0.9, 0.999, 1e-8, self.wd, self.step_t,
-                self.mode,
-            )
+            with _timer("next_batch"):
+                xb, yb = eng.next_batch()
+            with _timer("forward"):
+                eng.forward(xb)
+            with _timer("backward"):
+                wl = want_tl and pi == self.pits - 1
+                lb = eng.backward(xb, yb, want_loss=wl)
+                if wl:
+                    eng.update_tloss(lb)
+            with _timer("fused_step"):
+                self.step_t += 1
+                ext.fused_step(
+                    eng.theta, eng.grad, self.duals, self.s, deg,
+                    None if self.mode == 2 else self.m,
+                    None if self.mode == 2 else self.v,
+                    self.rho, lr, 0.9, 0.999, 1e-8, self.wd,
+                    self.step_t, self.mode,
+                )
 
     def _opt_state(self):
         return {
