@@ -104,14 +104,20 @@ class _StreamSampler:
     """
 
     def __init__(self, lengths, batch, device, seed, epoch_cb,
-                 stream_batches=256):
+                 stream_batches=1024):
         self.lengths = lengths
         self.B = batch
         self.S = stream_batches * batch
         self.device = device
         self.epoch_cb = epoch_cb
+        # device-resident generators on GPU: permutations are produced
+        # where they are consumed, no host randperm + H2D per epoch
+        # (host refills were the top cost of the MNIST round — see
+        # profiles/README.md)
         self.gens = [
-            torch.Generator().manual_seed(seed * 100003 + i)
+            torch.Generator(device=device).manual_seed(
+                seed * 100003 + i
+            )
             for i in range(len(lengths))
         ]
         self.consumed = [0] * len(lengths)
@@ -122,14 +128,15 @@ class _StreamSampler:
         self.pos = self.S  # force initial fill
 
     def _refill(self):
-        rows = []
         for li, n in enumerate(self.lengths):
-            parts, tot = [], 0
-            while tot < self.S:
-                parts.append(torch.randperm(n, generator=self.gens[li]))
-                tot += n
-            rows.append(torch.cat(parts)[: self.S])
-        self.stream.copy_(torch.stack(rows).to(self.device))
+            filled = 0
+            while filled < self.S:
+                take = min(n, self.S - filled)
+                perm = torch.randperm(
+                    n, generator=self.gens[li], device=self.device
+                )
+                self.stream[li, filled : filled + take] = perm[:take]
+                filled += take
         self.pos = 0
 
     def next_view(self) -> torch.Tensor:
@@ -138,12 +145,14 @@ class _StreamSampler:
             self._refill()
         view = self.stream[:, self.pos : self.pos + self.B]
         self.pos += self.B
+        # epoch bookkeeping: cheap integer math, callbacks on crossings
         for li, n in enumerate(self.lengths):
             self.consumed[li] += self.B
             ep = self.consumed[li] // n
-            while self.reported_epochs[li] < ep:
-                self.reported_epochs[li] += 1
-                self.epoch_cb(li)
+            if self.reported_epochs[li] < ep:
+                for _ in range(ep - self.reported_epochs[li]):
+                    self.epoch_cb(li)
+                self.reported_epochs[li] = ep
         return view
 
 
