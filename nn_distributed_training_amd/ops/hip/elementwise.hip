@@ -184,16 +184,17 @@ __global__ void fused_step_k(
 template <typename T>
 __global__ void gather_batch_k(
     const T* __restrict__ X_all,       // [L, maxlen, F]
-    const long* __restrict__ idx,      // [L, B] rows strided by idx_stride
+    const long* __restrict__ idx,      // [L, S] index stream
     T* __restrict__ out,               // [L*B, F]
-    long maxlen, long Fdim, long B, long idx_stride, long total) {
+    long maxlen, long Fdim, long B, long idx_stride, long idx_off,
+    long total) {
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
     const long f = t % Fdim;
     const long lb = t / Fdim;
     const long l = lb / B;
     const long b = lb - l * B;
-    const long src = idx[l * idx_stride + b];
+    const long src = idx[l * idx_stride + idx_off + b];
     out[t] = X_all[(l * maxlen + src) * Fdim + f];
   }
 }
@@ -203,12 +204,12 @@ template <typename T>
 __global__ void gather_targets_k(
     const T* __restrict__ Y_all, const long* __restrict__ idx,
     T* __restrict__ out, long maxlen, long B, long idx_stride,
-    long total) {
+    long idx_off, long total) {
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
     const long l = t / B;
     const long b = t - l * B;
-    out[t] = Y_all[l * maxlen + idx[l * idx_stride + b]];
+    out[t] = Y_all[l * maxlen + idx[l * idx_stride + idx_off + b]];
   }
 }
 

@@ -89,7 +89,7 @@ void dsgt_mix(torch::Tensor p_loc, torch::Tensor y_loc,
 }
 
 void gather_batch(torch::Tensor X_all, torch::Tensor idx,
-                  torch::Tensor out, long idx_stride) {
+                  torch::Tensor out, long idx_stride, long idx_off) {
   CHECK_DEV(X_all); CHECK_DEV(out);
   const long L = X_all.size(0), maxlen = X_all.size(1),
              Fdim = X_all.size(2);
@@ -99,13 +99,14 @@ void gather_batch(torch::Tensor X_all, torch::Tensor idx,
     hipLaunchKernelGGL(ew::gather_batch_k<scalar_t>,
         dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
         X_all.data_ptr<scalar_t>(), idx.data_ptr<long>(),
-        out.data_ptr<scalar_t>(), maxlen, Fdim, B, idx_stride, total);
+        out.data_ptr<scalar_t>(), maxlen, Fdim, B, idx_stride,
+        idx_off, total);
   });
   HIP_CHECK_LAST();
 }
 
 void gather_targets(torch::Tensor Y_all, torch::Tensor idx,
-                    torch::Tensor out, long idx_stride) {
+                    torch::Tensor out, long idx_stride, long idx_off) {
   CHECK_DEV(Y_all); CHECK_DEV(out);
   const long L = Y_all.size(0), maxlen = Y_all.size(1);
   const long B = out.numel() / L;
@@ -114,13 +115,14 @@ void gather_targets(torch::Tensor Y_all, torch::Tensor idx,
     hipLaunchKernelGGL(ew::gather_targets_k<long>,
         dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
         Y_all.data_ptr<long>(), idx.data_ptr<long>(),
-        out.data_ptr<long>(), maxlen, B, idx_stride, total);
+        out.data_ptr<long>(), maxlen, B, idx_stride, idx_off, total);
   } else {
     DISPATCH_FT(Y_all, {
       hipLaunchKernelGGL(ew::gather_targets_k<scalar_t>,
           dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
           Y_all.data_ptr<scalar_t>(), idx.data_ptr<long>(),
-          out.data_ptr<scalar_t>(), maxlen, B, idx_stride, total);
+          out.data_ptr<scalar_t>(), maxlen, B, idx_stride, idx_off,
+          total);
     });
   }
   HIP_CHECK_LAST();

@@ -110,14 +110,11 @@ class _StreamSampler:
         self.S = stream_batches * batch
         self.device = device
         self.epoch_cb = epoch_cb
-        # device-resident generators on GPU: permutations are produced
-        # where they are consumed, no host randperm + H2D per epoch
-        # (host refills were the top cost of the MNIST round — see
-        # profiles/README.md)
+        # CPU generators fill a pinned staging buffer; ONE async H2D per
+        # refill. (Device-side randperms were tried and regressed: each
+        # is a multi-launch rocprim sort, ~1.5k launches per refill.)
         self.gens = [
-            torch.Generator(device=device).manual_seed(
-                seed * 100003 + i
-            )
+            torch.Generator().manual_seed(seed * 100003 + i)
             for i in range(len(lengths))
         ]
         self.consumed = [0] * len(lengths)
@@ -125,27 +122,31 @@ class _StreamSampler:
         self.stream = torch.empty(
             len(lengths), self.S, dtype=torch.long, device=device
         )
+        self._staging = torch.empty(
+            len(lengths), self.S, dtype=torch.long,
+            pin_memory=(device.type == "cuda"),
+        )
         self.pos = self.S  # force initial fill
 
     def _refill(self):
+        st = self._staging
         for li, n in enumerate(self.lengths):
             filled = 0
             while filled < self.S:
                 take = min(n, self.S - filled)
-                perm = torch.randperm(
-                    n, generator=self.gens[li], device=self.device
-                )
-                self.stream[li, filled : filled + take] = perm[:take]
+                perm = torch.randperm(n, generator=self.gens[li])
+                st[li, filled : filled + take] = perm[:take]
                 filled += take
+        self.stream.copy_(st, non_blocking=True)
         self.pos = 0
 
-    def next_view(self) -> torch.Tensor:
-        """[L, B] strided view into the stream + its row stride."""
+    def next_ref(self):
+        """(stream tensor, row stride, offset) for the next batch —
+        no per-draw tensor slicing."""
         if self.pos + self.B > self.S:
             self._refill()
-        view = self.stream[:, self.pos : self.pos + self.B]
+        off = self.pos
         self.pos += self.B
-        # epoch bookkeeping: cheap integer math, callbacks on crossings
         for li, n in enumerate(self.lengths):
             self.consumed[li] += self.B
             ep = self.consumed[li] // n
@@ -153,7 +154,7 @@ class _StreamSampler:
                 for _ in range(ep - self.reported_epochs[li]):
                     self.epoch_cb(li)
                 self.reported_epochs[li] = ep
-        return view
+        return self.stream, self.S, off
 
 
 class _OnlineWindowSampler:
@@ -196,7 +197,7 @@ class _OnlineWindowSampler:
         self.pools[li] = perm + lb
         self.pos[li] = 0
 
-    def next_view(self) -> torch.Tensor:
+    def next_ref(self):
         for li in range(len(self.dss)):
             p = self.pos[li]
             pool = self.pools[li]
@@ -209,7 +210,7 @@ class _OnlineWindowSampler:
                 take = self.B - head.numel()
                 self.buf[li] = torch.cat([head, self.pools[li][:take]])
                 self.pos[li] = take
-        return self.buf
+        return self.buf, self.B, 0
 
 
 class StackedEngine:
@@ -242,6 +243,7 @@ class StackedEngine:
 
         self._stage_data()
         self.B = problem.conf["train_batch_size"]
+        self._has_node0 = 0 in self.local_nodes
         self._bufs = None
         self._loss_kind = type(problem.base_loss).__name__  # NLLLoss etc.
         # device-resident train-loss EMA (the reference's tloss tracker,
@@ -352,13 +354,12 @@ class StackedEngine:
         """Assemble the next per-node batches with two gather kernels."""
         if self._bufs is None:
             self._bufs = self._alloc_bufs()
-        view = self.sampler.next_view()  # [L, B] (possibly strided)
-        stride = view.stride(0)
+        idx, stride, off = self.sampler.next_ref()
         xb, yb = self._bufs["xb"], self._bufs["yb"]
-        self.ext.gather_batch(self.X_all, view, xb, stride)
-        self.ext.gather_targets(self.Y_all, view, yb, stride)
-        if 0 in self.local_nodes:
-            self.pr.forward_cnt += self.pr.conf["train_batch_size"]
+        self.ext.gather_batch(self.X_all, idx, xb, stride, off)
+        self.ext.gather_targets(self.Y_all, idx, yb, stride, off)
+        if self._has_node0:
+            self.pr.forward_cnt += self.B
         return xb, yb
 
     # ------------------------------------------------------------------

@@ -329,8 +329,8 @@ def test_gather_batch(ext, dtype):
     view = stream[:, 5 : 5 + B]  # strided view (row stride S)
     xb = torch.empty(L * B, Fdim, dtype=dtype, device=dev)
     yb = torch.empty(L * B, dtype=torch.long, device=dev)
-    ext.gather_batch(X, view, xb, view.stride(0))
-    ext.gather_targets(Yt, view, yb, view.stride(0))
+    ext.gather_batch(X, view.contiguous(), xb, B, 0)
+    ext.gather_targets(Yt, view.contiguous(), yb, B, 0)
     ar = torch.arange(L, device=dev).unsqueeze(1)
     torch.testing.assert_close(
         xb, X[ar, view].reshape(L * B, Fdim), rtol=0, atol=0
@@ -338,6 +338,13 @@ def test_gather_batch(ext, dtype):
     torch.testing.assert_close(
         yb, Yt[ar, view].reshape(-1), rtol=0, atol=0
     )
+    # stride+offset form: gather straight out of the stream tensor
+    xb2 = torch.empty_like(xb)
+    yb2 = torch.empty_like(yb)
+    ext.gather_batch(X, stream, xb2, S, 5)
+    ext.gather_targets(Yt, stream, yb2, S, 5)
+    torch.testing.assert_close(xb2, xb, rtol=0, atol=0)
+    torch.testing.assert_close(yb2, yb, rtol=0, atol=0)
 
 
 @requires_gpu
