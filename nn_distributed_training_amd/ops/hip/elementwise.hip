@@ -142,6 +142,10 @@ __global__ void dsgt_y_update_k(
 // dinno.py:74-91 builds it through torch.cdist + autograd every primal
 // iteration).  mode: 0=Adam 1=AdamW 2=SGD (matching dinno.py:55-72);
 // with_penalty=false gives the plain local step (DSGD's dsgd.py:49-58).
+// first_step: treat the Adam moments as zero without reading them —
+// the per-round m/v zero-fill kernels of the non-persistent DiNNO mode
+// fold away (reference recreates the Adam optimizer each round,
+// dinno.py:57-72).
 template <typename T, int MODE, bool WITH_PENALTY>
 __global__ void fused_step_k(
     T* __restrict__ theta, const T* __restrict__ grad,
@@ -150,7 +154,7 @@ __global__ void fused_step_k(
     const int* __restrict__ deg,  // [L], null unless WITH_PENALTY
     T* __restrict__ m, T* __restrict__ v,  // Adam state (null for SGD)
     T rho, T lr, T beta1, T beta2, T eps, T wd, T bc1, T bc2,
-    long n, long L) {
+    int first_step, long n, long L) {
   const long total = L * n;
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
@@ -169,8 +173,10 @@ __global__ void fused_step_k(
     } else if (wd != T(0)) {  // Adam L2
       g += wd * th;
     }
-    const T mt = beta1 * m[t] + (T(1) - beta1) * g;
-    const T vt = beta2 * v[t] + (T(1) - beta2) * g * g;
+    const T m_prev = first_step ? T(0) : m[t];
+    const T v_prev = first_step ? T(0) : v[t];
+    const T mt = beta1 * m_prev + (T(1) - beta1) * g;
+    const T vt = beta2 * v_prev + (T(1) - beta2) * g * g;
     m[t] = mt;
     v[t] = vt;
     // bc1 = 1-beta1^t, bc2 = 1-beta2^t (host-computed per step)

@@ -148,7 +148,8 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
                 c10::optional<torch::Tensor> m,
                 c10::optional<torch::Tensor> v,
                 double rho, double lr, double beta1, double beta2,
-                double eps, double wd, long step_t, long mode) {
+                double eps, double wd, long step_t, long mode,
+                bool first_step) {
   CHECK_DEV(theta); CHECK_DEV(grad);
   const long L = theta.size(0), n = theta.size(1);
   const bool pen = dual.has_value();
@@ -169,7 +170,8 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
           m.has_value() ? m->data_ptr<scalar_t>() : nullptr,
           v.has_value() ? v->data_ptr<scalar_t>() : nullptr,
           (scalar_t)rho, (scalar_t)lr, (scalar_t)beta1,
-          (scalar_t)beta2, (scalar_t)eps, (scalar_t)wd, bc1, bc2, n, L);
+          (scalar_t)beta2, (scalar_t)eps, (scalar_t)wd, bc1, bc2,
+          first_step ? 1 : 0, n, L);
     };
     using c0 = std::integral_constant<int, 0>;
     using c1 = std::integral_constant<int, 1>;
@@ -347,14 +349,16 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
   CHECK_DEV(dY); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
   TORCH_CHECK(K <= 7, "conv_pool_bwd supports kernel size <= 7");
-  // chunk the batch so the grid fills the chip; accumulation is atomic
-  // (caller zeroes the grad stack). nchunk trades occupancy against
-  // f64-atomic contention on the 26 dw/db words per (l, f): 32 chunks
-  // serialized ~100 us/call on the MNIST bench, 8 chunks ~4x less.
-  const int nchunk = (int)std::min<long>(B, 8);
+  // images per block bounded by LDS (4 fp64 28x28 images = 25 KB);
+  // chunks spread the batch over the chip and bound atomic contention
   DISPATCH_FT(dY, {
+    const long max_imgs =
+        (120 * 1024 / sizeof(scalar_t)) / (IMG * IMG);
+    const int cb = (int)std::max<long>(1, std::min<long>(4, max_imgs));
+    const int nchunk = (int)((B + cb - 1) / cb);
+    const size_t shmem = (size_t)cb * IMG * IMG * sizeof(scalar_t);
     hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
-        dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
+        dim3(L * nchunk), dim3(256), shmem, cur_stream(),
         dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
         X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
         n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
