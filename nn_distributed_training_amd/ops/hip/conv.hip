@@ -107,9 +107,7 @@ __global__ void conv_pool_bwd_k(
   }
   __syncthreads();
 
-  __shared__ T red[4];
   const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x / WAVE;
 
   for (int f = 0; f < F; ++f) {
     T dw[KMAX * KMAX];
@@ -139,21 +137,20 @@ __global__ void conv_pool_bwd_k(
       }
     }
 
+    // barrier-free: shuffle-reduce within each wave, wave leaders add
+    // straight to global (few dozen atomics per block, spread over 26
+    // addresses — cheaper than 26 serialized block-tree reductions)
     T* wslice = gstack + (long)l * n + w_off + (long)f * K * K;
     for (int i = 0; i < K * K + 1; ++i) {
       T v = (i < K * K) ? dw[i] : db;
       v = wave_reduce_sum(v);
-      if (lane == 0) red[wid] = v;
-      __syncthreads();
-      if (threadIdx.x == 0) {
-        T tot = red[0] + red[1] + red[2] + red[3];
+      if (lane == 0) {
         if (i < K * K) {
-          atomicAdd(&wslice[i], tot);
+          atomicAdd(&wslice[i], v);
         } else {
-          atomicAdd(&gstack[(long)l * n + b_off + f], tot);
+          atomicAdd(&gstack[(long)l * n + b_off + f], v);
         }
       }
-      __syncthreads();
     }
   }
 }
