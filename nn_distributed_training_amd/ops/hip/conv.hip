@@ -72,86 +72,76 @@ __global__ void conv_pool_fwd_k(
 
 // Backward to weights/bias: route each pooled grad to its argmax conv
 // position, multiply by the image window. dY is the grad AFTER the relu
-// mask (pooled output > 0), applied by the caller (act_grad or the
-// fused dX epilogue).
+// mask (pooled output > 0), applied by the caller via act_grad.
 //
-// Grid: (l, batch-chunk); each block STAGES its CB images in LDS
-// (uncoalesced per-entry window reads from HBM were ~1/3 of the MNIST
-// round), then loops filters: per-thread private dW[K*K]+db over the
-// chunk's pooled entries, wave shuffle-reduce + LDS combine, one
-// atomicAdd per value (the caller zeroes the conv grad slice first).
+// Grid: (l, f, batch-chunk) so the chip fills (the v1 one-block-per-
+// (l,f) version was 24 blocks on 256 CUs and 41% of round time);
+// each thread accumulates a private dW[K*K]+db over its strided share
+// of the chunk, waves shuffle-reduce, wave leaders combine in LDS and
+// lane 0 atomically adds into the grad stack (the caller zeroes the
+// conv slice first).
 template <typename T, int KMAX>
 __global__ void conv_pool_bwd_k(
     const T* __restrict__ dY, const unsigned char* __restrict__ idx,
     const T* __restrict__ X, T* __restrict__ gstack,
     long n, long w_off, long b_off, int B, int F, int K, int IMG,
     int nchunk) {
-  extern __shared__ __align__(16) unsigned char smem_raw[];
-  T* imgs = reinterpret_cast<T*>(smem_raw);   // [CB * IMG * IMG]
-
   const int chunk = blockIdx.x % nchunk;
-  const int l = blockIdx.x / nchunk;
+  const int f = (blockIdx.x / nchunk) % F;
+  const int l = blockIdx.x / (nchunk * F);
   const int conv_out = IMG - (K - 1);
   const int P = conv_out / 2;
   const int npool = F * P * P;
-  const int cb = (B + nchunk - 1) / nchunk;
+  const int cb = (B + nchunk - 1) / nchunk;       // images per chunk
   const int b0 = chunk * cb;
   const int b1 = min(B, b0 + cb);
-  const int nimg = b1 - b0;
 
-  // coalesced image staging
-  const long img_elems = (long)nimg * IMG * IMG;
-  const T* Xl = X + ((long)l * B + b0) * IMG * IMG;
-  for (long t = threadIdx.x; t < img_elems; t += blockDim.x) {
-    imgs[t] = Xl[t];
+  T dw[KMAX * KMAX];
+  T db = T(0);
+  #pragma unroll
+  for (int i = 0; i < KMAX * KMAX; ++i) dw[i] = T(0);
+
+  const int work = (b1 - b0) * P * P;
+  for (int t = threadIdx.x; t < work; t += blockDim.x) {
+    const int b = b0 + t / (P * P);
+    const int py = (t / P) % P;
+    const int px = t % P;
+    const long lb = (long)l * B + b;
+    const long o = lb * npool + f * P * P + py * P + px;
+    const T g = dY[o];
+    if (g == T(0)) continue;
+    const int d = idx[o];
+    const int cy = 2 * py + (d >> 1);
+    const int cx = 2 * px + (d & 1);
+    const T* img = X + lb * IMG * IMG;
+    db += g;
+    for (int ky = 0; ky < K; ++ky) {
+      const T* row = img + (cy + ky) * IMG + cx;
+      for (int kx = 0; kx < K; ++kx) {
+        dw[ky * K + kx] += g * row[kx];
+      }
+    }
   }
-  __syncthreads();
 
+  // reduce the K*K+1 partials: shuffle within waves, LDS across waves
+  __shared__ T red[4];  // one slot per wave (256 threads = 4 waves)
   const int lane = threadIdx.x & (WAVE - 1);
-
-  for (int f = 0; f < F; ++f) {
-    T dw[KMAX * KMAX];
-    T db = T(0);
-    #pragma unroll
-    for (int i = 0; i < KMAX * KMAX; ++i) dw[i] = T(0);
-
-    const int work = nimg * P * P;
-    for (int t = threadIdx.x; t < work; t += blockDim.x) {
-      const int b = t / (P * P);
-      const int py = (t / P) % P;
-      const int px = t % P;
-      const long lb = (long)l * B + b0 + b;
-      const long o = lb * npool + f * P * P + py * P + px;
-      const T g = dY[o];
-      if (g == T(0)) continue;
-      const int d = idx[o];
-      const int cy = 2 * py + (d >> 1);
-      const int cx = 2 * px + (d & 1);
-      const T* img = imgs + (long)b * IMG * IMG;
-      db += g;
-      for (int ky = 0; ky < K; ++ky) {
-        const T* row = img + (cy + ky) * IMG + cx;
-        for (int kx = 0; kx < K; ++kx) {
-          dw[ky * K + kx] += g * row[kx];
-        }
+  const int wid = threadIdx.x / WAVE;
+  T* wslice = gstack + (long)l * n + w_off + (long)f * K * K;
+  for (int i = 0; i < K * K + 1; ++i) {
+    T v = (i < K * K) ? dw[i] : db;
+    v = wave_reduce_sum(v);
+    if (lane == 0) red[wid] = v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      T tot = red[0] + red[1] + red[2] + red[3];
+      if (i < K * K) {
+        atomicAdd(&wslice[i], tot);
+      } else {
+        atomicAdd(&gstack[(long)l * n + b_off + f], tot);
       }
     }
-
-    // barrier-free: shuffle-reduce within each wave, wave leaders add
-    // straight to global (few dozen atomics per block, spread over 26
-    // addresses — cheaper than 26 serialized block-tree reductions)
-    T* wslice = gstack + (long)l * n + w_off + (long)f * K * K;
-    for (int i = 0; i < K * K + 1; ++i) {
-      T v = (i < K * K) ? dw[i] : db;
-      v = wave_reduce_sum(v);
-      if (lane == 0) {
-        if (i < K * K) {
-          atomicAdd(&wslice[i], v);
-        } else {
-          atomicAdd(&gstack[(long)l * n + b_off + f], v);
-        }
-      }
-    }
+    __syncthreads();
   }
 }
 

@@ -349,16 +349,13 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
   CHECK_DEV(dY); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
   TORCH_CHECK(K <= 7, "conv_pool_bwd supports kernel size <= 7");
-  // images per block bounded by LDS (4 fp64 28x28 images = 25 KB);
-  // chunks spread the batch over the chip and bound atomic contention
+  // batch-chunked (l, f, chunk) grid with block-tree reduction +
+  // atomics; measured best of three structures on the MNIST bench
+  // (LDS-staged image variant was 2x slower — profiles/README.md)
+  const int nchunk = (int)std::min<long>(B, 32);
   DISPATCH_FT(dY, {
-    const long max_imgs =
-        (120 * 1024 / sizeof(scalar_t)) / (IMG * IMG);
-    const int cb = (int)std::max<long>(1, std::min<long>(4, max_imgs));
-    const int nchunk = (int)((B + cb - 1) / cb);
-    const size_t shmem = (size_t)cb * IMG * IMG * sizeof(scalar_t);
     hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
-        dim3(L * nchunk), dim3(256), shmem, cur_stream(),
+        dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
         dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
         X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
         n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
