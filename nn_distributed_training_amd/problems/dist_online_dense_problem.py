@@ -126,16 +126,20 @@ class DistOnlineDensityProblem(ProblemBase):
         if self.conf.get("save_models", False):
             if self.stacked is not None:
                 self.stacked.flush_to_models()
-            # each rank contributes its local nodes' states via rank-0
-            # gather-free path: in single-rank mode this is all nodes
+            # every rank writes its local nodes' states; rank 0 keeps
+            # the reference's file name (single-rank = all nodes there),
+            # other ranks add per-rank shards
+            state_dicts = {
+                i: self.models[i].state_dict() for i in self.local_nodes
+            }
             if self.is_root:
-                state_dicts = {
-                    i: self.models[i].state_dict() for i in self.local_nodes
-                }
                 file_name = self.conf["problem_name"] + "_models.pt"
-                torch.save(
-                    state_dicts, os.path.join(output_dir, file_name)
+            else:
+                file_name = (
+                    f"{self.conf['problem_name']}_models_rank"
+                    f"{self.comm.rank}.pt"
                 )
+            torch.save(state_dicts, os.path.join(output_dir, file_name))
 
     # ------------------------------------------------------------------
     def validate(self, i):

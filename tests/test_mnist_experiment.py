@@ -167,3 +167,67 @@ def test_dinno_learns_synthetic_mnist(tmp_path):
     peak = max(ce[1].amax() for ce in res["consensus_error"][:-1])
     last = res["consensus_error"][-1][1].amax()
     assert last < peak
+
+
+def test_scaling_driver(tmp_path):
+    """dist_mnist_scaling sweep (tiny): per-trial graphs + results +
+    rounds/sec summary (reference experiments/dist_mnist_scaling.py)."""
+    import yaml as _yaml
+
+    from nn_distributed_training_amd.experiments import dist_mnist_scaling
+
+    conf = {
+        "experiment": {
+            "name": "tiny_scaling",
+            "output_metadir": str(tmp_path / "out"),
+            "use_cuda": False,
+            "writeout": True,
+            "data_source": "synthetic",
+            "train_samples": 400,
+            "val_samples": 100,
+            "precision": "fp64",
+            "engine": "torch",
+            "seed": 0,
+            "loss": "NLL",
+            "sweep": {"type": "nodes", "num_nodes": [3, 5],
+                      "fiedler": 1.0},
+            "model": {"num_filters": 3, "kernel_size": 5,
+                      "linear_width": 64},
+        },
+        "problem_configs": {
+            "p1": {
+                "problem_name": "dinno",
+                "train_batch_size": 16,
+                "val_batch_size": 64,
+                "verbose_evals": False,
+                "metrics": ["top1_accuracy", "consensus_error"],
+                "metrics_config": {"evaluate_frequency": 100},
+                "optimizer_config": {
+                    "alg_name": "dinno", "rho_init": 0.5,
+                    "rho_scaling": 1.0003, "outer_iterations": 3,
+                    "primal_iterations": 1,
+                    "primal_optimizer": "adam",
+                    "persistant_primal_opt": False,
+                    "primal_lr_start": 0.005,
+                    "primal_lr_finish": 0.001,
+                    "lr_decay_type": "log", "profile": False,
+                },
+            }
+        },
+    }
+    pth = tmp_path / "scaling.yaml"
+    with open(pth, "w") as f:
+        _yaml.safe_dump(conf, f)
+    dist_mnist_scaling.experiment(str(pth))
+
+    runs = list((tmp_path / "out").iterdir())
+    files = {p.name for p in runs[0].iterdir()}
+    assert "0.gpickle" in files and "1.gpickle" in files
+    assert "0_dinno_results.pt" in files and "1_dinno_results.pt" in files
+    summary = torch.load(
+        runs[0] / "scaling_summary.pt", weights_only=False
+    )
+    assert {(0, "dinno"), (1, "dinno")} <= set(summary)
+    assert summary[(0, "dinno")]["N"] == 3
+    assert summary[(1, "dinno")]["N"] == 5
+    assert summary[(0, "dinno")]["rounds_per_sec"] > 0
