@@ -580,59 +580,34 @@ class StackedEngine:
             )
 
     # ------------------------------------------------------------------
-    # neighbor-table plumbing (packed remote buffers, CSR over rows)
+    # neighbor-table plumbing (delegates to parallel/schedule.py so the
+    # multi-rank plan is unit-testable on CPU with gloo)
     def remote_plan(self, width_factor=1):
-        """(remote_nodes sorted, remote_buf [R, w*n], dests {node: row})
-        for the current graph; None remote_buf when nothing is remote."""
-        if self.pr.comm.world == 1:
-            return [], None, {}
-        _, recv_nodes = self.pr.comm.edge_transfers(
-            self.pr.layout, list(self.pr.graph.edges())
+        from ..parallel import schedule
+
+        return schedule.remote_plan(
+            self.pr.comm, self.pr.layout, self.pr.graph, self.n,
+            self.device, self.dtype, width_factor,
         )
-        remote_nodes = sorted(recv_nodes)
-        if not remote_nodes:
-            return [], None, {}
-        buf = torch.empty(
-            len(remote_nodes), width_factor * self.n,
-            device=self.device, dtype=self.dtype,
-        )
-        dests = {j: buf[r] for r, j in enumerate(remote_nodes)}
-        return remote_nodes, buf, dests
 
     def build_csr(self, row_of, include_self=False, W=None):
-        offs = [0]
-        idx = []
-        wts = []
-        for i in self.local_nodes:
-            if include_self:
-                idx.append(row_of[i])
-                if W is not None:
-                    wts.append(float(W[i, i]))
-            for j in self.pr.graph.neighbors(i):
-                idx.append(row_of[j])
-                if W is not None:
-                    wts.append(float(W[i, j]))
-            offs.append(len(idx))
-        dev = self.device
-        offs_t = torch.tensor(offs, dtype=torch.int32, device=dev)
-        idx_t = torch.tensor(idx, dtype=torch.int32, device=dev)
-        w_t = (
-            torch.tensor(wts, dtype=self.dtype, device=dev)
-            if W is not None
-            else None
+        from ..parallel import schedule
+
+        return schedule.build_csr(
+            self.pr.graph, self.local_nodes, row_of, self.device,
+            self.dtype, include_self=include_self, W=W,
         )
-        return offs_t, idx_t, w_t
 
     def row_map(self, remote_nodes):
-        row_of = {i: li for li, i in enumerate(self.local_nodes)}
-        for r, j in enumerate(remote_nodes):
-            row_of[j] = self.L + r
-        return row_of
+        from ..parallel import schedule
+
+        return schedule.row_map(self.local_nodes, remote_nodes)
 
     def degrees(self) -> torch.Tensor:
-        return torch.tensor(
-            [self.pr.graph.degree(i) for i in self.local_nodes],
-            dtype=torch.int32, device=self.device,
+        from ..parallel import schedule
+
+        return schedule.degrees(
+            self.pr.graph, self.local_nodes, self.device
         )
 
 
