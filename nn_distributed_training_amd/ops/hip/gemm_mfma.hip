@@ -259,8 +259,11 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
   acc_t acc[2][2] = {};
 
   for (int k0 = mlo; k0 < mhi; k0 += BK) {
+    // o-fastest staging order: consecutive threads read consecutive
+    // dZ-row elements (k-fastest was a stride-O gather — uncoalesced,
+    // and dW was the #1 kernel of the density round)
     for (int t = tid; t < BM * BK; t += 256) {
-      const int o = t / BK, k = t % BK;
+      const int k = t / BM, o = t % BM;
       As[k][o] = (o0 + o < O && k0 + k < mhi)
                      ? Gl[(long)(k0 + k) * O + (o0 + o)]
                      : T(0);
