@@ -52,6 +52,11 @@ def parse_args():
         help="mnist = DiNNO MNIST 8-node (headline, BASELINE cfg 3/5); "
              "density = DSGT online implicit density (BASELINE cfg 4)",
     )
+    p.add_argument(
+        "--alg", default=None, choices=["dinno", "dsgd", "dsgt"],
+        help="mnist workload algorithm (default dinno; dsgd with "
+             "--nodes 4 = BASELINE cfg 2)",
+    )
     return p.parse_args()
 
 
@@ -172,13 +177,14 @@ def _build_mnist(args, device, use_hip):
         split_train_set,
     )
     from nn_distributed_training_amd.models import MNISTConvNet
-    from nn_distributed_training_amd.optimizers.dinno import DiNNO
+    from nn_distributed_training_amd.optimizers import build_optimizer
     from nn_distributed_training_amd.problems.dist_mnist_problem import (
         DistMNISTProblem,
     )
 
     N = args.nodes
     B = args.batch or 64
+    alg = args.alg or "dinno"
     graph = _make_graph(args, N)
     train = SyntheticMNIST(args.samples_per_node * N, seed=0)
     val = SyntheticMNIST(1024, seed=1)
@@ -186,21 +192,28 @@ def _build_mnist(args, device, use_hip):
                               "hetero_sorted")
     base_model = MNISTConvNet(3, 5, 64)
 
-    opt_conf = {
-        "alg_name": "dinno",
-        "rho_init": 0.5,
-        "rho_scaling": 1.0003,
-        "outer_iterations": args.warmup + args.steps,
-        "primal_iterations": 2,
-        "primal_optimizer": "adam",
-        "persistant_primal_opt": False,
-        "primal_lr_start": 0.005,
-        "primal_lr_finish": 0.0005,
-        "lr_decay_type": "log",
-        "profile": False,
+    oits = args.warmup + args.steps
+    opt_confs = {
+        "dinno": {
+            "alg_name": "dinno", "rho_init": 0.5,
+            "rho_scaling": 1.0003, "outer_iterations": oits,
+            "primal_iterations": 2, "primal_optimizer": "adam",
+            "persistant_primal_opt": False, "primal_lr_start": 0.005,
+            "primal_lr_finish": 0.0005, "lr_decay_type": "log",
+            "profile": False,
+        },
+        "dsgd": {
+            "alg_name": "dsgd", "outer_iterations": oits,
+            "alpha0": 0.005, "mu": 0.001, "profile": False,
+        },
+        "dsgt": {
+            "alg_name": "dsgt", "outer_iterations": oits,
+            "alpha": 0.005, "init_grads": True, "profile": False,
+        },
     }
+    opt_conf = opt_confs[alg]
     prob_conf = {
-        "problem_name": "bench_dinno",
+        "problem_name": f"bench_{alg}",
         "train_batch_size": B,
         "val_batch_size": 256,
         "data_seed": 0,
@@ -217,24 +230,80 @@ def _build_mnist(args, device, use_hip):
     if use_hip:
         from nn_distributed_training_amd.ops.stacked import (
             DiNNOStackedDriver,
+            DSGDStackedDriver,
+            DSGTStackedDriver,
             StackedEngine,
         )
 
         pr.stacked = StackedEngine(pr)
-        driver = DiNNOStackedDriver(DiNNO(pr, device, opt_conf), pr)
+        dopt = build_optimizer(pr, device, opt_conf)
+        drv_cls = {
+            "dinno": DiNNOStackedDriver,
+            "dsgd": DSGDStackedDriver,
+            "dsgt": DSGTStackedDriver,
+        }[alg]
+        driver = drv_cls(dopt, pr)
         driver.prepare()
         step_fn = driver.step_round
     else:
-        step_fn = _golden_round_fn(DiNNO(pr, device, opt_conf), pr)
+        dopt = build_optimizer(pr, device, opt_conf)
+        if alg == "dinno":
+            step_fn = _golden_round_fn(dopt, pr)
+        elif alg == "dsgt":
+            state = {"ready": False}
+
+            def step_fn(k, _d=dopt):
+                if not state["ready"]:
+                    if _d.conf["init_grads"]:
+                        for i in pr.local_nodes:
+                            g = _d._local_grad_vector(i)
+                            _d.y[i] = g.clone()
+                            _d.g[i] = g.clone()
+                    state["ready"] = True
+                _golden_dsgt_round(_d, pr)
+        else:
+            state = {"alph": dopt.alph0}
+
+            def step_fn(k, _d=dopt):
+                _golden_dsgd_round(_d, pr, state)
     cfg = {
         "model": "MNISTConvNet(3,5,64) n=28440",
-        "alg": "dinno",
+        "alg": alg,
         "nodes": N,
         "graph": args.graph,
         "global_batch": B * N,
-        "primal_iterations": 2,
+        "primal_iterations": 2 if alg == "dinno" else None,
     }
     return pr, step_fn, cfg
+
+
+def _golden_dsgd_round(opt, pr, state):
+    import torch as _t
+
+    from nn_distributed_training_amd.optimizers.neighbors import (
+        gather_neighbor_stacks,
+    )
+    from nn_distributed_training_amd.utils import graph_generation
+
+    pr.update_graph()
+    W = graph_generation.get_metropolis(pr.graph).to(pr.device)
+    state["alph"] = state["alph"] * (1 - opt.mu * state["alph"])
+    ths = pr.local_params_stack()
+    neigh = gather_neighbor_stacks(pr, ths)
+    for li, i in enumerate(pr.local_nodes):
+        mixed = W[i, i] * ths[li]
+        for row, j in zip(neigh[i], pr.graph.neighbors(i)):
+            mixed = mixed + W[i, j] * row
+        _t.nn.utils.vector_to_parameters(
+            mixed, pr.models[i].parameters()
+        )
+    for i in pr.local_nodes:
+        bloss = pr.local_batch_loss(i)
+        bloss.backward()
+        with _t.no_grad():
+            for p in pr.models[i].parameters():
+                p.add_(p.grad, alpha=-state["alph"])
+                p.grad.zero_()
 
 
 def _build_density(args, device, use_hip):
