@@ -184,6 +184,74 @@ __global__ void fused_step_k(
   }
 }
 
+// Graph-capture variants: the round scalars (rho, lr, per-pit Adam
+// bias corrections) live in a device buffer `sched` laid out as
+// [rho, lr, bc1_pit0, bc2_pit0, bc1_pit1, ...] and batch offsets in an
+// int64 buffer — the host updates them with one async pinned copy per
+// round and replays a captured hipGraph with CONSTANT kernel args.
+template <typename T>
+__global__ void dinno_dual_threg_sched_k(
+    const T* __restrict__ local, const T* __restrict__ remote,
+    const int* __restrict__ offs, const int* __restrict__ idx,
+    T* __restrict__ duals, T* __restrict__ s_out,
+    const T* __restrict__ sched, long n, long L) {
+  const T rho = sched[0];
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
+    const int k0 = offs[l], k1 = offs[l + 1];
+    T S = T(0);
+    for (int k = k0; k < k1; ++k) {
+      S += table_row(local, remote, L, n, idx[k])[e];
+    }
+    const T th = local[l * n + e];
+    const T deg = T(k1 - k0);
+    duals[t] += rho * (deg * th - S);
+    s_out[t] = (deg * th + S) * T(0.5);
+  }
+}
+
+template <typename T, int MODE, bool WITH_PENALTY>
+__global__ void fused_step_sched_k(
+    T* __restrict__ theta, const T* __restrict__ grad,
+    const T* __restrict__ dual, const T* __restrict__ s,
+    const int* __restrict__ deg, T* __restrict__ m, T* __restrict__ v,
+    const T* __restrict__ sched, int pit,
+    T beta1, T beta2, T eps, T wd, int first_step, long n, long L) {
+  const T rho = sched[0];
+  const T lr = sched[1];
+  const T bc1 = sched[2 + 2 * pit];
+  const T bc2 = sched[3 + 2 * pit];
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    T th = theta[t];
+    T g = grad[t];
+    if (WITH_PENALTY) {
+      const long l = t / n;
+      g += dual[t] + T(2) * rho * (T(deg[l]) * th - s[t]);
+    }
+    if (MODE == 2) {
+      theta[t] = th - lr * g;
+      continue;
+    }
+    if (MODE == 1) {
+      th -= lr * wd * th;
+    } else if (wd != T(0)) {
+      g += wd * th;
+    }
+    const T m_prev = first_step ? T(0) : m[t];
+    const T v_prev = first_step ? T(0) : v[t];
+    const T mt = beta1 * m_prev + (T(1) - beta1) * g;
+    const T vt = beta2 * v_prev + (T(1) - beta2) * g * g;
+    m[t] = mt;
+    v[t] = vt;
+    theta[t] = th - lr * (mt / bc1) / (::sqrt(vt / bc2) + eps);
+  }
+}
+
 // Batch assembly: out[l*B + b, :] = X_all[l, idx[l, b], :] in one
 // launch (replaces torch arange + advanced indexing + copies). idx rows
 // may be strided views into a longer per-node index stream.
@@ -211,6 +279,39 @@ __global__ void gather_targets_k(
     const T* __restrict__ Y_all, const long* __restrict__ idx,
     T* __restrict__ out, long maxlen, long B, long idx_stride,
     long idx_off, long total) {
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / B;
+    const long b = t - l * B;
+    out[t] = Y_all[l * maxlen + idx[l * idx_stride + idx_off + b]];
+  }
+}
+
+// Gather with the batch offset read from a device buffer (slot `pit`
+// of offs_dev) — hipGraph-replayable.
+template <typename T>
+__global__ void gather_batch_dev_k(
+    const T* __restrict__ X_all, const long* __restrict__ idx,
+    T* __restrict__ out, const long* __restrict__ offs_dev, int pit,
+    long maxlen, long Fdim, long B, long idx_stride, long total) {
+  const long idx_off = offs_dev[pit];
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long f = t % Fdim;
+    const long lb = t / Fdim;
+    const long l = lb / B;
+    const long b = lb - l * B;
+    const long src = idx[l * idx_stride + idx_off + b];
+    out[t] = X_all[(l * maxlen + src) * Fdim + f];
+  }
+}
+
+template <typename T>
+__global__ void gather_targets_dev_k(
+    const T* __restrict__ Y_all, const long* __restrict__ idx,
+    T* __restrict__ out, const long* __restrict__ offs_dev, int pit,
+    long maxlen, long B, long idx_stride, long total) {
+  const long idx_off = offs_dev[pit];
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
     const long l = t / B;

@@ -185,6 +185,108 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
   HIP_CHECK_LAST();
 }
 
+void dinno_dual_threg_sched(torch::Tensor local,
+                            c10::optional<torch::Tensor> remote,
+                            torch::Tensor offs, torch::Tensor idx,
+                            torch::Tensor duals, torch::Tensor s_out,
+                            torch::Tensor sched) {
+  CHECK_DEV(local); CHECK_DEV(duals); CHECK_DEV(s_out);
+  const long L = duals.size(0), n = duals.size(1);
+  DISPATCH_FT(local, {
+    hipLaunchKernelGGL(ew::dinno_dual_threg_sched_k<scalar_t>,
+        dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+        local.data_ptr<scalar_t>(),
+        remote.has_value() ? remote->data_ptr<scalar_t>() : nullptr,
+        offs.data_ptr<int>(), idx.data_ptr<int>(),
+        duals.data_ptr<scalar_t>(), s_out.data_ptr<scalar_t>(),
+        sched.data_ptr<scalar_t>(), n, L);
+  });
+  HIP_CHECK_LAST();
+}
+
+void fused_step_sched(torch::Tensor theta, torch::Tensor grad,
+                      c10::optional<torch::Tensor> dual,
+                      c10::optional<torch::Tensor> s,
+                      c10::optional<torch::Tensor> deg,
+                      c10::optional<torch::Tensor> m,
+                      c10::optional<torch::Tensor> v,
+                      torch::Tensor sched, long pit,
+                      double beta1, double beta2, double eps, double wd,
+                      long mode, bool first_step) {
+  CHECK_DEV(theta); CHECK_DEV(grad);
+  const long L = theta.size(0), n = theta.size(1);
+  const bool pen = dual.has_value();
+  DISPATCH_FT(theta, {
+    auto launch = [&](auto mode_c, auto pen_c) {
+      hipLaunchKernelGGL(
+          (ew::fused_step_sched_k<scalar_t, decltype(mode_c)::value,
+                                  decltype(pen_c)::value>),
+          dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+          theta.data_ptr<scalar_t>(), grad.data_ptr<scalar_t>(),
+          pen ? dual->data_ptr<scalar_t>() : nullptr,
+          pen ? s->data_ptr<scalar_t>() : nullptr,
+          pen ? deg->data_ptr<int>() : nullptr,
+          m.has_value() ? m->data_ptr<scalar_t>() : nullptr,
+          v.has_value() ? v->data_ptr<scalar_t>() : nullptr,
+          sched.data_ptr<scalar_t>(), (int)pit,
+          (scalar_t)beta1, (scalar_t)beta2, (scalar_t)eps,
+          (scalar_t)wd, first_step ? 1 : 0, n, L);
+    };
+    using c0 = std::integral_constant<int, 0>;
+    using c1 = std::integral_constant<int, 1>;
+    using c2 = std::integral_constant<int, 2>;
+    using bt = std::integral_constant<bool, true>;
+    using bf = std::integral_constant<bool, false>;
+    if (mode == 0) { pen ? launch(c0{}, bt{}) : launch(c0{}, bf{}); }
+    else if (mode == 1) { pen ? launch(c1{}, bt{}) : launch(c1{}, bf{}); }
+    else { pen ? launch(c2{}, bt{}) : launch(c2{}, bf{}); }
+  });
+  HIP_CHECK_LAST();
+}
+
+void gather_batch_dev(torch::Tensor X_all, torch::Tensor idx,
+                      torch::Tensor out, torch::Tensor offs_dev,
+                      long pit, long idx_stride) {
+  CHECK_DEV(X_all); CHECK_DEV(out);
+  const long L = X_all.size(0), maxlen = X_all.size(1),
+             Fdim = X_all.size(2);
+  const long B = out.size(0) / L;
+  const long total = out.numel();
+  DISPATCH_FT(X_all, {
+    hipLaunchKernelGGL(ew::gather_batch_dev_k<scalar_t>,
+        dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+        X_all.data_ptr<scalar_t>(), idx.data_ptr<long>(),
+        out.data_ptr<scalar_t>(), offs_dev.data_ptr<long>(), (int)pit,
+        maxlen, Fdim, B, idx_stride, total);
+  });
+  HIP_CHECK_LAST();
+}
+
+void gather_targets_dev(torch::Tensor Y_all, torch::Tensor idx,
+                        torch::Tensor out, torch::Tensor offs_dev,
+                        long pit, long idx_stride) {
+  CHECK_DEV(Y_all); CHECK_DEV(out);
+  const long L = Y_all.size(0), maxlen = Y_all.size(1);
+  const long B = out.numel() / L;
+  const long total = out.numel();
+  if (Y_all.scalar_type() == torch::kLong) {
+    hipLaunchKernelGGL(ew::gather_targets_dev_k<long>,
+        dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+        Y_all.data_ptr<long>(), idx.data_ptr<long>(),
+        out.data_ptr<long>(), offs_dev.data_ptr<long>(), (int)pit,
+        maxlen, B, idx_stride, total);
+  } else {
+    DISPATCH_FT(Y_all, {
+      hipLaunchKernelGGL(ew::gather_targets_dev_k<scalar_t>,
+          dim3(grid_1d(total)), dim3(ew::BLOCK), 0, cur_stream(),
+          Y_all.data_ptr<scalar_t>(), idx.data_ptr<long>(),
+          out.data_ptr<scalar_t>(), offs_dev.data_ptr<long>(),
+          (int)pit, maxlen, B, idx_stride, total);
+    });
+  }
+  HIP_CHECK_LAST();
+}
+
 void axpy(torch::Tensor x, torch::Tensor g, double alpha) {
   CHECK_DEV(x);
   DISPATCH_FT(x, {
@@ -440,6 +542,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("fused_step", &fused_step);
   mod.def("axpy", &axpy);
   mod.def("gather_batch", &gather_batch);
+  mod.def("gather_batch_dev", &gather_batch_dev);
+  mod.def("gather_targets_dev", &gather_targets_dev);
+  mod.def("dinno_dual_threg_sched", &dinno_dual_threg_sched);
+  mod.def("fused_step_sched", &fused_step_sched);
   mod.def("gather_targets", &gather_targets);
   mod.def("linear_fwd", &linear_fwd);
   mod.def("act_grad", &act_grad);

@@ -140,9 +140,9 @@ class _StreamSampler:
         self.stream.copy_(st, non_blocking=True)
         self.pos = 0
 
-    def next_ref(self):
-        """(stream tensor, row stride, offset) for the next batch —
-        no per-draw tensor slicing."""
+    def next_offset(self) -> int:
+        """Advance one batch; refill + epoch bookkeeping; returns the
+        stream column offset of the batch."""
         if self.pos + self.B > self.S:
             self._refill()
         off = self.pos
@@ -154,6 +154,12 @@ class _StreamSampler:
                 for _ in range(ep - self.reported_epochs[li]):
                     self.epoch_cb(li)
                 self.reported_epochs[li] = ep
+        return off
+
+    def next_ref(self):
+        """(stream tensor, row stride, offset) for the next batch —
+        no per-draw tensor slicing."""
+        off = self.next_offset()
         return self.stream, self.S, off
 
 
@@ -655,6 +661,35 @@ class DiNNOStackedDriver:
         self.step_t = 0
         self._plan_key = None
         self._plan = None
+        # hipGraph capture of the whole round (NDTA_GRAPHS=0 disables):
+        # round scalars travel through a device sched buffer so the
+        # captured launches have constant arguments
+        self._graph = None
+        self._graph_warm = 0
+        self._graph_failed = False
+        self.graph_mode = (
+            eng.device.type == "cuda"
+            and os.environ.get("NDTA_GRAPHS", "1") == "1"
+            and _graph_is_static(self.pr)
+            and isinstance(eng.sampler, _StreamSampler)
+            and not bool(getattr(self.pr, "track_tloss", False))
+        )
+        if self.graph_mode:
+            pits = self.pits
+            dt = eng.dtype
+            self._sched_dev = torch.zeros(
+                2 + 2 * pits, device=eng.device, dtype=dt
+            )
+            self._sched_host = torch.zeros(2 + 2 * pits, dtype=dt)
+            self._offs_dev = torch.zeros(
+                pits, device=eng.device, dtype=torch.long
+            )
+            self._offs_host = torch.zeros(pits, dtype=torch.long)
+            try:
+                self._sched_host = self._sched_host.pin_memory()
+                self._offs_host = self._offs_host.pin_memory()
+            except RuntimeError:
+                pass
 
     def _round_plan(self):
         key = _edge_key(self.pr)
@@ -669,9 +704,103 @@ class DiNNOStackedDriver:
         self._plan_key = key
         return self._plan
 
+    def _graph_body(self, rbuf, offs, idx, deg):
+        """One DiNNO round with all round-varying scalars read from
+        device buffers — every launch has constant args, so the whole
+        sequence is hipGraph-capturable."""
+        eng = self.eng
+        ext = eng.ext
+        bufs = eng._bufs
+        ext.dinno_dual_threg_sched(
+            eng.theta, rbuf, offs, idx, self.duals, self.s,
+            self._sched_dev,
+        )
+        xb, yb = bufs["xb"], bufs["yb"]
+        for pit in range(self.pits):
+            ext.gather_batch_dev(
+                eng.X_all, eng.sampler.stream, xb, self._offs_dev,
+                pit, eng.sampler.S,
+            )
+            ext.gather_targets_dev(
+                eng.Y_all, eng.sampler.stream, yb, self._offs_dev,
+                pit, eng.sampler.S,
+            )
+            eng.forward(xb)
+            eng.backward(xb, yb)
+            first = (not self.persistent) and pit == 0
+            ext.fused_step_sched(
+                eng.theta, eng.grad, self.duals, self.s, deg,
+                None if self.mode == 2 else self.m,
+                None if self.mode == 2 else self.v,
+                self._sched_dev, pit, 0.9, 0.999, 1e-8, self.wd,
+                self.mode, first,
+            )
+
+    def _step_round_graph(self, k):
+        import math as _math
+
+        opt, pr, eng = self.opt, self.pr, self.eng
+        self.rho *= opt.rho_scaling
+        pr.update_graph()
+        rbuf, dests, offs, idx, deg = self._round_plan()
+        if pr.comm.world > 1:
+            pr.comm.exchange_rows(
+                pr.layout, list(pr.graph.edges()), [eng.theta], [dests]
+            )
+        if eng._bufs is None:
+            eng._bufs = eng._alloc_bufs()
+
+        # host-side schedule for this round
+        lr = float(
+            opt.primal_lr[0] if self.persistent else opt.primal_lr[k]
+        )
+        sh = self._sched_host
+        sh[0] = self.rho
+        sh[1] = lr
+        for pit in range(self.pits):
+            if self.persistent:
+                self.step_t += 1
+                t = self.step_t
+            else:
+                t = pit + 1
+            sh[2 + 2 * pit] = 1.0 - _math.pow(0.9, t)
+            sh[3 + 2 * pit] = 1.0 - _math.pow(0.999, t)
+            self._offs_host[pit] = eng.sampler.next_offset()
+        if eng._has_node0:
+            pr.forward_cnt += eng.B * self.pits
+        self._sched_dev.copy_(self._sched_host, non_blocking=True)
+        self._offs_dev.copy_(self._offs_host, non_blocking=True)
+
+        if self._graph is not None:
+            self._graph.replay()
+            return
+        if self._graph_warm < 2:
+            # warmup executions on a side stream (these ARE real rounds)
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._graph_body(rbuf, offs, idx, deg)
+            torch.cuda.current_stream().wait_stream(side)
+            self._graph_warm += 1
+            return
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._graph_body(rbuf, offs, idx, deg)
+            self._graph = g
+            self._graph.replay()  # this round's work
+        except Exception as e:  # noqa: BLE001
+            print(f"[ndta] hipGraph capture failed ({e}); "
+                  "falling back to eager rounds", flush=True)
+            self.graph_mode = False
+            self._graph_failed = True
+            self._graph_body(rbuf, offs, idx, deg)
+
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
         ext = eng.ext
+        if self.graph_mode:
+            return self._step_round_graph(k)
 
         self.rho *= opt.rho_scaling
         with _timer("update_graph"):
