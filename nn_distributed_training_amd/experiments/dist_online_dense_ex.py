@@ -9,7 +9,6 @@ optimizers, solo baseline, seeded reproducibility.
 
 from __future__ import annotations
 
-import copy
 import os
 import sys
 
