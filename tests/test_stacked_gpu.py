@@ -114,9 +114,15 @@ def test_stacked_dinno_hipgraph_matches_golden(monkeypatch):
 
 
 @requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
 @pytest.mark.parametrize("alg", ["dinno", "dsgd", "dsgt"])
-def test_stacked_matches_golden(alg):
-    torch.set_default_dtype(torch.float64)
+def test_stacked_matches_golden(alg, dtype):
+    torch.set_default_dtype(dtype)
+    tol = (
+        dict(rtol=1e-8, atol=1e-8)
+        if dtype == torch.float64
+        else dict(rtol=5e-3, atol=5e-4)
+    )
     conf = _conf(copy.deepcopy(ALG_CONFS[alg]))
 
     # golden eager torch engine
@@ -132,7 +138,7 @@ def test_stacked_matches_golden(alg):
     opt_s.train()
     stacked = pr_s.stacked.theta
 
-    torch.testing.assert_close(stacked, golden, rtol=1e-8, atol=1e-8)
+    torch.testing.assert_close(stacked, golden, **tol)
 
 
 @requires_gpu
