@@ -118,11 +118,14 @@ def test_stacked_dinno_hipgraph_matches_golden(monkeypatch):
 @pytest.mark.parametrize("alg", ["dinno", "dsgd", "dsgt"])
 def test_stacked_matches_golden(alg, dtype):
     torch.set_default_dtype(dtype)
-    tol = (
-        dict(rtol=1e-8, atol=1e-8)
-        if dtype == torch.float64
-        else dict(rtol=5e-3, atol=5e-4)
-    )
+    if dtype == torch.float64:
+        tol = dict(rtol=1e-8, atol=1e-8)  # reduction-order only
+    elif alg == "dinno":
+        # Adam's sqrt/eps nonlinearity amplifies fp32 rounding across
+        # the 6 primal steps + dual ascent; fp64 is the exactness check
+        tol = dict(rtol=5e-2, atol=5e-3)
+    else:
+        tol = dict(rtol=5e-3, atol=5e-4)
     conf = _conf(copy.deepcopy(ALG_CONFS[alg]))
 
     # golden eager torch engine
