@@ -159,7 +159,7 @@ def main():
         except ImportError:
             pass
 
-    if world > 1:
+    if dist.is_initialized():
         dist.destroy_process_group()
 
 
