@@ -493,6 +493,28 @@ void nll_bwd(torch::Tensor logp, torch::Tensor y, torch::Tensor dZ,
   HIP_CHECK_LAST();
 }
 
+void nll_fused(torch::Tensor Z, torch::Tensor Y_all,
+               torch::Tensor idx, torch::Tensor dZ,
+               c10::optional<torch::Tensor> loss,
+               c10::optional<torch::Tensor> offs_dev, long pit,
+               long idx_stride, long idx_off, long C, long B,
+               double loss_scale) {
+  CHECK_DEV(Z); CHECK_DEV(dZ);
+  const long maxlen = Y_all.size(1);
+  const long M = Z.numel() / C;
+  DISPATCH_FT(Z, {
+    hipLaunchKernelGGL(losses::nll_fused_k<scalar_t>,
+        dim3(grid_1d(M)), dim3(256), 0, cur_stream(),
+        Z.data_ptr<scalar_t>(), Y_all.data_ptr<long>(),
+        idx.data_ptr<long>(), dZ.data_ptr<scalar_t>(),
+        loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr,
+        offs_dev.has_value() ? offs_dev->data_ptr<long>() : nullptr,
+        (int)pit, maxlen, idx_stride, idx_off, M, (int)C, (int)B,
+        (scalar_t)loss_scale);
+  });
+  HIP_CHECK_LAST();
+}
+
 void bce_bwd(torch::Tensor p, torch::Tensor tgt, torch::Tensor dZ,
              c10::optional<torch::Tensor> loss, long B,
              double loss_scale) {
@@ -555,6 +577,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("conv_pool_bwd", &conv_pool_bwd);
   mod.def("logsoftmax", &logsoftmax);
   mod.def("nll_bwd", &nll_bwd);
+  mod.def("nll_fused", &nll_fused);
   mod.def("bce_bwd", &bce_bwd);
   mod.def("regression_bwd", &regression_bwd);
 }

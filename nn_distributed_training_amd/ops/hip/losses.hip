@@ -94,4 +94,43 @@ __global__ void regression_bwd_k(
   }
 }
 
+// Fully fused classification backward: from the LOGITS, compute the
+// row log-sum-exp, gather the target label from the resident dataset
+// (via the index stream + offset — no separate target-gather kernel,
+// no materialized log-probs in the training path), and write
+// dZ = (softmax - onehot) * loss_scale / B. Offset comes from either
+// the literal idx_off or slot `pit` of offs_dev (hipGraph mode).
+template <typename T>
+__global__ void nll_fused_k(
+    const T* __restrict__ Z, const long* __restrict__ Y_all,
+    const long* __restrict__ idx, T* __restrict__ dZ,
+    T* __restrict__ loss,  // may be null, [L]
+    const long* __restrict__ offs_dev, int pit,
+    long maxlen, long idx_stride, long idx_off,
+    long M, int C, int B, T loss_scale) {
+  const T w = loss_scale / T(B);
+  const long off = offs_dev ? offs_dev[pit] : idx_off;
+  for (long m = blockIdx.x * 256L + threadIdx.x; m < M;
+       m += (long)gridDim.x * 256L) {
+    const long l = m / B;
+    const long b = m - l * B;
+    const long src = idx[l * idx_stride + off + b];
+    const int t = (int)Y_all[l * maxlen + src];
+
+    const T* z = Z + m * C;
+    T* g = dZ + m * C;
+    T mx = z[0];
+    for (int c = 1; c < C; ++c) mx = z[c] > mx ? z[c] : mx;
+    T sum = T(0);
+    for (int c = 0; c < C; ++c) sum += ::exp(z[c] - mx);
+    const T lse = mx + ::log(sum);
+    for (int c = 0; c < C; ++c) {
+      g[c] = (::exp(z[c] - lse) - (c == t ? T(1) : T(0))) * w;
+    }
+    if (loss != nullptr) {
+      atomicAdd(&loss[l], -(z[t] - lse) / T(B));
+    }
+  }
+}
+
 }  // namespace losses

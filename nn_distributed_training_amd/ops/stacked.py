@@ -364,13 +364,18 @@ class StackedEngine:
         idx, stride, off = self.sampler.next_ref()
         xb, yb = self._bufs["xb"], self._bufs["yb"]
         self.ext.gather_batch(self.X_all, idx, xb, stride, off)
-        self.ext.gather_targets(self.Y_all, idx, yb, stride, off)
+        self._last_off = off
+        if self.classification:
+            # targets stay resident: the fused NLL kernel gathers them
+            yb = None
+        else:
+            self.ext.gather_targets(self.Y_all, idx, yb, stride, off)
         if self._has_node0:
             self.pr.forward_cnt += self.B
         return xb, yb
 
     # ------------------------------------------------------------------
-    def forward(self, xb, bufs=None):
+    def forward(self, xb, bufs=None, train_skip_logp=False):
         if self._bufs is None:
             self._bufs = self._alloc_bufs()
         bufs = bufs if bufs is not None else self._bufs
@@ -393,7 +398,11 @@ class StackedEngine:
                         layer.b_off, M, layer.in_dim, layer.out_dim,
                         ACT_IDS["none"], 1.0,
                     )
-                    ext.logsoftmax(out, bufs["logp"], layer.out_dim)
+                    # training backward consumes the LOGITS through the
+                    # fused NLL kernel; log-probs are only materialized
+                    # for evaluation
+                    if not train_skip_logp:
+                        ext.logsoftmax(out, bufs["logp"], layer.out_dim)
                 else:
                     ext.linear_fwd(
                         cur, self.theta, out, bufs["zs"][li],
@@ -404,7 +413,8 @@ class StackedEngine:
         return cur
 
     # ------------------------------------------------------------------
-    def backward(self, xb, yb, loss_scale=1.0, want_loss=False):
+    def backward(self, xb, yb, loss_scale=1.0, want_loss=False,
+                 graph_offs=None, pit=0):
         bufs = self._bufs
         ext = self.ext
         layers = self.spec.layers
@@ -439,9 +449,20 @@ class StackedEngine:
         last = layers[-1]
         dz = bufs["dzs"][nl - 1]
         if self.classification:
-            ext.nll_bwd(
-                bufs["logp"], yb, dz, loss_buf, last.out_dim, M,
-                loss_scale,
+            # fused: logits -> row LSE -> resident-target gather -> dZ
+            idx_t = (
+                self.sampler.stream
+                if hasattr(self.sampler, "stream")
+                else self.sampler.buf
+            )
+            stride = (
+                self.sampler.S if hasattr(self.sampler, "S") else self.B
+            )
+            ext.nll_fused(
+                bufs["acts"][nl - 1], self.Y_all, idx_t, dz, loss_buf,
+                graph_offs, pit, stride,
+                0 if graph_offs is not None else self._last_off,
+                last.out_dim, M, loss_scale,
             )
         else:
             kind = self._loss_kind
@@ -720,18 +741,16 @@ class DiNNOStackedDriver:
             eng.theta, rbuf, offs, idx, self.duals, self.s,
             self._sched_dev,
         )
-        xb, yb = bufs["xb"], bufs["yb"]
+        xb = bufs["xb"]
         for pit in range(self.pits):
             ext.gather_batch_dev(
                 eng.X_all, eng.sampler.stream, xb, self._offs_dev,
                 pit, eng.sampler.S,
             )
-            ext.gather_targets_dev(
-                eng.Y_all, eng.sampler.stream, yb, self._offs_dev,
-                pit, eng.sampler.S,
+            eng.forward(xb, train_skip_logp=True)
+            eng.backward(
+                xb, None, graph_offs=self._offs_dev, pit=pit
             )
-            eng.forward(xb)
-            eng.backward(xb, yb)
             first = (not self.persistent) and pit == 0
             ext.fused_step_sched(
                 eng.theta, eng.grad, self.duals, self.s, deg,
@@ -835,7 +854,7 @@ class DiNNOStackedDriver:
             with _timer("next_batch"):
                 xb, yb = eng.next_batch()
             with _timer("forward"):
-                eng.forward(xb)
+                eng.forward(xb, train_skip_logp=True)
             with _timer("backward"):
                 wl = want_tl and pi == self.pits - 1
                 lb = eng.backward(xb, yb, want_loss=wl)
@@ -912,7 +931,7 @@ class DSGDStackedDriver:
         eng.theta, self.theta_next = self.theta_next, eng.theta
 
         xb, yb = eng.next_batch()
-        eng.forward(xb)
+        eng.forward(xb, train_skip_logp=True)
         want_tl = bool(getattr(pr, "track_tloss", False))
         lb = eng.backward(xb, yb, want_loss=want_tl)
         if want_tl:
@@ -951,7 +970,7 @@ class DSGTStackedDriver:
         self._plan = None
         if self.opt.conf["init_grads"]:
             xb, yb = eng.next_batch()
-            eng.forward(xb)
+            eng.forward(xb, train_skip_logp=True)
             eng.backward(xb, yb)
             self.y.copy_(eng.grad)
             self.g.copy_(eng.grad)
@@ -1000,7 +1019,7 @@ class DSGTStackedDriver:
         with _timer("next_batch"):
             xb, yb = eng.next_batch()
         with _timer("forward"):
-            eng.forward(xb)
+            eng.forward(xb, train_skip_logp=True)
         with _timer("backward"):
             want_tl = bool(getattr(pr, "track_tloss", False))
             lb = eng.backward(xb, yb, want_loss=want_tl)
