@@ -212,6 +212,43 @@ def test_logsoftmax_nll(ext, dtype):
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+def test_nll_fused(ext, dtype):
+    """Fused logits->LSE->resident-target-gather->dZ kernel vs torch."""
+    torch.manual_seed(14)
+    L, B, C, maxlen, S = 2, 16, 10, 40, 64
+    M = L * B
+    dev = _dev()
+    Z = torch.randn(M, C, dtype=dtype, device=dev)
+    Y_all = torch.randint(0, C, (L, maxlen), device=dev)
+    stream = torch.randint(0, maxlen, (L, S), device=dev)
+    off = 7
+    dZ = torch.empty_like(Z)
+    loss = torch.zeros(L, dtype=dtype, device=dev)
+    ext.nll_fused(Z, Y_all, stream, dZ, loss, None, 0, S, off, C, B,
+                  1.0)
+
+    ar = torch.arange(L, device=dev).unsqueeze(1)
+    y = Y_all[ar, stream[:, off : off + B]].reshape(-1)
+    for l in range(L):
+        zl = Z[l * B : (l + 1) * B].detach().requires_grad_()
+        ref = torch.nn.functional.nll_loss(
+            torch.log_softmax(zl, dim=1), y[l * B : (l + 1) * B]
+        )
+        ref.backward()
+        torch.testing.assert_close(dZ[l * B : (l + 1) * B], zl.grad,
+                                   **TOL[dtype])
+        torch.testing.assert_close(loss[l], ref.detach(), **TOL[dtype])
+
+    # device-offset form (hipGraph mode) must agree
+    offs_dev = torch.tensor([off], device=dev)
+    dZ2 = torch.empty_like(Z)
+    ext.nll_fused(Z, Y_all, stream, dZ2, None, offs_dev, 0, S, 0, C, B,
+                  1.0)
+    torch.testing.assert_close(dZ2, dZ, rtol=0, atol=0)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
 def test_bce_bwd(ext, dtype):
     torch.manual_seed(5)
     L, B = 2, 32
