@@ -398,12 +398,22 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
           dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
           gstack.data_ptr<scalar_t>(), n, w_off, b_off,
           (int)M, (int)I, (int)O, (int)nchunk);
-      long nb = std::max<long>(1, 1024 / std::max<long>(1, L));
-      nb = std::min<long>(nb, (M + 255) / 256);
-      hipLaunchKernelGGL(gmfma::bias_grad_chunked_k<scalar_t>,
-          dim3((O + 255) / 256, 1, L * nb), dim3(256), 0, cur_stream(),
-          dZ.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
-          n, b_off, (int)M, (int)O, (int)nb);
+    } else if (M > 2048 && I <= 4 && O <= 256) {
+      long nchunk = std::max<long>(1, 1024 / std::max<long>(1, L));
+      nchunk = std::min<long>(nchunk, (M + 255) / 256);
+      hipLaunchKernelGGL((gemm::dw_skinny_i_k<scalar_t, 4>),
+          dim3(1, 1, L * nchunk), dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+          gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)nchunk);
+    } else if (M > 2048 && O <= 4 && I <= 256) {
+      long nchunk = std::max<long>(1, 1024 / std::max<long>(1, L));
+      nchunk = std::min<long>(nchunk, (M + 255) / 256);
+      hipLaunchKernelGGL((gemm::dw_skinny_o_k<scalar_t, 4>),
+          dim3(1, 1, L * nchunk), dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+          gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)nchunk);
     } else if (M > 2048) {
       const long total = O * I + O;
       long nchunk = std::max<long>(

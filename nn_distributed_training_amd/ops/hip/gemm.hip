@@ -203,4 +203,118 @@ __global__ void dw_small_chunked_k(
   }
 }
 
+// Skinny-layer dW with fully coalesced row reads (the generic
+// dw_small_chunked walks columns with stride-O/stride-I gathers and was
+// 8% of the density round). Two shapes:
+//   I <= IMAX (FourierNet encode, dW[256, 2]): lanes own o-columns,
+//     dZ rows read coalesced, the tiny X row broadcast-loaded;
+//   O <= OMAX (density head, dW[1, 64]): lanes own i-columns, X rows
+//     read coalesced, the tiny dZ row broadcast-loaded.
+// M is chunked over blocks; atomic accumulation (grad slice zeroed).
+template <typename T, int IMAX>
+__global__ void dw_skinny_i_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O, int nchunk) {
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const int mc = (M + nchunk - 1) / nchunk;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+  const T* dZl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int NO = (O + WAVE - 1) / WAVE;
+
+  T acc[4][IMAX + 1] = {};  // [o-slot][i or bias]; NO <= 4 enforced
+  for (int m = mlo + wid; m < mhi; m += 4) {
+    T xv[IMAX];
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      xv[i] = (i < I) ? Xl[(long)m * I + i] : T(0);  // broadcast load
+    }
+    for (int no = 0; no < NO; ++no) {
+      const int o = no * WAVE + lane;
+      if (o < O) {
+        const T g = dZl[(long)m * O + o];  // coalesced
+        acc[no][IMAX] += g;
+#pragma unroll
+        for (int i = 0; i < IMAX; ++i) {
+          if (i < I) acc[no][i] += g * xv[i];
+        }
+      }
+    }
+  }
+  for (int no = 0; no < NO; ++no) {
+    const int o = no * WAVE + lane;
+    if (o < O) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
+          atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
+                    acc[no][i]);
+        }
+      }
+      atomicAdd(&gstack[l * n + b_off + o], acc[no][IMAX]);
+    }
+  }
+}
+
+template <typename T, int OMAX>
+__global__ void dw_skinny_o_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O, int nchunk) {
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const int mc = (M + nchunk - 1) / nchunk;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+  const T* dZl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int NI = (I + WAVE - 1) / WAVE;
+
+  T acc[OMAX][4] = {};  // [o][i-slot]; NI <= 4 enforced
+  T bacc[OMAX] = {};
+  for (int m = mlo + wid; m < mhi; m += 4) {
+    T gv[OMAX];
+#pragma unroll
+    for (int o = 0; o < OMAX; ++o) {
+      gv[o] = (o < O) ? dZl[(long)m * O + o] : T(0);  // broadcast
+      if (o < O && lane == 0) bacc[o] += gv[o];
+    }
+    for (int ni = 0; ni < NI; ++ni) {
+      const int i = ni * WAVE + lane;
+      if (i < I) {
+        const T x = Xl[(long)m * I + i];  // coalesced
+#pragma unroll
+        for (int o = 0; o < OMAX; ++o) {
+          if (o < O) acc[o][ni] += gv[o] * x;
+        }
+      }
+    }
+  }
+  for (int ni = 0; ni < NI; ++ni) {
+    const int i = ni * WAVE + lane;
+    if (i < I) {
+#pragma unroll
+      for (int o = 0; o < OMAX; ++o) {
+        if (o < O) {
+          atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
+                    acc[o][ni]);
+        }
+      }
+    }
+  }
+  if (lane == 0) {
+#pragma unroll
+    for (int o = 0; o < OMAX; ++o) {
+      if (o < O) atomicAdd(&gstack[l * n + b_off + o], bacc[o]);
+    }
+  }
+}
+
 }  // namespace gemm

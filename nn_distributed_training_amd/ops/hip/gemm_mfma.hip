@@ -257,6 +257,12 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
   const int wn = (wid & 1) * 32;    // i-direction
 
   acc_t acc[2][2] = {};
+  // db rides along: the dZ tile is already staged COALESCED in As, so
+  // i-tile-0 blocks column-sum it instead of a separate bias kernel
+  // re-reading dZ with stride-O gathers (that kernel was 9% of the
+  // density round). Thread tid < BM owns column o = tid.
+  T db_acc = T(0);
+  const bool bias_block = (i0 == 0);
 
   for (int k0 = mlo; k0 < mhi; k0 += BK) {
     // o-fastest staging order: consecutive threads read consecutive
@@ -288,7 +294,14 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
         }
       }
     }
+    if (bias_block && tid < BM) {
+#pragma unroll
+      for (int k = 0; k < BK; ++k) db_acc += As[k][tid];
+    }
     __syncthreads();
+  }
+  if (bias_block && tid < BM && o0 + tid < O) {
+    atomicAdd(&gstack[l * n + b_off + o0 + tid], db_acc);
   }
 
 #pragma unroll
