@@ -680,8 +680,7 @@ class DiNNOStackedDriver:
         self.v = torch.zeros_like(eng.theta)
         self.rho = conf["rho_init"]
         self.step_t = 0
-        self._plan_key = None
-        self._plan = None
+        self._plans = {}
         # hipGraph capture of the whole round (NDTA_GRAPHS=1 enables).
         # Measured on MI355X: NEUTRAL to -4% vs eager at both 8-node and
         # 2-node packings — the eager host launches already hide under
@@ -719,16 +718,18 @@ class DiNNOStackedDriver:
 
     def _round_plan(self):
         key = _edge_key(self.pr)
-        if key == self._plan_key:
-            return self._plan
+        plan = self._plans.get(key)
+        if plan is not None:
+            return plan
         eng = self.eng
         remote_nodes, rbuf, dests = eng.remote_plan()
         row_of = eng.row_map(remote_nodes)
         offs, idx, _ = eng.build_csr(row_of, include_self=False)
         deg = eng.degrees()
-        self._plan = (rbuf, dests, offs, idx, deg)
-        self._plan_key = key
-        return self._plan
+        plan = (rbuf, dests, offs, idx, deg)
+        if len(self._plans) < 256:
+            self._plans[key] = plan
+        return plan
 
     def _graph_body(self, rbuf, offs, idx, deg):
         """One DiNNO round with all round-varying scalars read from
@@ -899,23 +900,24 @@ class DSGDStackedDriver:
     def prepare(self):
         self.alph = self.opt.alph0
         self.theta_next = torch.empty_like(self.eng.theta)
-        self._plan_key = None
-        self._plan = None
+        self._plans = {}
 
     def _round_plan(self):
         from ..utils import graph_generation
 
         key = _edge_key(self.pr)
-        if key == self._plan_key:
-            return self._plan
+        plan = self._plans.get(key)
+        if plan is not None:
+            return plan
         eng = self.eng
         W = graph_generation.get_metropolis(self.pr.graph)
         remote_nodes, rbuf, dests = eng.remote_plan()
         row_of = eng.row_map(remote_nodes)
         offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
-        self._plan = (rbuf, dests, offs, idx, w)
-        self._plan_key = key
-        return self._plan
+        plan = (rbuf, dests, offs, idx, w)
+        if len(self._plans) < 256:
+            self._plans[key] = plan
+        return plan
 
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng
@@ -966,8 +968,7 @@ class DSGTStackedDriver:
         self.g = torch.zeros_like(eng.theta)
         self.y_mix = torch.zeros_like(eng.theta)
         self.theta_next = torch.empty_like(eng.theta)
-        self._plan_key = None
-        self._plan = None
+        self._plans = {}
         if self.opt.conf["init_grads"]:
             xb, yb = eng.next_batch()
             eng.forward(xb, train_skip_logp=True)
@@ -979,8 +980,9 @@ class DSGTStackedDriver:
         from ..utils import graph_generation
 
         key = _edge_key(self.pr)
-        if key == self._plan_key:
-            return self._plan
+        plan = self._plans.get(key)
+        if plan is not None:
+            return plan
         eng = self.eng
         W = graph_generation.get_metropolis(self.pr.graph)
         remote_nodes, rbuf, _ = eng.remote_plan(width_factor=2)
@@ -992,9 +994,10 @@ class DSGTStackedDriver:
             if rbuf is not None else {}
         row_of = eng.row_map(remote_nodes)
         offs, idx, w = eng.build_csr(row_of, include_self=True, W=W)
-        self._plan = (rbuf, dests_p, dests_y, offs, idx, w)
-        self._plan_key = key
-        return self._plan
+        plan = (rbuf, dests_p, dests_y, offs, idx, w)
+        if len(self._plans) < 256:
+            self._plans[key] = plan
+        return plan
 
     def step_round(self, k):
         opt, pr, eng = self.opt, self.pr, self.eng

@@ -67,18 +67,25 @@ def build_csr(graph, local_nodes, row_of, device, dtype,
             if W is not None:
                 wts.append(float(W[i, j]))
         offs.append(len(idx))
-    offs_t = torch.tensor(offs, dtype=torch.int32, device=device)
-    idx_t = torch.tensor(idx, dtype=torch.int32, device=device)
-    w_t = (
-        torch.tensor(wts, dtype=dtype, device=device)
-        if W is not None
-        else None
-    )
+    # pinned staging + non_blocking H2D: a pageable tensor-to-device
+    # copy SYNCS the stream, and dynamic graphs rebuild the CSR often —
+    # the sync was serializing host and GPU on the density bench
+    def _to_dev(lst, dt):
+        t = torch.tensor(lst, dtype=dt)
+        if device is not None and torch.device(device).type == "cuda":
+            return t.pin_memory().to(device, non_blocking=True)
+        return t.to(device)
+
+    offs_t = _to_dev(offs, torch.int32)
+    idx_t = _to_dev(idx, torch.int32)
+    w_t = _to_dev(wts, dtype) if W is not None else None
     return offs_t, idx_t, w_t
 
 
 def degrees(graph, local_nodes, device):
-    return torch.tensor(
-        [graph.degree(i) for i in local_nodes],
-        dtype=torch.int32, device=device,
+    t = torch.tensor(
+        [graph.degree(i) for i in local_nodes], dtype=torch.int32
     )
+    if torch.device(device).type == "cuda":
+        return t.pin_memory().to(device, non_blocking=True)
+    return t.to(device)
