@@ -321,24 +321,4 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
   }
 }
 
-// db[O] = sum_m dZ[m, o], M-chunked over blocks, atomic accumulate.
-// (kept separate from mfma_dw_k: the bias reduction is over the A image
-// only and one light kernel beats complicating the MFMA epilogue)
-template <typename T>
-__global__ void bias_grad_chunked_k(
-    const T* __restrict__ dZ, T* __restrict__ gstack,
-    long n, long b_off, int M, int O, int nchunk) {
-  const int chunk = blockIdx.z % nchunk;
-  const long l = blockIdx.z / nchunk;
-  const int mc = (M + nchunk - 1) / nchunk;
-  const int mlo = chunk * mc;
-  const int mhi = min(M, mlo + mc);
-  const int o = blockIdx.x * blockDim.x + threadIdx.x;
-  if (o >= O) return;
-  const T* Gl = dZ + l * (long)M * O;
-  T acc = T(0);
-  for (int m = mlo; m < mhi; ++m) acc += Gl[(long)m * O + o];
-  atomicAdd(&gstack[l * n + b_off + o], acc);
-}
-
 }  // namespace gmfma

@@ -1,0 +1,48 @@
+"""bench.py driver-contract test: flags accepted, one JSON line with the
+required schema, whole-job value semantics."""
+
+import json
+import subprocess
+import sys
+
+REQUIRED = {
+    "metric", "value", "unit", "n_gpus", "steps", "warmup",
+    "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+    "dtype", "data", "config",
+}
+
+
+def _run(args):
+    out = subprocess.run(
+        [sys.executable, "bench.py", *args],
+        capture_output=True, text=True, timeout=420,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    return json.loads(line)
+
+
+def test_bench_default_contract():
+    d = _run(["--steps", "4", "--warmup", "1",
+              "--samples-per-node", "256"])
+    assert REQUIRED <= set(d)
+    assert d["metric"] == "comm_rounds_per_sec"
+    assert d["n_gpus"] == 1
+    assert d["steps"] == 4 and d["warmup"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "strong"
+    assert d["data"] == "synthetic"
+    assert d["dtype"] == "fp64"
+    assert d["value"] > 0
+    assert abs(d["ms_per_step"] - 4 / d["value"] * 1e3 / 4) < 1e6
+    cfg = d["config"]
+    assert cfg["nodes"] == 8 and cfg["alg"] == "dinno"
+    assert "consensus_err_max" in cfg and "val_acc_max" in cfg
+    assert "parallelism" in cfg
+
+
+def test_bench_dsgd_config2():
+    d = _run(["--alg", "dsgd", "--nodes", "4", "--steps", "3",
+              "--warmup", "1", "--samples-per-node", "256"])
+    assert d["config"]["alg"] == "dsgd"
+    assert d["config"]["nodes"] == 4
