@@ -227,6 +227,9 @@ __global__ void dw_skinny_i_k(
   const int wid = threadIdx.x / WAVE;
   const int NO = (O + WAVE - 1) / WAVE;
 
+  // every acc index is compile-time (full unroll + early break): a
+  // runtime-indexed accumulator array would live in SCRATCH, not
+  // registers (CDNA guide §5.4 rule 20 — measured 8x slower here)
   T acc[4][IMAX + 1] = {};  // [o-slot][i or bias]; NO <= 4 enforced
   for (int m = mlo + wid; m < mhi; m += 4) {
     T xv[IMAX];
@@ -234,7 +237,9 @@ __global__ void dw_skinny_i_k(
     for (int i = 0; i < IMAX; ++i) {
       xv[i] = (i < I) ? Xl[(long)m * I + i] : T(0);  // broadcast load
     }
-    for (int no = 0; no < NO; ++no) {
+#pragma unroll
+    for (int no = 0; no < 4; ++no) {
+      if (no >= NO) break;
       const int o = no * WAVE + lane;
       if (o < O) {
         const T g = dZl[(long)m * O + o];  // coalesced
@@ -246,7 +251,9 @@ __global__ void dw_skinny_i_k(
       }
     }
   }
-  for (int no = 0; no < NO; ++no) {
+#pragma unroll
+  for (int no = 0; no < 4; ++no) {
+    if (no >= NO) break;
     const int o = no * WAVE + lane;
     if (o < O) {
 #pragma unroll
@@ -286,7 +293,9 @@ __global__ void dw_skinny_o_k(
       gv[o] = (o < O) ? dZl[(long)m * O + o] : T(0);  // broadcast
       if (o < O && lane == 0) bacc[o] += gv[o];
     }
-    for (int ni = 0; ni < NI; ++ni) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      if (ni >= NI) break;
       const int i = ni * WAVE + lane;
       if (i < I) {
         const T x = Xl[(long)m * I + i];  // coalesced
@@ -297,7 +306,9 @@ __global__ void dw_skinny_o_k(
       }
     }
   }
-  for (int ni = 0; ni < NI; ++ni) {
+#pragma unroll
+  for (int ni = 0; ni < 4; ++ni) {
+    if (ni >= NI) break;
     const int i = ni * WAVE + lane;
     if (i < I) {
 #pragma unroll
