@@ -316,6 +316,16 @@ void linear_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
           X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), zp, n, w_off, b_off,
           (int)M, (int)I, (int)O, (int)act, (scalar_t)scale);
+    } else if (I <= 4 && M >= 1024) {
+      const long total = (long)M * O;
+      const long blocks =
+          std::min<long>((total + 255) / 256, 2048);
+      const size_t shmem = (size_t)(O * I + O) * sizeof(scalar_t);
+      hipLaunchKernelGGL((gemm::linear_fwd_smallk_k<scalar_t, 4>),
+          dim3(blocks, 1, L), dim3(256), shmem, cur_stream(),
+          X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), zp, n, w_off, b_off,
+          (int)M, (int)I, (int)O, (int)act, (scalar_t)scale);
     } else {
       dim3 grid((O + 15) / 16, (M + 15) / 16, L);
       hipLaunchKernelGGL(gemm::linear_fwd_k<scalar_t>,

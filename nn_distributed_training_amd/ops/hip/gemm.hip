@@ -57,6 +57,45 @@ __global__ void linear_fwd_k(
   }
 }
 
+// Small-K forward (FourierNet encode: I = 2): the tiled kernel's LDS
+// K-loop degenerates there. Weights+bias stay in LDS; one thread per
+// output element, the block's threads share one X row (broadcast) and
+// write coalesced.
+template <typename T, int IMAX>
+__global__ void linear_fwd_smallk_k(
+    const T* __restrict__ X, const T* __restrict__ theta,
+    T* __restrict__ Y, T* __restrict__ Z,
+    long n, long w_off, long b_off, int M, int I, int O,
+    int act, T scale) {
+  extern __shared__ __align__(16) unsigned char smem_raw[];
+  T* wb = reinterpret_cast<T*>(smem_raw);  // [O*I + O]
+  const long l = blockIdx.z;
+  const T* W = theta + l * n + w_off;
+  const T* bias = theta + l * n + b_off;
+  for (int t = threadIdx.x; t < O * I; t += blockDim.x) wb[t] = W[t];
+  for (int t = threadIdx.x; t < O; t += blockDim.x) {
+    wb[O * I + t] = bias[t];
+  }
+  __syncthreads();
+
+  const T* Xl = X + l * (long)M * I;
+  T* Yl = Y + l * (long)M * O;
+  T* Zl = Z ? Z + l * (long)M * O : nullptr;
+  const long total = (long)M * O;
+  for (long t = blockIdx.x * (long)blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const long m = t / O;
+    const int o = (int)(t - m * O);
+    T acc = wb[O * I + o];
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      if (i < I) acc += Xl[m * I + i] * wb[o * I + i];
+    }
+    if (Zl) Zl[t] = acc;
+    Yl[t] = act_fwd(act, acc, scale);
+  }
+}
+
 // dZ = dY * act'(z, y) — fused activation backward, one pass.
 template <typename T>
 __global__ void act_grad_k(

@@ -59,7 +59,7 @@ struct mfma_t<float> {
 
 constexpr int BM = 64;   // block tile rows
 constexpr int BN = 64;   // block tile cols
-constexpr int BK = 16;   // K per LDS stage
+constexpr int BK = 16;   // K per LDS stage (BK=32 measured neutral)
 
 // ---------------------------------------------------------------------
 // NT: Y[M,O] = act(X[M,I] @ W_l[O,I]^T + b_l)  [+ optional Z store]

@@ -58,10 +58,11 @@ def test_linear_fwd(ext, dtype, act, actname):
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
-def test_linear_fwd_sin_relu(ext, dtype):
+@pytest.mark.parametrize("M", [40, 2048])  # tiled vs small-K kernels
+def test_linear_fwd_sin_relu(ext, dtype, M):
     """FourierNet encode: relu(sin(scale * (Wx+b))) with Z saved."""
     torch.manual_seed(1)
-    L, M, I, O = 2, 40, 2, 24
+    L, I, O = 2, 2, 24
     n = I * O + O
     dev = _dev()
     scale = 0.05
