@@ -9,6 +9,7 @@
 #include "gemm_mfma.hip"
 #include "conv.hip"
 #include "losses.hip"
+#include "fused_mnist.hip"
 
 #include <ATen/hip/HIPContext.h>
 
@@ -486,6 +487,43 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
 }
 
 // ------------------------------------------------------------ losses --
+void mnist_train_step(
+    torch::Tensor X_all, torch::Tensor Y_all, torch::Tensor idx,
+    c10::optional<torch::Tensor> offs_dev, torch::Tensor theta,
+    torch::Tensor grad, c10::optional<torch::Tensor> loss,
+    long pit, long idx_off, long idx_stride,
+    long wc_off, long bc_off, long w1_off, long b1_off, long w2_off,
+    long b2_off, long B, long F, long K, long IMG, long H, long C,
+    long TI, double loss_scale) {
+  CHECK_DEV(X_all); CHECK_DEV(theta); CHECK_DEV(grad);
+  const long L = theta.size(0), n = theta.size(1);
+  const long maxlen = Y_all.size(1);
+  TORCH_CHECK(bc_off == wc_off + F * K * K,
+              "conv weight/bias must be contiguous in the flat layout");
+  TORCH_CHECK(K <= 7, "fused mnist step supports kernel size <= 7");
+  const long P = (IMG - (K - 1)) / 2;
+  const long PF = F * P * P;
+  dim3 grid((B + TI - 1) / TI, 1, L);
+  DISPATCH_FT(X_all, {
+    const size_t shmem =
+        (size_t)(TI * (IMG * IMG + PF + 2 * H + C) + F * K * K + F) *
+            sizeof(scalar_t) +
+        ((size_t)TI * PF + 15) / 16 * 16 + (size_t)TI * sizeof(long);
+    hipLaunchKernelGGL(fmnist::mnist_train_step_k<scalar_t>,
+        grid, dim3(256), shmem, cur_stream(),
+        X_all.data_ptr<scalar_t>(), Y_all.data_ptr<long>(),
+        idx.data_ptr<long>(),
+        offs_dev.has_value() ? offs_dev->data_ptr<long>() : nullptr,
+        theta.data_ptr<scalar_t>(), grad.data_ptr<scalar_t>(),
+        loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr,
+        (int)pit, idx_off, idx_stride, maxlen, n,
+        wc_off, bc_off, w1_off, b1_off, w2_off, b2_off,
+        (int)B, (int)F, (int)K, (int)IMG, (int)H, (int)C, (int)TI,
+        (scalar_t)loss_scale);
+  });
+  HIP_CHECK_LAST();
+}
+
 void logsoftmax(torch::Tensor Z, torch::Tensor P, long C) {
   CHECK_DEV(Z); CHECK_DEV(P);
   const long M = Z.numel() / C;
@@ -598,6 +636,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("logsoftmax", &logsoftmax);
   mod.def("nll_bwd", &nll_bwd);
   mod.def("nll_fused", &nll_fused);
+  mod.def("mnist_train_step", &mnist_train_step);
   mod.def("bce_bwd", &bce_bwd);
   mod.def("regression_bwd", &regression_bwd);
 }

@@ -1,0 +1,253 @@
+// Fully fused MNIST train step (CDNA4).
+//
+// One kernel performs, per tile of T images of one node replica:
+//   gather (index stream) -> conv+ReLU+maxpool -> fc1+ReLU ->
+//   fc2 logits -> log-sum-exp + NLL dZ (targets read from the resident
+//   label array) -> fc2 dW/db -> dh1 -> fc1 dW/db -> dpool ->
+//   conv dW/db
+// with every intermediate living in LDS. The layered engine ran this
+// as ~12 launches per primal iteration with every activation making an
+// HBM round trip; at n=28,440 the work per launch is tiny and the round
+// is kernel-boundary-bound (profiles/README.md), so fusing the chain is
+// the lever the per-kernel optimizations could not reach.
+//
+// Gradients accumulate ATOMICALLY into the [L, n] grad stack (caller
+// zeroes it); blocks are independent (one per (l, image-tile)), so no
+// inter-workgroup ordering is assumed anywhere. fc1's 27,648 weights
+// stay in L2 (221 KB fp64 — larger than LDS); each block streams them
+// for fc1-forward and once more for dpool.
+//
+// LDS budget at T=8, fp64: img 50.2KB + pool 27.6KB + h1/dh1 8KB +
+// small ~ 90KB -> 1 block/CU; T is a launch parameter (dynamic LDS).
+
+#include "common.h"
+
+namespace fmnist {
+
+template <typename T>
+__global__ __launch_bounds__(256) void mnist_train_step_k(
+    const T* __restrict__ X_all,      // [L, maxlen, IMG*IMG]
+    const long* __restrict__ Y_all,   // [L, maxlen]
+    const long* __restrict__ idx,     // [L, S] index stream
+    const long* __restrict__ offs_dev,  // nullable; slot `pit`
+    const T* __restrict__ theta,      // [L, n]
+    T* __restrict__ grad,             // [L, n] atomic accumulate
+    T* __restrict__ loss,             // nullable [L]
+    int pit, long idx_off, long idx_stride, long maxlen, long n,
+    long wc_off, long bc_off, long w1_off, long b1_off, long w2_off,
+    long b2_off,
+    int B, int F, int K, int IMG, int H, int C, int TI,
+    T loss_scale) {
+  extern __shared__ __align__(16) unsigned char smem_raw[];
+  const int conv_out = IMG - (K - 1);
+  const int P = conv_out / 2;
+  const int PF = F * P * P;          // pooled features (432)
+
+  T* img = reinterpret_cast<T*>(smem_raw);        // [TI][IMG*IMG]
+  T* pool = img + (long)TI * IMG * IMG;           // [TI][PF]
+  T* h1 = pool + (long)TI * PF;                   // [TI][H]
+  T* dh1 = h1 + (long)TI * H;                     // [TI][H]
+  T* dz2 = dh1 + (long)TI * H;                    // [TI][C]
+  T* wconv = dz2 + (long)TI * C;                  // [F*K*K + F]
+  unsigned char* pidx =
+      reinterpret_cast<unsigned char*>(wconv + F * K * K + F);
+  long* src = reinterpret_cast<long*>(
+      pidx + ((long)TI * PF + 15) / 16 * 16);     // [TI]
+
+  const long l = blockIdx.z;
+  const int tile = blockIdx.x;
+  const int t0 = tile * TI;                       // first image index
+  const int tcnt = min(TI, B - t0);
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+
+  const T* th = theta + l * n;
+  T* gr = grad + l * n;
+  const long off = offs_dev ? offs_dev[pit] : idx_off;
+
+  // ---- P0: resolve sources, stage images + conv weights ----
+  if (tid < tcnt) {
+    src[tid] = idx[l * idx_stride + off + t0 + tid];
+  }
+  for (int t = tid; t < F * K * K + F; t += 256) {
+    wconv[t] = th[wc_off + t];  // wc..|bc.. contiguous in the layout
+  }
+  __syncthreads();
+  for (int u = tid; u < tcnt * IMG * IMG; u += 256) {
+    const int t = u / (IMG * IMG);
+    img[u] = X_all[(l * maxlen + src[t]) * IMG * IMG
+                   + (u - t * IMG * IMG)];
+  }
+  __syncthreads();
+
+  // ---- P1: conv + ReLU + maxpool forward ----
+  for (int u = tid; u < tcnt * PF; u += 256) {
+    const int t = u / PF;
+    const int r = u - t * PF;
+    const int f = r / (P * P);
+    const int py = (r / P) % P;
+    const int px = r % P;
+    const T* wf = wconv + f * K * K;
+    const T bias = wconv[F * K * K + f];
+    const T* im = img + (long)t * IMG * IMG;
+    T best = T(0);
+    int best_i = 0;
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      const int cy = 2 * py + (d >> 1);
+      const int cx = 2 * px + (d & 1);
+      T acc = bias;
+      for (int ky = 0; ky < K; ++ky) {
+        const T* row = im + (cy + ky) * IMG + cx;
+        const T* wr = wf + ky * K;
+        for (int kx = 0; kx < K; ++kx) acc += row[kx] * wr[kx];
+      }
+      if (acc > best) { best = acc; best_i = d; }
+    }
+    pool[u] = best;
+    pidx[u] = (unsigned char)best_i;
+  }
+  __syncthreads();
+
+  // ---- P2: fc1 + ReLU (weights streamed from L2) ----
+  const T* W1 = th + w1_off;
+  for (int u = tid; u < tcnt * H; u += 256) {
+    const int t = u / H;
+    const int o = u - t * H;
+    const T* w = W1 + (long)o * PF;
+    const T* x = pool + (long)t * PF;
+    T acc = th[b1_off + o];
+    for (int i = 0; i < PF; ++i) acc += x[i] * w[i];
+    h1[u] = acc > T(0) ? acc : T(0);
+  }
+  __syncthreads();
+
+  // ---- P3: fc2 logits + LSE + NLL dZ ----
+  const T* W2 = th + w2_off;
+  for (int u = tid; u < tcnt * C; u += 256) {
+    const int t = u / C;
+    const int o = u - t * C;
+    const T* w = W2 + (long)o * H;
+    const T* x = h1 + (long)t * H;
+    T acc = th[b2_off + o];
+    for (int i = 0; i < H; ++i) acc += x[i] * w[i];
+    dz2[u] = acc;  // logits, converted in place below
+  }
+  __syncthreads();
+  const T wmean = loss_scale / T(B);
+  if (tid < tcnt) {
+    T* z = dz2 + (long)tid * C;
+    const int y = (int)Y_all[l * maxlen + src[tid]];
+    T mx = z[0];
+    for (int c = 1; c < C; ++c) mx = z[c] > mx ? z[c] : mx;
+    T sum = T(0);
+    for (int c = 0; c < C; ++c) sum += ::exp(z[c] - mx);
+    const T lse = mx + ::log(sum);
+    if (loss != nullptr) {
+      atomicAdd(&loss[l], -(z[y] - lse) / T(B));
+    }
+    for (int c = 0; c < C; ++c) {
+      z[c] = (::exp(z[c] - lse) - (c == y ? T(1) : T(0))) * wmean;
+    }
+  }
+  __syncthreads();
+
+  // ---- P4: fc2 dW/db ----
+  for (int u = tid; u < C * H + C; u += 256) {
+    T acc = T(0);
+    if (u < C * H) {
+      const int o = u / H;
+      const int i = u - o * H;
+      for (int t = 0; t < tcnt; ++t) {
+        acc += dz2[(long)t * C + o] * h1[(long)t * H + i];
+      }
+      atomicAdd(&gr[w2_off + u], acc);
+    } else {
+      const int o = u - C * H;
+      for (int t = 0; t < tcnt; ++t) acc += dz2[(long)t * C + o];
+      atomicAdd(&gr[b2_off + o], acc);
+    }
+  }
+
+  // ---- P5: dh1 = dz2 @ W2, ReLU' ----
+  for (int u = tid; u < tcnt * H; u += 256) {
+    const int t = u / H;
+    const int i = u - t * H;
+    T acc = T(0);
+    for (int o = 0; o < C; ++o) {
+      acc += dz2[(long)t * C + o] * W2[(long)o * H + i];
+    }
+    dh1[u] = (h1[u] > T(0)) ? acc : T(0);
+  }
+  __syncthreads();
+
+  // ---- P6: fc1 dW/db ----
+  for (int u = tid; u < H * PF; u += 256) {
+    const int o = u / PF;
+    const int i = u - o * PF;
+    T acc = T(0);
+    for (int t = 0; t < tcnt; ++t) {
+      acc += dh1[(long)t * H + o] * pool[(long)t * PF + i];
+    }
+    atomicAdd(&gr[w1_off + u], acc);
+  }
+  for (int u = tid; u < H; u += 256) {
+    T acc = T(0);
+    for (int t = 0; t < tcnt; ++t) acc += dh1[(long)t * H + u];
+    atomicAdd(&gr[b1_off + u], acc);
+  }
+  __syncthreads();
+
+  // ---- P7: dpool = dh1 @ W1, ReLU' mask; pool is reused as dpool
+  // (safe: P6's readers finished at the barrier above, and here each
+  // thread reads/writes only its own pool[u]) ----
+  for (int u = tid; u < tcnt * PF; u += 256) {
+    const int t = u / PF;
+    const int i = u - t * PF;
+    T acc = T(0);
+    const T* g = dh1 + (long)t * H;
+    for (int o = 0; o < H; ++o) {
+      acc += g[o] * W1[(long)o * PF + i];
+    }
+    pool[u] = (pool[u] > T(0)) ? acc : T(0);
+  }
+  __syncthreads();
+
+  // ---- P8: conv dW/db via argmax routing ----
+  for (int f = 0; f < F; ++f) {
+    T dw[7 * 7];
+    T db = T(0);
+#pragma unroll
+    for (int i = 0; i < 7 * 7; ++i) dw[i] = T(0);
+    for (int u = tid; u < tcnt * P * P; u += 256) {
+      const int t = u / (P * P);
+      const int py = (u / P) % P;
+      const int px = u % P;
+      const long e = (long)t * PF + f * P * P + py * P + px;
+      const T g = pool[e];
+      if (g == T(0)) continue;
+      const int d = pidx[e];
+      const int cy = 2 * py + (d >> 1);
+      const int cx = 2 * px + (d & 1);
+      const T* im = img + (long)t * IMG * IMG;
+      db += g;
+      for (int ky = 0; ky < K; ++ky) {
+        const T* row = im + (cy + ky) * IMG + cx;
+        for (int kx = 0; kx < K; ++kx) dw[ky * K + kx] += g * row[kx];
+      }
+    }
+    for (int i = 0; i < K * K + 1; ++i) {
+      T v = (i < K * K) ? dw[i] : db;
+      v = wave_reduce_sum(v);
+      if (lane == 0) {
+        if (i < K * K) {
+          atomicAdd(&gr[wc_off + (long)f * K * K + i], v);
+        } else {
+          atomicAdd(&gr[bc_off + f], v);
+        }
+      }
+    }
+  }
+}
+
+}  // namespace fmnist

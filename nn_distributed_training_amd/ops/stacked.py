@@ -357,6 +357,46 @@ class StackedEngine:
         return pack
 
     # ------------------------------------------------------------------
+    # fully fused MNIST train step (ops/hip/fused_mnist.hip): one launch
+    # per primal iteration instead of the ~12-kernel layered chain.
+    # NDTA_FUSED=0 falls back to the layered path (A/B lever).
+    def fused_step_available(self) -> bool:
+        if os.environ.get("NDTA_FUSED", "1") != "1":
+            return False
+        if self.spec.name != "mnist_conv" or not self.classification:
+            return False
+        if not isinstance(self.sampler, _StreamSampler):
+            return False
+        return self.device.type == "cuda"
+
+    def run_fused_mnist(self, off=0, graph_offs=None, pit=0,
+                        want_loss=False):
+        conv, fc1, fc2 = self.spec.layers
+        self.grad.zero_()  # all three layers accumulate atomically
+        loss_buf = None
+        if want_loss:
+            if self._bufs is None:
+                self._bufs = self._alloc_bufs()
+            self._bufs["loss"].zero_()
+            loss_buf = self._bufs["loss"]
+        ti = int(os.environ.get("NDTA_FUSED_TI", "8"))
+        self.ext.mnist_train_step(
+            self.X_all, self.Y_all, self.sampler.stream, graph_offs,
+            self.theta, self.grad, loss_buf, pit, off, self.sampler.S,
+            conv.w_off, conv.b_off, fc1.w_off, fc1.b_off, fc2.w_off,
+            fc2.b_off, self.B, conv.out_dim, conv.kernel_size,
+            conv.in_dim, fc1.out_dim, fc2.out_dim, ti, 1.0,
+        )
+        return loss_buf
+
+    def fused_advance(self):
+        """Advance the sampler for a fused step; returns the offset."""
+        off = self.sampler.next_offset()
+        if self._has_node0:
+            self.pr.forward_cnt += self.B
+        return off
+
+    # ------------------------------------------------------------------
     def next_batch(self):
         """Assemble the next per-node batches with two gather kernels."""
         if self._bufs is None:
@@ -742,16 +782,22 @@ class DiNNOStackedDriver:
             eng.theta, rbuf, offs, idx, self.duals, self.s,
             self._sched_dev,
         )
+        fused = eng.fused_step_available()
         xb = bufs["xb"]
         for pit in range(self.pits):
-            ext.gather_batch_dev(
-                eng.X_all, eng.sampler.stream, xb, self._offs_dev,
-                pit, eng.sampler.S,
-            )
-            eng.forward(xb, train_skip_logp=True)
-            eng.backward(
-                xb, None, graph_offs=self._offs_dev, pit=pit
-            )
+            if fused:
+                eng.run_fused_mnist(
+                    graph_offs=self._offs_dev, pit=pit
+                )
+            else:
+                ext.gather_batch_dev(
+                    eng.X_all, eng.sampler.stream, xb, self._offs_dev,
+                    pit, eng.sampler.S,
+                )
+                eng.forward(xb, train_skip_logp=True)
+                eng.backward(
+                    xb, None, graph_offs=self._offs_dev, pit=pit
+                )
             first = (not self.persistent) and pit == 0
             ext.fused_step_sched(
                 eng.theta, eng.grad, self.duals, self.s, deg,
@@ -851,16 +897,22 @@ class DiNNOStackedDriver:
         )
 
         want_tl = bool(getattr(pr, "track_tloss", False))
+        fused = eng.fused_step_available()
         for pi in range(self.pits):
-            with _timer("next_batch"):
-                xb, yb = eng.next_batch()
-            with _timer("forward"):
-                eng.forward(xb, train_skip_logp=True)
-            with _timer("backward"):
-                wl = want_tl and pi == self.pits - 1
-                lb = eng.backward(xb, yb, want_loss=wl)
-                if wl:
-                    eng.update_tloss(lb)
+            wl = want_tl and pi == self.pits - 1
+            if fused:
+                with _timer("train_step"):
+                    off = eng.fused_advance()
+                    lb = eng.run_fused_mnist(off=off, want_loss=wl)
+            else:
+                with _timer("next_batch"):
+                    xb, yb = eng.next_batch()
+                with _timer("forward"):
+                    eng.forward(xb, train_skip_logp=True)
+                with _timer("backward"):
+                    lb = eng.backward(xb, yb, want_loss=wl)
+            if wl and lb is not None:
+                eng.update_tloss(lb)
             with _timer("fused_step"):
                 self.step_t += 1
                 ext.fused_step(
@@ -932,11 +984,15 @@ class DSGDStackedDriver:
         ext.mix_rows(eng.theta, rbuf, offs, idx, w, self.theta_next)
         eng.theta, self.theta_next = self.theta_next, eng.theta
 
-        xb, yb = eng.next_batch()
-        eng.forward(xb, train_skip_logp=True)
         want_tl = bool(getattr(pr, "track_tloss", False))
-        lb = eng.backward(xb, yb, want_loss=want_tl)
-        if want_tl:
+        if eng.fused_step_available():
+            off = eng.fused_advance()
+            lb = eng.run_fused_mnist(off=off, want_loss=want_tl)
+        else:
+            xb, yb = eng.next_batch()
+            eng.forward(xb, train_skip_logp=True)
+            lb = eng.backward(xb, yb, want_loss=want_tl)
+        if want_tl and lb is not None:
             eng.update_tloss(lb)
         ext.axpy(eng.theta, eng.grad, -self.alph)
 
@@ -970,9 +1026,12 @@ class DSGTStackedDriver:
         self.theta_next = torch.empty_like(eng.theta)
         self._plans = {}
         if self.opt.conf["init_grads"]:
-            xb, yb = eng.next_batch()
-            eng.forward(xb, train_skip_logp=True)
-            eng.backward(xb, yb)
+            if eng.fused_step_available():
+                eng.run_fused_mnist(off=eng.fused_advance())
+            else:
+                xb, yb = eng.next_batch()
+                eng.forward(xb, train_skip_logp=True)
+                eng.backward(xb, yb)
             self.y.copy_(eng.grad)
             self.g.copy_(eng.grad)
 
@@ -1019,15 +1078,20 @@ class DSGTStackedDriver:
             )
             eng.theta, self.theta_next = self.theta_next, eng.theta
 
-        with _timer("next_batch"):
-            xb, yb = eng.next_batch()
-        with _timer("forward"):
-            eng.forward(xb, train_skip_logp=True)
-        with _timer("backward"):
-            want_tl = bool(getattr(pr, "track_tloss", False))
-            lb = eng.backward(xb, yb, want_loss=want_tl)
-            if want_tl:
-                eng.update_tloss(lb)
+        want_tl = bool(getattr(pr, "track_tloss", False))
+        if eng.fused_step_available():
+            with _timer("train_step"):
+                off = eng.fused_advance()
+                lb = eng.run_fused_mnist(off=off, want_loss=want_tl)
+        else:
+            with _timer("next_batch"):
+                xb, yb = eng.next_batch()
+            with _timer("forward"):
+                eng.forward(xb, train_skip_logp=True)
+            with _timer("backward"):
+                lb = eng.backward(xb, yb, want_loss=want_tl)
+        if want_tl and lb is not None:
+            eng.update_tloss(lb)
         with _timer("y_update"):
             ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
 

@@ -231,6 +231,39 @@ def test_stacked_checkpoint_resume(alg, tmp_path):
 
 
 @requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+def test_fused_mnist_step_matches_layered(dtype, monkeypatch):
+    """The one-launch fused train-step's gradient == the layered
+    kernel chain's gradient for the same batch."""
+    torch.set_default_dtype(dtype)
+    conf = _conf(copy.deepcopy(ALG_CONFS["dsgd"]))
+
+    def grads(fused):
+        monkeypatch.setenv("NDTA_FUSED", "1" if fused else "0")
+        pr = _build_problem(conf)
+        pr.stacked = StackedEngine(pr)
+        eng = pr.stacked
+        assert eng.fused_step_available() == fused
+        if fused:
+            off = eng.fused_advance()
+            eng.run_fused_mnist(off=off)
+        else:
+            xb, yb = eng.next_batch()
+            eng.forward(xb, train_skip_logp=True)
+            eng.backward(xb, yb)
+        return eng.grad.clone()
+
+    g_layered = grads(False)
+    g_fused = grads(True)
+    tol = (
+        dict(rtol=1e-10, atol=1e-12)
+        if dtype == torch.float64
+        else dict(rtol=2e-3, atol=1e-5)
+    )
+    torch.testing.assert_close(g_fused, g_layered, **tol)
+
+
+@requires_gpu
 def test_stacked_validation_matches_eager():
     """Stacked batched validation == eager per-node torch validation."""
     torch.set_default_dtype(torch.float64)
