@@ -501,12 +501,17 @@ void mnist_train_step(
   TORCH_CHECK(bc_off == wc_off + F * K * K,
               "conv weight/bias must be contiguous in the flat layout");
   TORCH_CHECK(K <= 7, "fused mnist step supports kernel size <= 7");
+  TORCH_CHECK(TI >= 1 && TI <= 8,
+              "fused mnist step register tiling assumes TI <= 8");
+  TORCH_CHECK(H <= 64 * TI || TI * H <= 512,
+              "fused mnist step per-thread output budget exceeded");
   const long P = (IMG - (K - 1)) / 2;
   const long PF = F * P * P;
   dim3 grid((B + TI - 1) / TI, 1, L);
   DISPATCH_FT(X_all, {
     const size_t shmem =
-        (size_t)(TI * (IMG * IMG + PF + 2 * H + C) + F * K * K + F) *
+        (size_t)(TI * (IMG * IMG + PF + 2 * H + C) + F * K * K + F +
+                 C * H + C + 16 * 65) *
             sizeof(scalar_t) +
         ((size_t)TI * PF + 15) / 16 * 16 + (size_t)TI * sizeof(long);
     hipLaunchKernelGGL(fmnist::mnist_train_step_k<scalar_t>,

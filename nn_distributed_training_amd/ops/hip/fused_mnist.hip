@@ -18,7 +18,8 @@
 // for fc1-forward and once more for dpool.
 //
 // LDS budget at T=8, fp64: img 50.2KB + pool 27.6KB + h1/dh1 8KB +
-// small ~ 90KB -> 1 block/CU; T is a launch parameter (dynamic LDS).
+// W2 5.2KB + W1 tile 8.3KB + small ~ 104KB -> 1 block/CU; T is a
+// launch parameter (dynamic LDS).
 
 #include "common.h"
 
@@ -49,8 +50,10 @@ __global__ __launch_bounds__(256) void mnist_train_step_k(
   T* dh1 = h1 + (long)TI * H;                     // [TI][H]
   T* dz2 = dh1 + (long)TI * H;                    // [TI][C]
   T* wconv = dz2 + (long)TI * C;                  // [F*K*K + F]
+  T* w2s = wconv + F * K * K + F;                 // [C*H + C]
+  T* wt = w2s + (long)C * H + C;                  // [16][65] W1 tiles
   unsigned char* pidx =
-      reinterpret_cast<unsigned char*>(wconv + F * K * K + F);
+      reinterpret_cast<unsigned char*>(wt + 16 * 65);
   long* src = reinterpret_cast<long*>(
       pidx + ((long)TI * PF + 15) / 16 * 16);     // [TI]
 
@@ -72,6 +75,8 @@ __global__ __launch_bounds__(256) void mnist_train_step_k(
   for (int t = tid; t < F * K * K + F; t += 256) {
     wconv[t] = th[wc_off + t];  // wc..|bc.. contiguous in the layout
   }
+  for (int t = tid; t < C * H; t += 256) w2s[t] = th[w2_off + t];
+  for (int t = tid; t < C; t += 256) w2s[C * H + t] = th[b2_off + t];
   __syncthreads();
   for (int u = tid; u < tcnt * IMG * IMG; u += 256) {
     const int t = u / (IMG * IMG);
@@ -109,27 +114,56 @@ __global__ __launch_bounds__(256) void mnist_train_step_k(
   }
   __syncthreads();
 
-  // ---- P2: fc1 + ReLU (weights streamed from L2) ----
+  // ---- P2: fc1 + ReLU as an LDS-tiled mini-GEMM. Per 16-deep
+  // K-tile: stage wt[k][o] = W1[o][k0+k] with consecutive threads
+  // reading consecutive W1-row elements (the v1 per-thread serial dot
+  // walked W1 with a 3.5KB stride across lanes — the dominant cost of
+  // the first fused version), then accumulate (t, o) outputs.
   const T* W1 = th + w1_off;
-  for (int u = tid; u < tcnt * H; u += 256) {
-    const int t = u / H;
-    const int o = u - t * H;
-    const T* w = W1 + (long)o * PF;
-    const T* x = pool + (long)t * PF;
-    T acc = th[b1_off + o];
-    for (int i = 0; i < PF; ++i) acc += x[i] * w[i];
-    h1[u] = acc > T(0) ? acc : T(0);
+  {
+    T acc[2] = {};  // (t, o) outputs per thread: TI*H/256 <= 2 at TI=8
+    for (int k0 = 0; k0 < PF; k0 += 16) {
+      for (int u = tid; u < H * 16; u += 256) {
+        const int o = u / 16, k = u & 15;
+        wt[k * 65 + o] =
+            (k0 + k < PF) ? W1[(long)o * PF + k0 + k] : T(0);
+      }
+      __syncthreads();
+      const int kmax = min(16, PF - k0);
+#pragma unroll
+      for (int a = 0; a < 2; ++a) {
+        const int u = tid + a * 256;
+        if (u < tcnt * H) {
+          const int t = u / H;
+          const int o = u - t * H;
+          const T* x = pool + (long)t * PF + k0;
+          T sacc = T(0);
+          for (int k = 0; k < kmax; ++k) {
+            sacc += x[k] * wt[k * 65 + o];
+          }
+          acc[a] += sacc;
+        }
+      }
+      __syncthreads();
+    }
+#pragma unroll
+    for (int a = 0; a < 2; ++a) {
+      const int u = tid + a * 256;
+      if (u < tcnt * H) {
+        const T z = acc[a] + th[b1_off + (u % H)];
+        h1[u] = z > T(0) ? z : T(0);
+      }
+    }
   }
   __syncthreads();
 
-  // ---- P3: fc2 logits + LSE + NLL dZ ----
-  const T* W2 = th + w2_off;
+  // ---- P3: fc2 logits (W2 is LDS-resident) + LSE + NLL dZ ----
   for (int u = tid; u < tcnt * C; u += 256) {
     const int t = u / C;
     const int o = u - t * C;
-    const T* w = W2 + (long)o * H;
+    const T* w = w2s + (long)o * H;
     const T* x = h1 + (long)t * H;
-    T acc = th[b2_off + o];
+    T acc = w2s[C * H + o];
     for (int i = 0; i < H; ++i) acc += x[i] * w[i];
     dz2[u] = acc;  // logits, converted in place below
   }
@@ -175,7 +209,7 @@ __global__ __launch_bounds__(256) void mnist_train_step_k(
     const int i = u - t * H;
     T acc = T(0);
     for (int o = 0; o < C; ++o) {
-      acc += dz2[(long)t * C + o] * W2[(long)o * H + i];
+      acc += dz2[(long)t * C + o] * w2s[(long)o * H + i];
     }
     dh1[u] = (h1[u] > T(0)) ? acc : T(0);
   }
@@ -198,18 +232,53 @@ __global__ __launch_bounds__(256) void mnist_train_step_k(
   }
   __syncthreads();
 
-  // ---- P7: dpool = dh1 @ W1, ReLU' mask; pool is reused as dpool
-  // (safe: P6's readers finished at the barrier above, and here each
-  // thread reads/writes only its own pool[u]) ----
-  for (int u = tid; u < tcnt * PF; u += 256) {
-    const int t = u / PF;
-    const int i = u - t * PF;
-    T acc = T(0);
-    const T* g = dh1 + (long)t * H;
-    for (int o = 0; o < H; ++o) {
-      acc += g[o] * W1[(long)o * PF + i];
+  // ---- P7: dpool = dh1 @ W1 as a tiled GEMM over 64-wide i-tiles and
+  // 16-deep o-tiles: wt[o][i] staged with consecutive threads reading
+  // consecutive W1 elements (coalesced), accumulators in registers.
+  // pool is reused as dpool (safe: P6's readers hit the barrier above;
+  // the ReLU' mask is applied from the saved sign via a fresh read of
+  // pool BEFORE overwrite within the same thread's element).
+  {
+    for (int i0 = 0; i0 < PF; i0 += 64) {
+      const int imax = min(64, PF - i0);
+      T acc[2] = {};  // (t, ii) outputs: TI*64/256 = 2 at TI = 8
+      for (int o0 = 0; o0 < H; o0 += 16) {
+        // wt[o][ii] = W1[o0+o][i0+ii]; thread u: o = u/64, ii = u%64
+        for (int u = tid; u < 16 * 64; u += 256) {
+          const int o = u / 64, ii = u & 63;
+          wt[o * 65 + ii] = (ii < imax)
+                                ? W1[(long)(o0 + o) * PF + i0 + ii]
+                                : T(0);
+        }
+        __syncthreads();
+#pragma unroll
+        for (int a = 0; a < 2; ++a) {
+          const int u = tid + a * 256;
+          const int t = u / 64;
+          const int ii = u & 63;
+          if (t < tcnt) {
+            const T* g = dh1 + (long)t * H + o0;
+            T sacc = T(0);
+            for (int o = 0; o < 16; ++o) {
+              sacc += g[o] * wt[o * 65 + ii];
+            }
+            acc[a] += sacc;
+          }
+        }
+        __syncthreads();
+      }
+#pragma unroll
+      for (int a = 0; a < 2; ++a) {
+        const int u = tid + a * 256;
+        const int t = u / 64;
+        const int ii = u & 63;
+        if (t < tcnt && ii < imax) {
+          const long e = (long)t * PF + i0 + ii;
+          pool[e] = (pool[e] > T(0)) ? acc[a] : T(0);
+        }
+      }
+      __syncthreads();
     }
-    pool[u] = (pool[u] > T(0)) ? acc : T(0);
   }
   __syncthreads();
 
