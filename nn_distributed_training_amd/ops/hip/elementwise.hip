@@ -146,6 +146,25 @@ __global__ void dsgt_y_update_k(
 // the per-round m/v zero-fill kernels of the non-persistent DiNNO mode
 // fold away (reference recreates the Adam optimizer each round,
 // dinno.py:57-72).
+// Reduce per-tile gradient slabs [L, P, n] into [L, n] (consumers that
+// are not the fused step: DSGD's axpy, DSGT's tracker update).
+template <typename T>
+__global__ void reduce_parts_k(const T* __restrict__ parts,
+                               T* __restrict__ out, int nparts,
+                               long n, long L) {
+  const long total = L * n;
+  for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
+       t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
+    T acc = T(0);
+    for (int p = 0; p < nparts; ++p) {
+      acc += parts[(l * nparts + p) * n + e];
+    }
+    out[t] = acc;
+  }
+}
+
 template <typename T, int MODE, bool WITH_PENALTY>
 __global__ void fused_step_k(
     T* __restrict__ theta, const T* __restrict__ grad,
@@ -154,14 +173,23 @@ __global__ void fused_step_k(
     const int* __restrict__ deg,  // [L], null unless WITH_PENALTY
     T* __restrict__ m, T* __restrict__ v,  // Adam state (null for SGD)
     T rho, T lr, T beta1, T beta2, T eps, T wd, T bc1, T bc2,
-    int first_step, long n, long L) {
+    int first_step, int nparts, long n, long L) {
   const long total = L * n;
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
     T th = theta[t];
-    T g = grad[t];
+    T g;
+    if (nparts > 1) {  // per-tile slabs from the fused train step
+      g = T(0);
+      for (int p = 0; p < nparts; ++p) {
+        g += grad[(l * nparts + p) * n + e];
+      }
+    } else {
+      g = grad[t];
+    }
     if (WITH_PENALTY) {
-      const long l = t / n;
       g += dual[t] + T(2) * rho * (T(deg[l]) * th - s[t]);
     }
     if (MODE == 2) {  // SGD
@@ -219,7 +247,7 @@ __global__ void fused_step_sched_k(
     const T* __restrict__ dual, const T* __restrict__ s,
     const int* __restrict__ deg, T* __restrict__ m, T* __restrict__ v,
     const T* __restrict__ sched, int pit,
-    T beta1, T beta2, T eps, T wd, int first_step, long n, long L) {
+    T beta1, T beta2, T eps, T wd, int first_step, int nparts, long n, long L) {
   const T rho = sched[0];
   const T lr = sched[1];
   const T bc1 = sched[2 + 2 * pit];
@@ -227,10 +255,19 @@ __global__ void fused_step_sched_k(
   const long total = L * n;
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
+    const long l = t / n;
+    const long e = t - l * n;
     T th = theta[t];
-    T g = grad[t];
+    T g;
+    if (nparts > 1) {  // per-tile slabs from the fused train step
+      g = T(0);
+      for (int p = 0; p < nparts; ++p) {
+        g += grad[(l * nparts + p) * n + e];
+      }
+    } else {
+      g = grad[t];
+    }
     if (WITH_PENALTY) {
-      const long l = t / n;
       g += dual[t] + T(2) * rho * (T(deg[l]) * th - s[t]);
     }
     if (MODE == 2) {

@@ -150,9 +150,10 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
                 c10::optional<torch::Tensor> v,
                 double rho, double lr, double beta1, double beta2,
                 double eps, double wd, long step_t, long mode,
-                bool first_step) {
+                bool first_step, long nparts) {
   CHECK_DEV(theta); CHECK_DEV(grad);
   const long L = theta.size(0), n = theta.size(1);
+  TORCH_CHECK(grad.numel() == L * nparts * n, "grad/nparts mismatch");
   const bool pen = dual.has_value();
   DISPATCH_FT(theta, {
     const scalar_t bc1 =
@@ -172,7 +173,7 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
           v.has_value() ? v->data_ptr<scalar_t>() : nullptr,
           (scalar_t)rho, (scalar_t)lr, (scalar_t)beta1,
           (scalar_t)beta2, (scalar_t)eps, (scalar_t)wd, bc1, bc2,
-          first_step ? 1 : 0, n, L);
+          first_step ? 1 : 0, (int)nparts, n, L);
     };
     using c0 = std::integral_constant<int, 0>;
     using c1 = std::integral_constant<int, 1>;
@@ -213,9 +214,10 @@ void fused_step_sched(torch::Tensor theta, torch::Tensor grad,
                       c10::optional<torch::Tensor> v,
                       torch::Tensor sched, long pit,
                       double beta1, double beta2, double eps, double wd,
-                      long mode, bool first_step) {
+                      long mode, bool first_step, long nparts) {
   CHECK_DEV(theta); CHECK_DEV(grad);
   const long L = theta.size(0), n = theta.size(1);
+  TORCH_CHECK(grad.numel() == L * nparts * n, "grad/nparts mismatch");
   const bool pen = dual.has_value();
   DISPATCH_FT(theta, {
     auto launch = [&](auto mode_c, auto pen_c) {
@@ -231,7 +233,7 @@ void fused_step_sched(torch::Tensor theta, torch::Tensor grad,
           v.has_value() ? v->data_ptr<scalar_t>() : nullptr,
           sched.data_ptr<scalar_t>(), (int)pit,
           (scalar_t)beta1, (scalar_t)beta2, (scalar_t)eps,
-          (scalar_t)wd, first_step ? 1 : 0, n, L);
+          (scalar_t)wd, first_step ? 1 : 0, (int)nparts, n, L);
     };
     using c0 = std::integral_constant<int, 0>;
     using c1 = std::integral_constant<int, 1>;
@@ -285,6 +287,20 @@ void gather_targets_dev(torch::Tensor Y_all, torch::Tensor idx,
           (int)pit, maxlen, B, idx_stride, total);
     });
   }
+  HIP_CHECK_LAST();
+}
+
+void reduce_parts(torch::Tensor parts, torch::Tensor out,
+                  long nparts) {
+  CHECK_DEV(parts); CHECK_DEV(out);
+  const long L = out.size(0), n = out.size(1);
+  TORCH_CHECK(parts.numel() == L * nparts * n, "parts shape mismatch");
+  DISPATCH_FT(out, {
+    hipLaunchKernelGGL(ew::reduce_parts_k<scalar_t>,
+        dim3(grid_1d(L * n)), dim3(ew::BLOCK), 0, cur_stream(),
+        parts.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+        (int)nparts, n, L);
+  });
   HIP_CHECK_LAST();
 }
 
@@ -497,6 +513,9 @@ void mnist_train_step(
     long TI, double loss_scale) {
   CHECK_DEV(X_all); CHECK_DEV(theta); CHECK_DEV(grad);
   const long L = theta.size(0), n = theta.size(1);
+  const long NT = (B + TI - 1) / TI;
+  TORCH_CHECK(grad.numel() == L * NT * n,
+              "grad must be the [L, NT, n] per-tile slab buffer");
   const long maxlen = Y_all.size(1);
   TORCH_CHECK(bc_off == wc_off + F * K * K,
               "conv weight/bias must be contiguous in the flat layout");
@@ -524,7 +543,7 @@ void mnist_train_step(
         (int)pit, idx_off, idx_stride, maxlen, n,
         wc_off, bc_off, w1_off, b1_off, w2_off, b2_off,
         (int)B, (int)F, (int)K, (int)IMG, (int)H, (int)C, (int)TI,
-        (scalar_t)loss_scale);
+        (int)NT, (scalar_t)loss_scale);
   });
   HIP_CHECK_LAST();
 }
@@ -626,6 +645,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("dsgt_y_update", &dsgt_y_update);
   mod.def("fused_step", &fused_step);
   mod.def("axpy", &axpy);
+  mod.def("reduce_parts", &reduce_parts);
   mod.def("gather_batch", &gather_batch);
   mod.def("gather_batch_dev", &gather_batch_dev);
   mod.def("gather_targets_dev", &gather_targets_dev);

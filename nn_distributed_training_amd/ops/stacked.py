@@ -371,23 +371,40 @@ class StackedEngine:
 
     def run_fused_mnist(self, off=0, graph_offs=None, pit=0,
                         want_loss=False):
+        """One-launch train step writing per-tile gradient SLABS
+        (plain stores — no atomics, no pre-zero); the fused optimizer
+        step reduces the slabs on the fly (nparts), other consumers
+        call reduce_fused_grad()."""
         conv, fc1, fc2 = self.spec.layers
-        self.grad.zero_()  # all three layers accumulate atomically
+        ti = int(os.environ.get("NDTA_FUSED_TI", "8"))
+        nt = (self.B + ti - 1) // ti
+        if (getattr(self, "grad_parts", None) is None
+                or self.grad_parts.shape[1] != nt):
+            self.grad_parts = torch.empty(
+                self.L, nt, self.n, device=self.device,
+                dtype=self.dtype,
+            )
+        self.fused_nparts = nt
         loss_buf = None
         if want_loss:
             if self._bufs is None:
                 self._bufs = self._alloc_bufs()
             self._bufs["loss"].zero_()
             loss_buf = self._bufs["loss"]
-        ti = int(os.environ.get("NDTA_FUSED_TI", "8"))
         self.ext.mnist_train_step(
             self.X_all, self.Y_all, self.sampler.stream, graph_offs,
-            self.theta, self.grad, loss_buf, pit, off, self.sampler.S,
+            self.theta, self.grad_parts, loss_buf, pit, off,
+            self.sampler.S,
             conv.w_off, conv.b_off, fc1.w_off, fc1.b_off, fc2.w_off,
             fc2.b_off, self.B, conv.out_dim, conv.kernel_size,
             conv.in_dim, fc1.out_dim, fc2.out_dim, ti, 1.0,
         )
         return loss_buf
+
+    def reduce_fused_grad(self):
+        self.ext.reduce_parts(
+            self.grad_parts, self.grad, self.fused_nparts
+        )
 
     def fused_advance(self):
         """Advance the sampler for a fused step; returns the offset."""
@@ -789,6 +806,7 @@ class DiNNOStackedDriver:
                 eng.run_fused_mnist(
                     graph_offs=self._offs_dev, pit=pit
                 )
+                grad_t, nparts = eng.grad_parts, eng.fused_nparts
             else:
                 ext.gather_batch_dev(
                     eng.X_all, eng.sampler.stream, xb, self._offs_dev,
@@ -798,13 +816,14 @@ class DiNNOStackedDriver:
                 eng.backward(
                     xb, None, graph_offs=self._offs_dev, pit=pit
                 )
+                grad_t, nparts = eng.grad, 1
             first = (not self.persistent) and pit == 0
             ext.fused_step_sched(
-                eng.theta, eng.grad, self.duals, self.s, deg,
+                eng.theta, grad_t, self.duals, self.s, deg,
                 None if self.mode == 2 else self.m,
                 None if self.mode == 2 else self.v,
                 self._sched_dev, pit, 0.9, 0.999, 1e-8, self.wd,
-                self.mode, first,
+                self.mode, first, nparts,
             )
 
     def _step_round_graph(self, k):
@@ -904,6 +923,7 @@ class DiNNOStackedDriver:
                 with _timer("train_step"):
                     off = eng.fused_advance()
                     lb = eng.run_fused_mnist(off=off, want_loss=wl)
+                grad_t, nparts = eng.grad_parts, eng.fused_nparts
             else:
                 with _timer("next_batch"):
                     xb, yb = eng.next_batch()
@@ -911,16 +931,17 @@ class DiNNOStackedDriver:
                     eng.forward(xb, train_skip_logp=True)
                 with _timer("backward"):
                     lb = eng.backward(xb, yb, want_loss=wl)
+                grad_t, nparts = eng.grad, 1
             if wl and lb is not None:
                 eng.update_tloss(lb)
             with _timer("fused_step"):
                 self.step_t += 1
                 ext.fused_step(
-                    eng.theta, eng.grad, self.duals, self.s, deg,
+                    eng.theta, grad_t, self.duals, self.s, deg,
                     None if self.mode == 2 else self.m,
                     None if self.mode == 2 else self.v,
                     self.rho, lr, 0.9, 0.999, 1e-8, self.wd,
-                    self.step_t, self.mode, self.step_t == 1,
+                    self.step_t, self.mode, self.step_t == 1, nparts,
                 )
 
     def _opt_state(self):
@@ -988,6 +1009,7 @@ class DSGDStackedDriver:
         if eng.fused_step_available():
             off = eng.fused_advance()
             lb = eng.run_fused_mnist(off=off, want_loss=want_tl)
+            eng.reduce_fused_grad()
         else:
             xb, yb = eng.next_batch()
             eng.forward(xb, train_skip_logp=True)
@@ -1028,6 +1050,7 @@ class DSGTStackedDriver:
         if self.opt.conf["init_grads"]:
             if eng.fused_step_available():
                 eng.run_fused_mnist(off=eng.fused_advance())
+                eng.reduce_fused_grad()
             else:
                 xb, yb = eng.next_batch()
                 eng.forward(xb, train_skip_logp=True)
@@ -1083,6 +1106,7 @@ class DSGTStackedDriver:
             with _timer("train_step"):
                 off = eng.fused_advance()
                 lb = eng.run_fused_mnist(off=off, want_loss=want_tl)
+                eng.reduce_fused_grad()
         else:
             with _timer("next_batch"):
                 xb, yb = eng.next_batch()

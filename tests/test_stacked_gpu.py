@@ -247,6 +247,7 @@ def test_fused_mnist_step_matches_layered(dtype, monkeypatch):
         if fused:
             off = eng.fused_advance()
             eng.run_fused_mnist(off=off)
+            eng.reduce_fused_grad()
         else:
             xb, yb = eng.next_batch()
             eng.forward(xb, train_skip_logp=True)
