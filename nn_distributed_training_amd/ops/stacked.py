@@ -359,9 +359,15 @@ class StackedEngine:
     # ------------------------------------------------------------------
     # fully fused MNIST train step (ops/hip/fused_mnist.hip): one launch
     # per primal iteration instead of the ~12-kernel layered chain.
-    # NDTA_FUSED=0 falls back to the layered path (A/B lever).
+    # DEFAULT OFF — measured SLOWER than the layered path on MI355X
+    # (best 2263 vs 3650 rounds/s at the 8-node bench): per-tile fusion
+    # re-streams fc1's 221KB weight panel per block (cost grows as tiles
+    # shrink) while large tiles leave 3/4 of the 256-CU chip idle at
+    # 1 block/CU. The layered GEMMs share weight panels across the whole
+    # batch and fill the chip. Kept as a validated alternative
+    # (NDTA_FUSED=1; numerics-tested against the layered path).
     def fused_step_available(self) -> bool:
-        if os.environ.get("NDTA_FUSED", "1") != "1":
+        if os.environ.get("NDTA_FUSED", "0") != "1":
             return False
         if self.spec.name != "mnist_conv" or not self.classification:
             return False
