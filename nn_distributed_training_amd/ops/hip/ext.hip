@@ -493,11 +493,21 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
   // (LDS-staged image variant was 2x slower — profiles/README.md)
   const int nchunk = (int)std::min<long>(B, 32);
   DISPATCH_FT(dY, {
-    hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
-        dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
-        dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
-        X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
-        n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
+    // KMAX sizes the per-thread dW register block: the K<=5 variant
+    // saves 48 VGPRs over the generic K<=7 one (occupancy)
+    if (K <= 5) {
+      hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 5>),
+          dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
+          dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
+    } else {
+      hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
+          dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
+          dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
+    }
   });
   HIP_CHECK_LAST();
 }
