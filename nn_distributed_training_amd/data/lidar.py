@@ -276,6 +276,10 @@ class OnlineTrajectoryLidarDataset(torch.utils.data.Dataset):
         self.curr_scan_idx = 0
         self.curr_pos = self.scan_locs[0]
         self.curr_idx_list: list = []
+        # dataset-local RNG: window shuffles must not depend on global
+        # random-module state (which diverges with rank packing — the
+        # same node must draw the same stream on any rank)
+        self._shuffle_rng = random.Random(0xD5)
         self._advance_window()
 
     def _advance_window(self):
@@ -295,7 +299,7 @@ class OnlineTrajectoryLidarDataset(torch.utils.data.Dataset):
         if self.curr_idx_list is None:
             lb, ub = self.window_bounds
             self.curr_idx_list = list(range(lb, ub))
-            random.shuffle(self.curr_idx_list)
+            self._shuffle_rng.shuffle(self.curr_idx_list)
         return self.curr_idx_list
 
     # the reference pops shuffled indices until the window empties, then

@@ -128,3 +128,92 @@ def test_two_rank_matches_single_process(alg, tmp_path):
     dist_stack = torch.stack([pieces[i] for i in range(N_NODES)])
 
     torch.testing.assert_close(dist_stack, golden, rtol=0, atol=0)
+
+
+# ----------------------------------------------------------------------
+# dynamic-graph multirank coverage: the online density problem's
+# position all-gather + per-round disk-graph rebuild must produce
+# identical schedules (and therefore identical parameters) on 2 ranks
+def _run_online(alg="dsgd"):
+    import numpy as np
+
+    from nn_distributed_training_amd.data.floorplan import (
+        synthetic_floorplan,
+        synthetic_waypoints,
+    )
+    from nn_distributed_training_amd.data.lidar import (
+        Lidar2D,
+        OnlineTrajectoryLidarDataset,
+        RandomPoseLidarDataset,
+    )
+    from nn_distributed_training_amd.models import FourierNet
+    from nn_distributed_training_amd.problems.dist_online_dense_problem \
+        import DistOnlineDensityProblem
+
+    torch.set_default_dtype(torch.float64)
+    torch.manual_seed(3)
+    np.random.seed(3)
+    import random as _random
+
+    _random.seed(3)
+    img = synthetic_floorplan(nx=96, ny=96, num_walls=3,
+                              border_width=10, seed=0)
+    lidar = Lidar2D(img, 6, 0.25, 8, 1.0, 20, 3)
+    wps = synthetic_waypoints(img, 4, seed=1)
+    sets = [
+        OnlineTrajectoryLidarDataset(lidar, wp, 2, 2) for wp in wps
+    ]
+    val = RandomPoseLidarDataset(lidar, 4)
+    conf = {
+        "problem_name": "odense",
+        "train_batch_size": 32,
+        "val_batch_size": 64,
+        "comm_radius": 500.0,
+        "dynamic_graph": True,
+        "save_models": False,
+        "data_seed": 9,
+        "verbose_evals": False,
+        "metrics": ["consensus_error"],
+        "metrics_config": {"evaluate_frequency": 1000},
+        "optimizer_config": {
+            "alg_name": alg, "outer_iterations": 4, "alpha0": 0.002,
+            "mu": 0.001, "profile": False,
+        },
+    }
+    pr = DistOnlineDensityProblem(
+        FourierNet([2, 12, 6, 1], scale=0.05), torch.nn.BCELoss(),
+        sets, val, torch.device("cpu"), conf,
+    )
+    dopt = build_optimizer(pr, pr.device, conf["optimizer_config"])
+    dopt.train()
+    return pr.local_params_stack(), pr.local_nodes
+
+
+def _online_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        stack, nodes = _run_online()
+        with open(os.path.join(out_dir, f"od{rank}.pkl"), "wb") as f:
+            pickle.dump((list(nodes), stack.numpy()), f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_online_density_two_rank_matches_single(tmp_path):
+    golden, nodes = _run_online()
+    assert list(nodes) == list(range(4))
+
+    mp.start_processes(
+        _online_worker, args=(2, 29725, str(tmp_path)), nprocs=2,
+        join=True, start_method="spawn",
+    )
+    pieces = {}
+    for r in range(2):
+        with open(tmp_path / f"od{r}.pkl", "rb") as f:
+            local_nodes, stack = pickle.load(f)
+        for li, i in enumerate(local_nodes):
+            pieces[i] = torch.from_numpy(stack[li])
+    dist_stack = torch.stack([pieces[i] for i in range(4)])
+    torch.testing.assert_close(dist_stack, golden, rtol=0, atol=0)
