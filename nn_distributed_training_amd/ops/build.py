@@ -22,11 +22,14 @@ def build(verbose: bool = False):
     os.makedirs(_BUILD_DIR, exist_ok=True)
     from torch.utils.cpp_extension import load
 
+    extra = ["-O3", "--offload-arch=gfx950"]
+    if os.environ.get("NDTA_BUILD_DEFINES"):
+        extra += os.environ["NDTA_BUILD_DEFINES"].split()
     return load(
         name=_EXT_NAME,
         sources=_SRC,
         build_directory=_BUILD_DIR,
-        extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
+        extra_cuda_cflags=extra,
         extra_cflags=["-O3"],
         verbose=verbose,
     )
