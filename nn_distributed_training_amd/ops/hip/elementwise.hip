@@ -179,6 +179,11 @@ __global__ void fused_step_k(
        t += (long)gridDim.x * BLOCK) {
     const long l = t / n;
     const long e = t - l * n;
+    // isolated node under a dynamic graph: the golden engine skips the
+    // whole primal update (the reference crashes on torch.stack of an
+    // empty neighbor list, so "frozen" is this framework's defined
+    // behavior) — freeze theta AND the Adam moments to match exactly
+    if (WITH_PENALTY && deg[l] == 0) continue;
     T th = theta[t];
     T g;
     if (nparts > 1) {  // per-tile slabs from the fused train step
@@ -257,6 +262,8 @@ __global__ void fused_step_sched_k(
        t += (long)gridDim.x * BLOCK) {
     const long l = t / n;
     const long e = t - l * n;
+    // freeze isolated nodes (see fused_step_k)
+    if (WITH_PENALTY && deg[l] == 0) continue;
     T th = theta[t];
     T g;
     if (nparts > 1) {  // per-tile slabs from the fused train step

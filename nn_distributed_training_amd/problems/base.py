@@ -151,5 +151,8 @@ class ProblemBase:
         torch.save(self.metrics, os.path.join(output_dir, file_name))
 
     def _print(self, msg):
-        if self.is_root:
+        # honors the documented `verbose_evals` switch (reference
+        # README.md:176 documents it but problems/*.py:210 prints
+        # unconditionally — we follow the documentation)
+        if self.is_root and self.conf.get("verbose_evals", True):
             print(msg, flush=True)

@@ -145,6 +145,50 @@ def test_stacked_matches_golden(alg, dtype):
 
 
 @requires_gpu
+def test_stacked_dinno_isolated_node_matches_golden():
+    """A degree-0 node must stay FROZEN in both engines.
+
+    The golden engine skips the whole primal update for a node with no
+    neighbors (the reference crashes there: optimizers/dinno.py:121
+    torch.stack of an empty list); the fused-step kernel freezes
+    deg==0 rows to match. Dynamic disk graphs can produce isolation,
+    so this is a real runtime case, not just an API corner.
+    """
+    torch.set_default_dtype(torch.float64)
+    conf = _conf(copy.deepcopy(ALG_CONFS["dinno"]))
+
+    def _build_isolated(conf):
+        torch.manual_seed(11)
+        graph = nx.path_graph(N_NODES - 1)  # 0-1-2 chain
+        graph.add_node(N_NODES - 1)         # node 3 isolated
+        train = SyntheticMNIST(PER_NODE * N_NODES, seed=0)
+        val = SyntheticMNIST(64, seed=1)
+        subsets = split_train_set(train, N_NODES, "random")
+        base_model = MNISTConvNet(3, 5, 64)
+        return DistMNISTProblem(
+            graph, base_model, torch.nn.NLLLoss(), subsets, val,
+            torch.device("cuda"), conf,
+        )
+
+    pr_g = _build_isolated(conf)
+    init = pr_g.local_params_stack().clone()
+    opt_g = build_optimizer(pr_g, pr_g.device, conf["optimizer_config"])
+    opt_g.train()
+    golden = pr_g.local_params_stack()
+    # golden really froze the isolated node (guards the premise)
+    torch.testing.assert_close(golden[-1], init[-1], rtol=0, atol=0)
+    assert not torch.allclose(golden[0], init[0])
+
+    pr_s = _build_isolated(conf)
+    pr_s.stacked = StackedEngine(pr_s)
+    opt_s = build_optimizer(pr_s, pr_s.device, conf["optimizer_config"])
+    opt_s.train()
+    torch.testing.assert_close(
+        pr_s.stacked.theta, golden, rtol=1e-8, atol=1e-8
+    )
+
+
+@requires_gpu
 @pytest.mark.parametrize("alg", ["dinno", "dsgt"])
 def test_stacked_density_matches_golden(alg):
     """FourierNet + BCE density problem: stacked engine vs golden torch
