@@ -181,21 +181,7 @@ class Communicator:
             return {}
 
         r = self.rank
-        # nodes I must receive: remote neighbor j of a local node i
-        # nodes I must send: local node i that has a remote neighbor on peer p
-        recv_nodes = set()
-        send_pairs = set()  # (local node, peer rank)
-        for a, b in edges:
-            ra, rb = layout.rank_of(a), layout.rank_of(b)
-            if ra == rb:
-                continue
-            if ra == r:
-                recv_nodes.add(b)
-                send_pairs.add((a, rb))
-            elif rb == r:
-                recv_nodes.add(a)
-                send_pairs.add((b, ra))
-
+        send_pairs, recv_nodes = self.edge_transfers(layout, edges)
         n = local_stack.shape[1]
         recv_bufs: Dict[int, torch.Tensor] = {
             j: torch.empty(n, dtype=local_stack.dtype, device=self.device)
