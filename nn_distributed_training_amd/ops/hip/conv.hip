@@ -12,8 +12,9 @@ namespace conv {
 // X: [L*B, 28*28]; theta: [L, n] flat stack, W at w_off ([F,1,k,k]
 // row-major = [F, k*k]), b at b_off; Y: [L*B, F*P*P] pooled+ReLU output;
 // idx: [L*B, F*P*P] argmax position (0..3) for pool backward.
-// Grid: one block per image, 256 threads.
-template <typename T>
+// Grid: one block per image, 256 threads. KMAX bounds the tap loops at
+// compile time (K<=KMAX checked by the caller) so they fully unroll.
+template <typename T, int KMAX>
 __global__ void conv_pool_fwd_k(
     const T* __restrict__ X, const T* __restrict__ theta,
     T* __restrict__ Y, unsigned char* __restrict__ idx,
@@ -56,10 +57,14 @@ __global__ void conv_pool_fwd_k(
       const int cy = 2 * py + (d >> 1);
       const int cx = 2 * px + (d & 1);
       T acc = bias;
-      for (int ky = 0; ky < K; ++ky) {
+#pragma unroll
+      for (int ky = 0; ky < KMAX; ++ky) {
+        if (ky >= K) break;
         const T* row = img + (cy + ky) * IMG + cx;
         const T* wr = wf + ky * K;
-        for (int kx = 0; kx < K; ++kx) {
+#pragma unroll
+        for (int kx = 0; kx < KMAX; ++kx) {
+          if (kx >= K) break;
           acc += row[kx] * wr[kx];
         }
       }
@@ -115,10 +120,18 @@ __global__ void conv_pool_bwd_k(
     const int cx = 2 * px + (d & 1);
     const T* img = X + lb * IMG * IMG;
     db += g;
-    for (int ky = 0; ky < K; ++ky) {
+    // compile-time bounds + KMAX-strided indices: a runtime `ky*K+kx`
+    // subscript makes dw[] dynamically indexed and the compiler spills
+    // the whole accumulator to scratch (guide §5.4 rule 20 — the same
+    // trap dw_skinny_*_k hit); this kernel was 45 us/call that way
+#pragma unroll
+    for (int ky = 0; ky < KMAX; ++ky) {
+      if (ky >= K) break;
       const T* row = img + (cy + ky) * IMG + cx;
-      for (int kx = 0; kx < K; ++kx) {
-        dw[ky * K + kx] += g * row[kx];
+#pragma unroll
+      for (int kx = 0; kx < KMAX; ++kx) {
+        if (kx >= K) break;
+        dw[ky * KMAX + kx] += g * row[kx];
       }
     }
   }
@@ -128,17 +141,21 @@ __global__ void conv_pool_bwd_k(
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   T* wslice = gstack + (long)l * n + w_off + (long)f * K * K;
-  for (int i = 0; i < K * K + 1; ++i) {
-    T v = (i < K * K) ? dw[i] : db;
+#pragma unroll
+  for (int i = 0; i < KMAX * KMAX + 1; ++i) {
+    const int ky = i / KMAX, kx = i % KMAX;
+    const bool is_db = (i == KMAX * KMAX);
+    if (!is_db && (ky >= K || kx >= K)) continue;  // uniform across block
+    T v = is_db ? db : dw[i];
     v = wave_reduce_sum(v);
     if (lane == 0) red[wid] = v;
     __syncthreads();
     if (threadIdx.x == 0) {
       T tot = red[0] + red[1] + red[2] + red[3];
-      if (i < K * K) {
-        atomicAdd(&wslice[i], tot);
-      } else {
+      if (is_db) {
         atomicAdd(&gstack[(long)l * n + b_off + f], tot);
+      } else {
+        atomicAdd(&wslice[ky * K + kx], tot);
       }
     }
     __syncthreads();

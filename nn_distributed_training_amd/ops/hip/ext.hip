@@ -470,14 +470,23 @@ void conv_pool_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
                    long F, long K, long IMG) {
   CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
   const long L = theta.size(0), n = theta.size(1);
+  TORCH_CHECK(K <= 7, "conv_pool_fwd supports kernel size <= 7");
   DISPATCH_FT(X, {
     const size_t shmem =
         (IMG * IMG + F * K * K + F) * sizeof(scalar_t);
-    hipLaunchKernelGGL(conv::conv_pool_fwd_k<scalar_t>,
-        dim3(L * B), dim3(256), shmem, cur_stream(),
-        X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
-        Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
-        n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+    if (K <= 5) {
+      hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 5>),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
+          X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+    } else {
+      hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 7>),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
+          X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+    }
   });
   HIP_CHECK_LAST();
 }

@@ -141,9 +141,13 @@ def test_linear_bwd(ext, dtype, L, M, I, O):
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
-def test_conv_pool_fwd_bwd(ext, dtype):
+@pytest.mark.parametrize("K", [3, 5, 7])
+def test_conv_pool_fwd_bwd(ext, dtype, K):
+    # K=5/7 hit the exact KMAX template instantiations; K=3 exercises
+    # the K < KMAX case (early-broken unrolled loops, KMAX-strided
+    # register accumulators)
     torch.manual_seed(3)
-    L, B, F, K, IMG = 2, 5, 3, 5, 28
+    L, B, F, IMG = 2, 5, 3, 28
     P = (IMG - K + 1) // 2
     n = F * K * K + F
     dev = _dev()
