@@ -136,8 +136,12 @@ __global__ void conv_pool_bwd_k(
     }
   }
 
-  // reduce the K*K+1 partials: shuffle within waves, LDS across waves
-  __shared__ T red[4];  // one slot per wave (256 threads = 4 waves)
+  // reduce the K*K+1 partials: shuffle-reduce every tap within its
+  // wave (no barriers), park the per-wave sums in LDS, then ONE
+  // barrier and the first KMAX*KMAX+1 threads finish their tap in
+  // parallel. The previous per-tap LDS tree spent 2 barriers per tap
+  // (52 for K=5) and dominated the kernel.
+  __shared__ T red[(KMAX * KMAX + 1) * 4];  // [tap][wave]
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   T* wslice = gstack + (long)l * n + w_off + (long)f * K * K;
@@ -145,20 +149,22 @@ __global__ void conv_pool_bwd_k(
   for (int i = 0; i < KMAX * KMAX + 1; ++i) {
     const int ky = i / KMAX, kx = i % KMAX;
     const bool is_db = (i == KMAX * KMAX);
-    if (!is_db && (ky >= K || kx >= K)) continue;  // uniform across block
-    T v = is_db ? db : dw[i];
-    v = wave_reduce_sum(v);
-    if (lane == 0) red[wid] = v;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      T tot = red[0] + red[1] + red[2] + red[3];
-      if (is_db) {
-        atomicAdd(&gstack[(long)l * n + b_off + f], tot);
-      } else {
-        atomicAdd(&wslice[ky * K + kx], tot);
-      }
+    if (!is_db && (ky >= K || kx >= K)) continue;  // uniform
+    const T v = wave_reduce_sum(is_db ? db : dw[i]);
+    if (lane == 0) red[i * 4 + wid] = v;
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < KMAX * KMAX + 1; i += blockDim.x) {
+    const int ky = i / KMAX, kx = i % KMAX;
+    const bool is_db = (i == KMAX * KMAX);
+    if (!is_db && (ky >= K || kx >= K)) continue;
+    const T tot =
+        red[i * 4] + red[i * 4 + 1] + red[i * 4 + 2] + red[i * 4 + 3];
+    if (is_db) {
+      atomicAdd(&gstack[(long)l * n + b_off + f], tot);
+    } else {
+      atomicAdd(&wslice[ky * K + kx], tot);
     }
-    __syncthreads();
   }
 }
 
