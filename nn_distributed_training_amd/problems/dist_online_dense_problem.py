@@ -180,7 +180,7 @@ class DistOnlineDensityProblem(ProblemBase):
                     )
                 val_losses = self.gather_per_node(vl)
                 self.metrics[met_name].append(val_losses)
-                evalprint += "Val Loss: {:.4f} - {:.4} - {:.4f} | ".format(
+                evalprint += "Val Loss: {:.4f} - {:.4f} - {:.4f} | ".format(
                     val_losses.amin().item(),
                     val_losses.mean().item(),
                     val_losses.amax().item(),
