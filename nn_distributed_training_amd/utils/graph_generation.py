@@ -119,9 +119,11 @@ def disk_with_fied(N: int, targ: float, num_restarts: int = 50) -> nx.Graph:
         if abs(ubf - targ) < tol:
             return nx.random_geometric_graph(N, ubr, pos=pos)
         if not ubf > lbf:
-            raise NameError(
-                f"Degenerate Fiedler bounds: lb={lbf} ub={ubf}"
-            )
+            # disconnected even at the outer radius (can happen for any
+            # N on an unlucky position draw, and always for tiny N when
+            # two points land far apart) — try a fresh draw rather than
+            # aborting the whole sweep
+            continue
         if targ > ubf or targ < lbf:
             # target not bracketed for this position draw; try a new draw
             continue

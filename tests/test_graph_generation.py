@@ -63,6 +63,17 @@ def test_delaunay():
 
 
 @pytest.mark.slow
+def test_disk_with_fied_unreachable_target_raises_cleanly():
+    # N=2 only ever has Fiedler 0 (disconnected) or 2 (one edge): a
+    # target of 1.0 is unreachable. The bisection must retry draws and
+    # fail with its terminal error, not die on a degenerate bracket.
+    import random
+
+    random.seed(0)
+    with pytest.raises(NameError, match="Never found"):
+        gg.disk_with_fied(2, 1.0, num_restarts=5)
+
+
 def test_disk_with_fied():
     g = gg.disk_with_fied(12, 1.0)
     fied = nx.linalg.algebraic_connectivity(g, tol=1e-3, method="lanczos")
