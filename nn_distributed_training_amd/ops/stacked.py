@@ -472,7 +472,14 @@ class StackedEngine:
                         layer.w_off, layer.b_off, M, layer.in_dim,
                         layer.out_dim, ACT_IDS[act], layer.scale,
                     )
-            cur = out if layer.activation != "logsoftmax" else bufs["logp"]
+            # under train_skip_logp the logp buffer was never written —
+            # hand back the logits (the training path consumes those
+            # through the fused NLL kernel anyway)
+            cur = (
+                out
+                if layer.activation != "logsoftmax" or train_skip_logp
+                else bufs["logp"]
+            )
         return cur
 
     # ------------------------------------------------------------------
@@ -723,7 +730,16 @@ def _edge_key(pr):
 
 class DiNNOStackedDriver:
     """DiNNO outer loop with the fused kernels (same math as
-    optimizers/dinno.py)."""
+    optimizers/dinno.py).
+
+    Isolated nodes (possible under dynamic graphs) are frozen whole —
+    theta, Adam moments and duals — matching the golden engine's skip
+    (the reference crashes on them). One deliberate corner: with
+    ``persistant_primal_opt`` the bias-correction step count is global,
+    so a node that reconnects after an isolated stretch sees a larger
+    t than the golden engine's per-node Adam would give it; the
+    non-persistent mode (the paper configs) matches exactly.
+    """
 
     def __init__(self, dinno, pr):
         self.opt = dinno
