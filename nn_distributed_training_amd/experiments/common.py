@@ -22,8 +22,17 @@ from ..parallel.comm import init_from_env
 
 def setup_run(yaml_pth: str, exp_conf: dict, rank: int) -> str:
     """Create the run output dir (rank 0 only) and snapshot the config."""
+    import torch.distributed as dist
+
     output_metadir = exp_conf["output_metadir"]
     time_now = datetime.now().strftime("%Y-%m-%d_%H-%M")
+    if dist.is_available() and dist.is_initialized():
+        # all ranks must agree on the run dir (checkpoint shards live
+        # there); a rank clocking in across a minute boundary must not
+        # invent its own timestamp
+        box = [time_now]
+        dist.broadcast_object_list(box, src=0)
+        time_now = box[0]
     output_dir = os.path.join(output_metadir, time_now + "_" + exp_conf["name"])
     if exp_conf["writeout"] and rank == 0:
         os.makedirs(output_metadir, exist_ok=True)
