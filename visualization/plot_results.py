@@ -156,14 +156,76 @@ def plot_rl_dir(rl_dir: str, out_dir: str):
     print(f"RL figures -> {out_dir}")
 
 
+def animate_run_dir(run_dir: str, out_dir: str, node: int = 0,
+                    fps: int = 8):
+    """Render the per-eval mesh frames of an anim-config run.
+
+    Covers the reference's visualization/animations/{density_anim,
+    mnist_anim}.ipynb role: for every ``<problem>_results.pt`` whose
+    metrics carry per-eval ``mesh_grid_density`` frames (configs with
+    ``mesh_only_at_end: false``, e.g. configs/dist_online_dense_anim.
+    yaml), write one PNG per evaluation for ``node`` (robot positions
+    overlaid when ``current_position`` was recorded) plus an animated
+    GIF. Frames hold the per-rank LOCAL nodes — run the anim configs
+    single-rank for all-node frames.
+    """
+    os.makedirs(out_dir, exist_ok=True)
+    for res in sorted(glob.glob(os.path.join(run_dir, "*_results.pt"))):
+        name = os.path.basename(res)[: -len("_results.pt")]
+        metrics = torch.load(res, map_location="cpu", weights_only=False)
+        frames = [
+            f for f in metrics.get("mesh_grid_density", [])
+            if f.numel() > 0
+        ]
+        if len(frames) < 2:
+            continue  # PAPER configs only record the final mesh
+        positions = metrics.get("current_position", [])
+        side = int(round(frames[0].shape[1] ** 0.5))
+        fdir = os.path.join(out_dir, f"{name}_frames")
+        os.makedirs(fdir, exist_ok=True)
+        paths = []
+        for t, fr in enumerate(frames):
+            img = fr[node].reshape(side, side).numpy()
+            fig, ax = plt.subplots(figsize=(4, 4))
+            ax.imshow(img, origin="lower", cmap="viridis",
+                      vmin=0.0, vmax=1.0)
+            if t < len(positions) and positions[t] is not None:
+                pos = np.asarray(positions[t]) / 8.0  # mesh is ::8
+                ax.scatter(pos[:, 0], pos[:, 1], c="red", s=12,
+                           marker="o")
+            ax.set_title(f"{name} node {node} eval {t}")
+            ax.set_axis_off()
+            fig.tight_layout()
+            path = os.path.join(fdir, f"{t:04d}.png")
+            fig.savefig(path, dpi=110)
+            plt.close(fig)
+            paths.append(path)
+        try:
+            from PIL import Image
+
+            ims = [Image.open(p) for p in paths]
+            gif = os.path.join(out_dir, f"{name}_mesh.gif")
+            ims[0].save(gif, save_all=True, append_images=ims[1:],
+                        duration=int(1000 / fps), loop=0)
+            print(f"{len(paths)} frames -> {fdir}, gif -> {gif}")
+        except ImportError:
+            print(f"{len(paths)} frames -> {fdir} (no pillow: gif "
+                  "skipped)")
+
+
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
     p.add_argument("run_dir", nargs="?", default=None)
     p.add_argument("--rl", default=None)
+    p.add_argument("--animate", default=None,
+                   help="run dir of an anim config; renders mesh frames")
+    p.add_argument("--node", type=int, default=0)
     p.add_argument("--out", default="./figs")
     p.add_argument("--eval-every", type=int, default=20)
     args = p.parse_args()
     if args.rl:
         plot_rl_dir(args.rl, args.out)
+    if args.animate:
+        animate_run_dir(args.animate, args.out, node=args.node)
     if args.run_dir:
         plot_run_dir(args.run_dir, args.out, args.eval_every)
