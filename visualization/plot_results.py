@@ -157,7 +157,7 @@ def plot_rl_dir(rl_dir: str, out_dir: str):
 
 
 def animate_run_dir(run_dir: str, out_dir: str, node: int = 0,
-                    fps: int = 8):
+                    fps: int = 8, world_size: float = 256.0):
     """Render the per-eval mesh frames of an anim-config run.
 
     Covers the reference's visualization/animations/{density_anim,
@@ -184,13 +184,15 @@ def animate_run_dir(run_dir: str, out_dir: str, node: int = 0,
         fdir = os.path.join(out_dir, f"{name}_frames")
         os.makedirs(fdir, exist_ok=True)
         paths = []
+        half = world_size / 2.0  # lidar world box is [-nx/2, nx/2]
         for t, fr in enumerate(frames):
             img = fr[node].reshape(side, side).numpy()
             fig, ax = plt.subplots(figsize=(4, 4))
             ax.imshow(img, origin="lower", cmap="viridis",
-                      vmin=0.0, vmax=1.0)
+                      vmin=0.0, vmax=1.0,
+                      extent=[-half, half, -half, half])
             if t < len(positions) and positions[t] is not None:
-                pos = np.asarray(positions[t]) / 8.0  # mesh is ::8
+                pos = np.asarray(positions[t])  # world coords
                 ax.scatter(pos[:, 0], pos[:, 1], c="red", s=12,
                            marker="o")
             ax.set_title(f"{name} node {node} eval {t}")
@@ -220,12 +222,15 @@ if __name__ == "__main__":
     p.add_argument("--animate", default=None,
                    help="run dir of an anim config; renders mesh frames")
     p.add_argument("--node", type=int, default=0)
+    p.add_argument("--world-size", type=float, default=256.0,
+                   help="lidar world box side (= floorplan_size)")
     p.add_argument("--out", default="./figs")
     p.add_argument("--eval-every", type=int, default=20)
     args = p.parse_args()
     if args.rl:
         plot_rl_dir(args.rl, args.out)
     if args.animate:
-        animate_run_dir(args.animate, args.out, node=args.node)
+        animate_run_dir(args.animate, args.out, node=args.node,
+                        world_size=args.world_size)
     if args.run_dir:
         plot_run_dir(args.run_dir, args.out, args.eval_every)
