@@ -90,7 +90,11 @@ def split_train_set(train_set: Dataset, N: int, split_type: str):
     Parity with the reference driver (experiments/dist_mnist_ex.py:107-127):
       random — N equal random subsets;
       hetero — the 10 digit classes partitioned contiguously over nodes
-               (requires N <= 10);
+               (requires N <= 10). Quirk kept for parity: with N not
+               dividing 10, the trailing `10 - N*(10//N)` classes are
+               assigned to NO node (the reference's torch.split +
+               first-N-chunks loop does the same — its paper configs
+               use N = 10);
       hetero_sorted — label-sorted chunks for arbitrary N (the scaling
                driver's variant, experiments/dist_mnist_scaling.py:122-129).
     """
