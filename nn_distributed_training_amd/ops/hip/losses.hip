@@ -2,7 +2,11 @@
 // layer's PRE-activation in one pass (log-softmax+NLL and sigmoid+BCE
 // use their combined analytic forms), plus an optional per-node loss
 // value (for the train-loss EMA metric) — no autograd, no separate
-// softmax/log/clamp launches.
+// softmax/log/clamp launches. Loss semantics follow the reference's
+// problem layer: NLLLoss on log-softmax outputs (reference
+// problems/dist_mnist_problem.py:65-98), BCELoss on sigmoid outputs
+// and MSE/L1 regression (dist_dense_problem.py / experiments'
+// make_loss choices).
 
 #include "common.h"
 

@@ -3,7 +3,10 @@
 Turns the communication graph + a local [L, n] snapshot (optionally with
 per-node auxiliary vectors concatenated, e.g. DSGT's gradient tracker)
 into per-node neighbor stacks, fetching remote rows through the
-Communicator's batched P2P exchange.
+Communicator's batched P2P exchange. This is the multi-rank equivalent
+of the reference's in-process neighbor reads (`torch.stack([ths[j] for
+j in graph.neighbors(i)])`, reference optimizers/dinno.py:120-122 and
+the plist walks in dsgd.py:37-46 / dsgt.py:58-75).
 """
 
 from __future__ import annotations
