@@ -231,3 +231,47 @@ def test_scaling_driver(tmp_path):
     assert summary[(0, "dinno")]["N"] == 3
     assert summary[(1, "dinno")]["N"] == 5
     assert summary[(0, "dinno")]["rounds_per_sec"] > 0
+
+
+def test_fixed_seed_metric_streams_are_identical(tmp_path):
+    """fixed seed -> identical metric streams across two
+    independent runs of the same driver (the reference's only notion of
+    reproducibility is `torch.manual_seed` at the driver top)."""
+    import glob
+    import pickle
+
+    problems = {
+        "p1": _prob_conf(
+            "dsgd",
+            {
+                "alg_name": "dsgd", "outer_iterations": 6,
+                "alpha0": 0.005, "mu": 0.001, "profile": False,
+            },
+        )
+    }
+
+    def run(sub):
+        base = tmp_path / sub
+        os.makedirs(base)
+        pth = _write_conf(base, problems)
+        dist_mnist_ex.experiment(pth)
+        run_dir = glob.glob(str(base / "out" / "*_tiny_mnist"))[0]
+        res = torch.load(
+            os.path.join(run_dir, "dsgd_results.pt"), weights_only=False
+        )
+        with open(os.path.join(run_dir, "graph.gpickle"), "rb") as f:
+            graph = pickle.load(f)
+        return res, graph
+
+    res_a, g_a = run("a")
+    res_b, g_b = run("b")
+
+    assert sorted(g_a.edges()) == sorted(g_b.edges())
+    assert res_a.keys() == res_b.keys()
+    for key in res_a:
+        for ea, eb in zip(res_a[key], res_b[key]):
+            ta = torch.as_tensor(ea[1] if isinstance(ea, tuple) else ea,
+                                 dtype=torch.float64)
+            tb = torch.as_tensor(eb[1] if isinstance(eb, tuple) else eb,
+                                 dtype=torch.float64)
+            torch.testing.assert_close(ta, tb, rtol=0, atol=0)
