@@ -12,7 +12,13 @@ multi-rank design:
   round's transfers are issued in ONE ``dist.batch_isend_irecv`` group
   (ncclGroupStart/End under RCCL), so RCCL can drive each GPU's 7 xGMI
   links concurrently instead of serializing per edge — and deliberately
-  NOT as a global all-reduce (the algorithms are neighbor-local);
+  NOT as a global all-reduce (the algorithms are neighbor-local).
+  Explicit host-side edge-coloring (SURVEY.md §7 step 4 floated it) is
+  deliberately NOT done: within a grouped launch RCCL already places
+  each peer pair on its own channel/link, every pair here maps to a
+  distinct xGMI link on a single node anyway (≤7 neighbors per GPU),
+  and a hand-built coloring would serialize rounds of the schedule that
+  the grouped launch runs concurrently;
 * same-rank edges never touch the network: the consumer reads the local
   stack row directly;
 * the only collectives are an all-gather of the parameter stacks at
