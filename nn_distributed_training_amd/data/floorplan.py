@@ -19,7 +19,16 @@ def synthetic_floorplan(
     border_width: int = 16,
     seed: int = 0,
 ) -> np.ndarray:
-    """Occupancy image in [0,1], shape [ny, nx]; 1 = wall."""
+    """Occupancy image in [0,1], shape [ny, nx]; 1 = wall.
+
+    Square images only: the vertical-wall branch reuses the x-sampled
+    bounds as row indices (row/col symmetric when nx == ny, and all
+    shipped configs expose a single ``floorplan_size``). Kept as-is so
+    seeded maps stay identical to the committed example/benchmark
+    artifacts.
+    """
+    if nx != ny:
+        raise ValueError("synthetic_floorplan requires nx == ny")
     rng = np.random.default_rng(seed)
     img = np.zeros((ny, nx), dtype=float)
     img[:border_width, :] = 1.0
