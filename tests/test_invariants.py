@@ -1,0 +1,174 @@
+"""Property-based invariants for the math-critical helpers.
+
+Uses hypothesis to sweep shapes the example-based tests fix: Metropolis
+mixing matrices must be symmetric doubly-stochastic for ANY graph,
+NodeLayout must partition nodes for ANY (N, world), and the model specs'
+flat-vector offsets must tile the parameter vector exactly.
+"""
+
+import networkx as nx
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from nn_distributed_training_amd.models import (
+    FFReLUNet,
+    FourierNet,
+    MNISTConvNet,
+)
+from nn_distributed_training_amd.models.spec import model_spec, param_layout
+from nn_distributed_training_amd.parallel.comm import NodeLayout
+from nn_distributed_training_amd.utils import graph_generation
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    n=st.integers(2, 24),
+    p=st.floats(0.2, 1.0),
+    seed=st.integers(0, 10**6),
+)
+def test_metropolis_is_symmetric_doubly_stochastic(n, p, seed):
+    g = nx.erdos_renyi_graph(n, p, seed=seed)
+    W = graph_generation.get_metropolis(g)
+    assert W.shape == (n, n)
+    assert torch.all(W >= 0)
+    torch.testing.assert_close(W, W.T, rtol=0, atol=0)
+    torch.testing.assert_close(
+        W.sum(dim=1), torch.ones(n, dtype=W.dtype), rtol=0, atol=1e-12
+    )
+    # off-diagonal support exactly matches the edge set
+    for i, j in g.edges():
+        assert W[i, j] > 0
+    comp = nx.complement(g)
+    for i, j in comp.edges():
+        assert W[i, j] == 0
+
+
+@settings(max_examples=60, deadline=None)
+@given(N=st.integers(1, 64), world=st.integers(1, 16))
+def test_node_layout_partitions(N, world):
+    lay = NodeLayout(N, world)
+    seen = []
+    for r in range(world):
+        nodes = list(lay.nodes_of(r))
+        assert len(nodes) == lay.counts[r]
+        for node in nodes:
+            assert lay.rank_of(node) == r
+            assert lay.local_index(node, r) == node - lay.starts[r]
+        seen.extend(nodes)
+    assert seen == list(range(N))
+    # contiguous blocks, sizes differ by at most one
+    assert max(lay.counts) - min(lay.counts) <= 1
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    widths=st.lists(st.integers(1, 9), min_size=2, max_size=5),
+)
+def test_param_layout_tiles_the_vector(widths):
+    model = FFReLUNet(widths)
+    layout, n = param_layout(model)
+    vec = torch.nn.utils.parameters_to_vector(model.parameters())
+    assert n == vec.numel()
+    off = 0
+    for name, shape, o in layout:
+        assert o == off, "offsets must be contiguous in vector order"
+        off += int(torch.tensor(shape).prod()) if shape else 1
+    assert off == n
+
+
+def test_model_spec_offsets_match_named_parameters():
+    for model in (
+        MNISTConvNet(3, 5, 64),
+        FourierNet([2, 16, 8, 1], scale=0.05),
+        FFReLUNet([4, 8, 2]),
+    ):
+        spec = model_spec(model)
+        vec = torch.nn.utils.parameters_to_vector(model.parameters())
+        assert spec.n == vec.numel()
+        params = dict(model.named_parameters())
+        layout, _ = param_layout(model)
+        offs = {name: off for name, _, off in layout}
+        for layer in spec.layers:
+            # the weight slice at w_off must BE the layer's weight
+            w = [p for nm, p in params.items()
+                 if offs[nm] == layer.w_off][0]
+            torch.testing.assert_close(
+                vec[layer.w_off : layer.w_off + w.numel()],
+                w.reshape(-1),
+                rtol=0,
+                atol=0,
+            )
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    N=st.integers(2, 20),
+    world=st.integers(1, 6),
+    rank=st.integers(0, 5),
+    p=st.floats(0.2, 0.9),
+    seed=st.integers(0, 10**6),
+)
+def test_csr_plan_matches_graph(N, world, rank, p, seed):
+    """The kernel-facing CSR must enumerate exactly each local node's
+    neighborhood, with local stack rows first and remote rows mapped
+    behind them — for any graph and any node->rank packing."""
+    from nn_distributed_training_amd.parallel import schedule
+
+    rank = rank % world
+    g = nx.erdos_renyi_graph(N, p, seed=seed)
+    lay = NodeLayout(N, world)
+    local = list(lay.nodes_of(rank))
+
+    # remote nodes: what edge_transfers would deliver for this rank
+    remote = sorted(
+        {j for i in local for j in g.neighbors(i)
+         if lay.rank_of(j) != rank}
+    )
+    row_of = schedule.row_map(local, remote)
+    offs, idx, w = schedule.build_csr(
+        g, local, row_of, torch.device("cpu"), torch.float64,
+        include_self=True, W=graph_generation.get_metropolis(g),
+    )
+    offs = offs.tolist()
+    idx = idx.tolist()
+    assert offs[0] == 0 and offs[-1] == len(idx)
+    L = len(local)
+    for li, i in enumerate(local):
+        rows = idx[offs[li] : offs[li + 1]]
+        assert rows[0] == li, "self row first under include_self"
+        expect = {row_of[j] for j in g.neighbors(i)}
+        assert set(rows[1:]) == expect
+        for r in rows:
+            assert 0 <= r < L + len(remote)
+    # weights: each row's entries sum to 1 (self + Metropolis neighbors)
+    wl = w.tolist()
+    for li in range(L):
+        assert abs(sum(wl[offs[li] : offs[li + 1]]) - 1.0) < 1e-9
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    lengths=st.lists(st.integers(3, 40), min_size=1, max_size=4),
+    batch=st.integers(1, 16),
+    draws=st.integers(1, 60),
+)
+def test_stream_sampler_epoch_accounting(lengths, batch, draws):
+    """Every batch column is a valid index; epochs advance exactly at
+    consumed/len boundaries; each full permutation covers the dataset."""
+    from nn_distributed_training_amd.ops.stacked import _StreamSampler
+
+    epochs = [0] * len(lengths)
+
+    def cb(li):
+        epochs[li] += 1
+
+    s = _StreamSampler(lengths, batch, torch.device("cpu"), seed=5,
+                       epoch_cb=cb, stream_batches=8)
+    for _ in range(draws):
+        stream, stride, off = s.next_ref()
+        for li, n in enumerate(lengths):
+            col = stream[li, off : off + batch]
+            assert int(col.max()) < n and int(col.min()) >= 0
+    for li, n in enumerate(lengths):
+        assert epochs[li] == (draws * batch) // n
