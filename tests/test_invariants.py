@@ -172,3 +172,73 @@ def test_stream_sampler_epoch_accounting(lengths, batch, draws):
             assert int(col.max()) < n and int(col.min()) >= 0
     for li, n in enumerate(lengths):
         assert epochs[li] == (draws * batch) // n
+
+
+def test_golden_dsgd_round_matches_matrix_form():
+    """One DSGD round (full-batch, world=1) must equal the closed form
+    theta' = W @ theta - alpha * grad(W @ theta) — the mixing is
+    snapshot-synchronous by design (documented deviation from the
+    reference's in-place Gauss-Seidel walk, optimizers/dsgd.py)."""
+    from nn_distributed_training_amd.data.mnist import (
+        SyntheticMNIST,
+        split_train_set,
+    )
+    from nn_distributed_training_amd.optimizers.dsgd import DSGD
+    from nn_distributed_training_amd.problems.dist_mnist_problem import (
+        DistMNISTProblem,
+    )
+
+    torch.set_default_dtype(torch.float64)
+    torch.manual_seed(4)
+    N, B = 4, 32
+    g = nx.cycle_graph(N)
+    train = SyntheticMNIST(N * B, seed=0)
+    val = SyntheticMNIST(32, seed=1)
+    subsets = split_train_set(train, N, "random")
+    conf = {
+        "problem_name": "closedform",
+        "train_batch_size": B,  # == per-node size: full batch
+        "val_batch_size": 32,
+        "data_seed": 2,
+        "verbose_evals": False,
+        "metrics": ["consensus_error"],
+        "metrics_config": {"evaluate_frequency": 10**9},
+        "optimizer_config": {
+            "alg_name": "dsgd", "outer_iterations": 1,
+            "alpha0": 0.01, "mu": 0.001, "profile": False,
+        },
+    }
+    pr = DistMNISTProblem(
+        g, MNISTConvNet(3, 5, 64), torch.nn.NLLLoss(), subsets, val,
+        torch.device("cpu"), conf,
+    )
+    theta0 = pr.local_params_stack().clone()
+
+    opt = DSGD(pr, pr.device, conf["optimizer_config"])
+    opt.train()
+    got = pr.local_params_stack()
+
+    # closed form on a fresh problem (same seeds -> same batches)
+    pr2 = DistMNISTProblem(
+        g, MNISTConvNet(3, 5, 64), torch.nn.NLLLoss(), subsets, val,
+        torch.device("cpu"), conf,
+    )
+    for li, i in enumerate(pr2.local_nodes):
+        torch.nn.utils.vector_to_parameters(
+            theta0[li], pr2.models[i].parameters()
+        )
+    W = graph_generation.get_metropolis(g)
+    mixed = W @ theta0
+    alpha = 0.01 * (1 - 0.001 * 0.01)
+    want = torch.empty_like(mixed)
+    for li, i in enumerate(pr2.local_nodes):
+        torch.nn.utils.vector_to_parameters(
+            mixed[li], pr2.models[i].parameters()
+        )
+        loss = pr2.local_batch_loss(i)
+        loss.backward()
+        gvec = torch.cat(
+            [p.grad.reshape(-1) for p in pr2.models[i].parameters()]
+        )
+        want[li] = mixed[li] - alpha * gvec
+    torch.testing.assert_close(got, want, rtol=1e-12, atol=1e-12)
