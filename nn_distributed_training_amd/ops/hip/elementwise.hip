@@ -135,17 +135,6 @@ __global__ void dsgt_y_update_k(
 }
 
 // ---------------------------------------------------------------------
-// Fused primal step. The DiNNO penalty gradient is analytic
-// (d/dth [ rho * sum_j ||th - th_reg_j||^2 ] = 2 rho (deg th - s), plus
-// the dual term from <th, dual>), so it folds into the optimizer update
-// and the autograd-visible loss never materializes (reference
-// dinno.py:74-91 builds it through torch.cdist + autograd every primal
-// iteration).  mode: 0=Adam 1=AdamW 2=SGD (matching dinno.py:55-72);
-// with_penalty=false gives the plain local step (DSGD's dsgd.py:49-58).
-// first_step: treat the Adam moments as zero without reading them —
-// the per-round m/v zero-fill kernels of the non-persistent DiNNO mode
-// fold away (reference recreates the Adam optimizer each round,
-// dinno.py:57-72).
 // Reduce per-tile gradient slabs [L, P, n] into [L, n] (consumers that
 // are not the fused step: DSGD's axpy, DSGT's tracker update).
 template <typename T>
@@ -165,6 +154,18 @@ __global__ void reduce_parts_k(const T* __restrict__ parts,
   }
 }
 
+// ---------------------------------------------------------------------
+// Fused primal step. The DiNNO penalty gradient is analytic
+// (d/dth [ rho * sum_j ||th - th_reg_j||^2 ] = 2 rho (deg th - s), plus
+// the dual term from <th, dual>), so it folds into the optimizer update
+// and the autograd-visible loss never materializes (reference
+// dinno.py:74-91 builds it through torch.cdist + autograd every primal
+// iteration).  mode: 0=Adam 1=AdamW 2=SGD (matching dinno.py:55-72);
+// with_penalty=false gives the plain local step (DSGD's dsgd.py:49-58).
+// first_step: treat the Adam moments as zero without reading them —
+// the per-round m/v zero-fill kernels of the non-persistent DiNNO mode
+// fold away (reference recreates the Adam optimizer each round,
+// dinno.py:57-72).
 template <typename T, int MODE, bool WITH_PENALTY>
 __global__ void fused_step_k(
     T* __restrict__ theta, const T* __restrict__ grad,
