@@ -32,6 +32,8 @@ def _load_img(img, border_width: int) -> np.ndarray:
         img = img.copy()
         img[:, :border_width] = 1.0
         img[:border_width, :] = 1.0
+        # `-1` end quirk kept for parity: the reference leaves the very
+        # last row/column unbordered (floorplans/lidar/lidar.py:39-42)
         img[:, -border_width:-1] = 1.0
         img[-border_width:-1, :] = 1.0
     return img
