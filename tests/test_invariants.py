@@ -7,6 +7,7 @@ flat-vector offsets must tile the parameter vector exactly.
 """
 
 import networkx as nx
+import numpy as np
 import torch
 from hypothesis import given, settings
 from hypothesis import strategies as st
@@ -242,3 +243,31 @@ def test_golden_dsgd_round_matches_matrix_form():
         )
         want[li] = mixed[li] - alpha * gvec
     torch.testing.assert_close(got, want, rtol=1e-12, atol=1e-12)
+
+
+def test_lidar_scan_geometry():
+    """Scan points stay within beam length of the pose; output is
+    [num_beams * beam_samps, 3]; rounded densities are {0, 1}."""
+    from nn_distributed_training_amd.data.floorplan import (
+        synthetic_floorplan,
+    )
+    from nn_distributed_training_amd.data.lidar import Lidar2D
+
+    img = synthetic_floorplan(nx=96, ny=96, num_walls=3,
+                              border_width=10, seed=1)
+    lidar = Lidar2D(img, 8, 0.25, 10, 1.0, 30, 3)
+    # find a free pose
+    pos = None
+    for x in lidar.xs[::5]:
+        for y in lidar.ys[::5]:
+            if lidar.density.ev(x, y) < 0.4:
+                pos = np.array([x, y])
+                break
+        if pos is not None:
+            break
+    scan = lidar.scan(pos)
+    assert scan.shape == (8 * 10, 3)
+    d = np.linalg.norm(scan[:, :2] - pos, axis=1)
+    assert d.max() <= lidar.beam_len * 1.0001
+    # density channel within spline overshoot tolerance of [0, 1]
+    assert scan[:, 2].min() > -0.5 and scan[:, 2].max() < 1.5
