@@ -106,3 +106,32 @@ def test_checkpoint_write_and_resume(alg, tmp_path):
             payload["models"][i].items(),
         ):
             torch.testing.assert_close(pa, pb, rtol=0, atol=0)
+
+
+def test_resume_rejects_different_packing(tmp_path):
+    """A checkpoint written under one node->rank packing must refuse to
+    load into a problem with different local nodes (shards are
+    per-rank; silently mixing them would corrupt training)."""
+    from nn_distributed_training_amd.optimizers.checkpointing import (
+        load_checkpoint,
+        save_checkpoint,
+    )
+
+    pr, opt = _make("dsgd", tmp_path)
+    opt.train()
+    save_checkpoint(str(tmp_path), pr, 5, {"alph": 0.004})
+
+    pr2, _ = _make("dsgd", tmp_path)
+    pr2.local_nodes = [0, 1]  # simulate a different packing
+    with pytest.raises(AssertionError, match="packing"):
+        load_checkpoint(str(tmp_path), pr2)
+
+
+def test_checkpoint_every_without_dir_raises(tmp_path):
+    from nn_distributed_training_amd.optimizers.checkpointing import (
+        save_checkpoint,
+    )
+
+    pr, _ = _make("dsgd", tmp_path)
+    with pytest.raises(ValueError, match="checkpoint_every"):
+        save_checkpoint(None, pr, 0, {})
