@@ -64,6 +64,27 @@ def timing_report():
                 f"  {k:20s} {_tacc[k]*1e3:9.1f} ms  {_tcnt[k]:6d} calls"
             )
 
+
+def set_timing(on: bool):
+    """Enable/disable the section timers at runtime (bench.py turns
+    them on for its sustained window only, keeping the exactly-timed
+    headline region instrumentation-free)."""
+    global _TIMING
+    _TIMING = bool(on)
+
+
+def timing_reset():
+    _tacc.clear()
+    _tcnt.clear()
+
+
+def timing_dict():
+    """Accumulated section times as a JSON-ready dict (ms, calls)."""
+    return {
+        k: {"ms": round(_tacc[k] * 1e3, 3), "calls": _tcnt[k]}
+        for k in sorted(_tacc, key=lambda k: -_tacc[k])
+    }
+
 from ..models.spec import ModelSpec, model_spec
 from . import get_ext
 
@@ -1142,12 +1163,13 @@ class DSGTStackedDriver:
             ext.dsgt_y_update(self.y_mix, eng.grad, self.g, self.y)
 
     def _opt_state(self):
-        return {"y": self.y, "g": self.g, "y_mix": self.y_mix}
+        # y_mix is per-round scratch (overwritten by dsgt_mix before any
+        # read) — excluded from checkpoints
+        return {"y": self.y, "g": self.g}
 
     def _load_opt_state(self, st):
         self.y.copy_(st["y"])
         self.g.copy_(st["g"])
-        self.y_mix.copy_(st["y_mix"])
 
     def run(self, profiler=None):
         _run_rounds(self, profiler, skip_bootstrap_on_resume=True)

@@ -70,9 +70,15 @@ def load_checkpoint(resume_from, pr) -> tuple:
     )
     payload = torch.load(path, map_location=pr.device,
                          weights_only=False)
-    assert payload["local_nodes"] == list(pr.local_nodes), (
-        "checkpoint was written with a different node->rank packing"
-    )
+    if payload["local_nodes"] != list(pr.local_nodes):
+        # not an assert: must survive `python -O`, or a mismatched
+        # world size would load a wrong packing
+        raise RuntimeError(
+            f"checkpoint {path} was written with node->rank packing "
+            f"{payload['local_nodes']} but this rank hosts "
+            f"{list(pr.local_nodes)} — resume with the same world size "
+            "as the run that wrote the checkpoint"
+        )
     for i in pr.local_nodes:
         pr.models[i].load_state_dict(payload["models"][i])
     pr.epoch_tracker = payload["epoch_tracker"]

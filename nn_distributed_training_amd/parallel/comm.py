@@ -33,6 +33,8 @@ exchange degenerates to local reads, which is the single-GPU packed mode
 from __future__ import annotations
 
 import os
+import sys
+import time
 from datetime import timedelta
 from typing import Dict, List, Sequence
 
@@ -44,9 +46,49 @@ import torch.distributed as dist
 #: schedule shows up as a stuck recv). Override: NDTA_COMM_TIMEOUT_S.
 COMM_TIMEOUT_S = float(os.environ.get("NDTA_COMM_TIMEOUT_S", "300"))
 
+#: NDTA_COMM_DEBUG=1 logs every exchange's schedule (peers/ops/bytes)
+#: per rank to stderr — first-line diagnostic for a desynchronized or
+#: hung multi-GPU round.
+COMM_DEBUG = os.environ.get("NDTA_COMM_DEBUG", "0") not in ("0", "")
+
+
+def _comm_log(rank: int, msg: str):
+    if COMM_DEBUG:
+        print(f"[ndta-comm rank {rank} t={time.monotonic():.3f}] {msg}",
+              file=sys.stderr, flush=True)
+
 
 def _wait_all(works, what: str, rank: int):
-    """Wait for a batch of P2P works with the watchdog timeout."""
+    """Wait for a batch of P2P works with the watchdog timeout.
+
+    Backend split: under gloo, ``wait(timedelta)``
+    honors the timeout and returns False on expiry — we raise.  Under
+    NCCL/RCCL the per-work timeout is NOT honored by ``wait`` (it only
+    stream-orders); the guard there is the process group's own watchdog
+    thread, armed by ``init_from_env`` via the PG timeout +
+    TORCH_NCCL_ASYNC_ERROR_HANDLING, which tears the job down with a
+    rank-attributed error instead of hanging.  A host-side poll of
+    ``is_completed`` is additionally run when NDTA_COMM_BLOCKING=1 (the
+    debug mode) so a stuck transfer raises in *this* stack with `what`
+    attached.
+    """
+    blocking = os.environ.get("NDTA_COMM_BLOCKING", "0") not in ("0", "")
+    if blocking:
+        deadline = time.monotonic() + COMM_TIMEOUT_S
+        pending = list(works)
+        while pending:
+            pending = [w for w in pending if not w.is_completed()]
+            if not pending:
+                break
+            if time.monotonic() > deadline:
+                raise RuntimeError(
+                    f"[rank {rank}] neighbor-exchange watchdog: {what} "
+                    f"did not complete within {COMM_TIMEOUT_S}s — check "
+                    "graph connectivity / schedule symmetry "
+                    "(NDTA_COMM_DEBUG=1 logs the per-rank schedule)"
+                )
+            time.sleep(0.001)
+        return
     for w in works:
         try:
             ok = w.wait(timedelta(seconds=COMM_TIMEOUT_S))
@@ -161,8 +203,16 @@ class Communicator:
             for d in dests:
                 ops.append(dist.P2POp(dist.irecv, d[j], peer))
         if ops:
+            _comm_log(
+                self.rank,
+                f"exchange_rows: {len(ops)} ops, "
+                f"send={sorted(send_pairs, key=lambda t: (t[1], t[0]))}, "
+                f"recv={sorted(recv_nodes)}, "
+                f"stacks={[tuple(s.shape) for s in stacks]}",
+            )
             _wait_all(dist.batch_isend_irecv(ops),
                       f"{len(ops)} P2P ops", self.rank)
+            _comm_log(self.rank, "exchange_rows: done")
 
     # ------------------------------------------------------------------
     def exchange_node_vectors(
@@ -203,8 +253,15 @@ class Communicator:
         for j in sorted(recv_nodes, key=lambda j: (layout.rank_of(j), j)):
             ops.append(dist.P2POp(dist.irecv, recv_bufs[j], layout.rank_of(j)))
         if ops:
+            _comm_log(
+                self.rank,
+                f"exchange_node_vectors: {len(ops)} ops, "
+                f"send={sorted(send_pairs, key=lambda t: (t[1], t[0]))}, "
+                f"recv={sorted(recv_nodes)}, n={n}",
+            )
             _wait_all(dist.batch_isend_irecv(ops),
                       f"{len(ops)} P2P ops", self.rank)
+            _comm_log(self.rank, "exchange_node_vectors: done")
         return recv_bufs
 
     # ------------------------------------------------------------------
@@ -260,6 +317,12 @@ def init_from_env(backend: str | None = None) -> tuple:
     if not dist.is_initialized():
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if backend == "nccl":
+            # arm the NCCL watchdog: on a P2P/collective exceeding the
+            # PG timeout the watchdog tears the job down with a
+            # rank-attributed error instead of hanging forever.  Must be
+            # set BEFORE init_process_group (read at PG construction).
+            os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
         dist.init_process_group(
             backend=backend,
             timeout=timedelta(seconds=max(COMM_TIMEOUT_S, 60.0)),

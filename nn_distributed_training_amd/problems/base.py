@@ -143,6 +143,27 @@ class ProblemBase:
         full = self.comm.all_gather_rows(self.layout, col)
         return full.reshape(-1).cpu()
 
+    def gather_per_node_rows(self, local_rows: torch.Tensor) -> torch.Tensor:
+        """All-gather an [L, ...] per-node payload into [N, ...].
+
+        Used for metrics with a per-node vector/image payload
+        (validation_as_vector, mesh_grid_density): rank 0 is the only
+        rank that saves metrics, so payloads must cross ranks or a
+        multi-rank run would silently persist rank 0's nodes only
+        (ADVICE r1 item 1).
+        """
+        L = local_rows.shape[0]
+        tail = local_rows.shape[1:]
+        flat = local_rows.reshape(L, -1)
+        # collectives don't carry bool; round-trip via uint8
+        cast = flat.dtype == torch.bool
+        if cast:
+            flat = flat.to(torch.uint8)
+        full = self.comm.all_gather_rows(self.layout, flat.to(self.device))
+        if cast:
+            full = full.to(torch.bool)
+        return full.reshape(self.N, *tail).cpu()
+
     # ------------------------------------------------------------------
     def save_metrics(self, output_dir):
         if not self.is_root:

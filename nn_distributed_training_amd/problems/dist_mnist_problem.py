@@ -112,8 +112,20 @@ class DistMNISTProblem(ProblemBase):
                 )
             elif met_name in ("validation_as_vector",
                               "valdiation_as_vector"):
-                # per-sample correctness vectors of this rank's nodes
-                self.metrics[met_name].append(valid_vecs)
+                # per-sample correctness vectors, gathered across ranks
+                # so rank 0 (the saver) holds every node's vector
+                if valid_vecs:
+                    loc = torch.stack(
+                        [valid_vecs[i].reshape(-1)
+                         for i in self.local_nodes]
+                    )
+                else:
+                    loc = torch.zeros(0, len(self.val_set),
+                                      dtype=torch.bool)
+                full = self.gather_per_node_rows(loc)
+                self.metrics[met_name].append(
+                    {i: full[i].reshape(-1, 1) for i in range(self.N)}
+                )
             else:
                 raise NameError("Unknown metric.")
 

@@ -205,8 +205,17 @@ class DistOnlineDensityProblem(ProblemBase):
                     dens = [
                         self.mesh_grid_density(i) for i in self.local_nodes
                     ]
+                    loc = (
+                        torch.stack(dens)
+                        if dens
+                        else torch.zeros(
+                            0, *self.mesh_inputs.shape[:1], 1,
+                            device=self.device,
+                        )
+                    )
+                    # gather so rank 0 (the saver) has every node's mesh
                     self.metrics[met_name].append(
-                        torch.stack(dens).cpu() if dens else torch.zeros(0)
+                        self.gather_per_node_rows(loc)
                     )
             elif met_name == "forward_pass_count":
                 self.metrics[met_name].append(self.forward_cnt)

@@ -13,9 +13,12 @@ REQUIRED = {
 
 
 def _run(args):
+    import os
+
+    env = {**os.environ, "NDTA_BENCH_SUSTAIN_S": "0.5"}
     out = subprocess.run(
         [sys.executable, "bench.py", *args],
-        capture_output=True, text=True, timeout=420,
+        capture_output=True, text=True, timeout=420, env=env,
     )
     assert out.returncode == 0, out.stderr[-2000:]
     line = out.stdout.strip().splitlines()[-1]
@@ -40,6 +43,11 @@ def test_bench_default_contract():
     assert cfg["nodes"] == 8 and cfg["alg"] == "dinno"
     assert "consensus_err_max" in cfg and "val_acc_max" in cfg
     assert "parallelism" in cfg
+    # sustained-window evidence keys
+    assert d["timed_region_s"] > 0
+    assert d["sustained_rounds_per_sec"] > 0
+    assert d["sustained_s"] > 0 and d["sustained_steps"] >= 1
+    assert "per_step_ms_p50" in d and "per_step_ms_std" in d
 
 
 def test_bench_dsgd_config2():

@@ -123,7 +123,8 @@ def test_resume_rejects_different_packing(tmp_path):
 
     pr2, _ = _make("dsgd", tmp_path)
     pr2.local_nodes = [0, 1]  # simulate a different packing
-    with pytest.raises(AssertionError, match="packing"):
+    # RuntimeError (not AssertionError): must survive `python -O`
+    with pytest.raises(RuntimeError, match="packing"):
         load_checkpoint(str(tmp_path), pr2)
 
 
