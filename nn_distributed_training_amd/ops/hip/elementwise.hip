@@ -375,4 +375,42 @@ __global__ void axpy_k(T* __restrict__ x, const T* __restrict__ g,
   }
 }
 
+// Keyed Feistel permutation of [0, n): out[i] = lb + pi(i), pi a
+// bijection. ONE elementwise launch replaces the device
+// torch.randperm chain (rocprim radix sort + merges + arange/fill —
+// ~245 us/round amortized in the online density workload, see
+// profiles/dens_final_topk.txt). 4 Feistel rounds over the smallest
+// 2^(2*hb) >= n, cycle-walking back into [0, n).
+DEV_INLINE unsigned feistel_round_f(unsigned r, unsigned key) {
+  unsigned f = r * 0x9E3779B9u + key;
+  f ^= f >> 13;
+  f *= 0x85EBCA6Bu;
+  f ^= f >> 16;
+  return f;
+}
+
+__global__ void feistel_perm_k(long* __restrict__ out, long n, long lb,
+                               unsigned long long key, int half_bits) {
+  const unsigned mask = (1u << half_bits) - 1u;
+  for (long i = blockIdx.x * (long)BLOCK + threadIdx.x; i < n;
+       i += (long)gridDim.x * BLOCK) {
+    unsigned long long x = (unsigned long long)i;
+    do {
+      unsigned lhs = (unsigned)(x >> half_bits);
+      unsigned rhs = (unsigned)x & mask;
+#pragma unroll
+      for (int rnd = 0; rnd < 4; ++rnd) {
+        const unsigned f =
+            feistel_round_f(rhs, (unsigned)(key >> (16 * rnd)) + rnd)
+            & mask;
+        const unsigned nl = rhs;
+        rhs = lhs ^ f;
+        lhs = nl;
+      }
+      x = ((unsigned long long)lhs << half_bits) | rhs;
+    } while (x >= (unsigned long long)n);
+    out[i] = lb + (long)x;
+  }
+}
+
 }  // namespace ew

@@ -322,8 +322,9 @@ void linear_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
   CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
   const long L = theta.size(0), n = theta.size(1);
   // MFMA pays only when M fills the 64-row tiles across the chip;
-  // small-M layers (MNIST fc, per-node B=64) stay on the VALU kernels
-  const bool use_mfma = (M >= 128 && O >= 16 && I >= 8);
+  // small-M layers (MNIST fc, per-node B=64) stay on the VALU kernels.
+  // I must be even for the 16-byte staging loads (vec2 alignment).
+  const bool use_mfma = (M >= 128 && O >= 16 && I >= 8 && I % 2 == 0);
   DISPATCH_FT(X, {
     auto zp = Z.has_value() ? Z->data_ptr<scalar_t>() : nullptr;
     if (use_mfma) {
@@ -414,7 +415,28 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
   CHECK_DEV(dZ); CHECK_DEV(X); CHECK_DEV(gstack);
   const long L = gstack.size(0), n = gstack.size(1);
   DISPATCH_FT(dZ, {
-    if (M >= 256 && I >= 16 && O >= 16) {
+    if (M >= 256 && O % 64 == 0 && I % 16 == 0 && I <= 448 && O <= 512) {
+      // full-I tiles: dZ and X each fetched from HBM exactly once;
+      // direct-global fragment reads (both operands m-row-major), no
+      // LDS/barriers — see gemm_mfma.hip mfma_dw_direct_k rationale
+      const long otiles = O / 64;
+      long nchunk = std::max<long>(1, 512 / std::max<long>(1, otiles * L));
+      nchunk = std::min<long>(nchunk, (M + 63) / 64);
+      dim3 grid(1, otiles, L * nchunk);
+      if (I <= 256) {
+        hipLaunchKernelGGL((gmfma::mfma_dw_direct_k<scalar_t, 4>),
+            grid, dim3(512), 0, cur_stream(),
+            dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+            gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+            (int)M, (int)I, (int)O, (int)nchunk);
+      } else {
+        hipLaunchKernelGGL((gmfma::mfma_dw_direct_k<scalar_t, 7>),
+            grid, dim3(512), 0, cur_stream(),
+            dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+            gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+            (int)M, (int)I, (int)O, (int)nchunk);
+      }
+    } else if (M >= 256 && I >= 16 && O >= 16) {
       // fill the chip: tiles * L * nchunk ≈ 2048 blocks
       const long tiles = ((I + 63) / 64) * ((O + 63) / 64);
       long nchunk = std::max<long>(1, 2048 / std::max<long>(1, tiles * L));
@@ -655,6 +677,24 @@ void regression_bwd(torch::Tensor yhat, torch::Tensor tgt,
   HIP_CHECK_LAST();
 }
 
+// keyed bijection of [0, n) written into `out` (int64, contiguous):
+// the online-density sampler's shuffle (ops/stacked.py
+// _OnlineWindowSampler) — one launch instead of a device randperm's
+// rocprim sort chain
+void feistel_perm(torch::Tensor out, long n, long lb, long key) {
+  CHECK_DEV(out);
+  TORCH_CHECK(out.scalar_type() == torch::kLong,
+              "feistel_perm: out must be int64");
+  TORCH_CHECK(out.numel() >= n, "feistel_perm: out too small");
+  int bits = 1;
+  while ((1L << bits) < n) ++bits;
+  const int hb = std::max(1, (bits + 1) / 2);
+  hipLaunchKernelGGL(ew::feistel_perm_k,
+      dim3(grid_1d(n)), dim3(ew::BLOCK), 0, cur_stream(),
+      out.data_ptr<long>(), n, lb, (unsigned long long)key, hb);
+  HIP_CHECK_LAST();
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -683,4 +723,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("mnist_train_step", &mnist_train_step);
   mod.def("bce_bwd", &bce_bwd);
   mod.def("regression_bwd", &regression_bwd);
+  mod.def("feistel_perm", &feistel_perm);
 }

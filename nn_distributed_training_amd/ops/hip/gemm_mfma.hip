@@ -63,6 +63,17 @@ constexpr int BK = 16;   // K per LDS stage (BK=32 measured neutral)
 
 // ---------------------------------------------------------------------
 // NT: Y[M,O] = act(X[M,I] @ W_l[O,I]^T + b_l)  [+ optional Z store]
+//
+// Round-2 structure (measured 139 us vs ~105 roofline at BK=16
+// single-buffered): BK=32, 16-byte (2xf64) staging loads, and the T14
+// write-after-barrier register pipeline — ONE register set holds stage
+// s+1 while the MFMA loop runs on the LDS image of stage s, so the HBM
+// fetch of the next tile overlaps compute instead of serializing at
+// the barrier (cdna_hip_programming.md §5.5 T14 / G15).
+// Requires I % 2 == 0 for the vectorized path (dispatch falls back to
+// the element loop otherwise via the k-bounds checks: pairs are only
+// vector-loaded when both elements are in range).
+constexpr int FK = 32;  // fwd K per stage
 template <typename T>
 __global__ __launch_bounds__(256) void mfma_fwd_k(
     const T* __restrict__ X, const T* __restrict__ theta,
@@ -71,8 +82,9 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
     int act, T scale) {
   using MF = mfma_t<T>;
   using acc_t = typename MF::acc_t;
-  __shared__ T As[BK][BM + 1];   // A^T image: As[k][m]
-  __shared__ T Bs[BK][BN + 1];   // Bs[k][o] = W[o][k]
+  typedef T vec2 __attribute__((ext_vector_type(2)));
+  __shared__ T As[FK][BM + 1];   // A^T image: As[k][m]
+  __shared__ T Bs[FK][BN + 1];   // Bs[k][o] = W[o][k]
 
   const long l = blockIdx.z;
   const T* Xl = X + l * (long)M * I;
@@ -87,38 +99,95 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
   const int wm = (wid >> 1) * 32;
   const int wn = (wid & 1) * 32;
 
+  // staging assignment: pair u = tid + 256*q covers row u/16 of the
+  // tile, k-pair u%16 (16 threads stream 256 contiguous bytes per row)
+  const int sm = tid / 16;           // A row (m) / B row (o) of pair 0
+  const int skp = tid % 16;          // k-pair within the row
   acc_t acc[2][2] = {};
+  vec2 ra[4], rb[4];
 
-  for (int k0 = 0; k0 < I; k0 += BK) {
-    // stage A^T: 64 rows x 16 k  (1024 elems, 4 per thread)
-    for (int t = tid; t < BM * BK; t += 256) {
-      const int m = t / BK, k = t % BK;
-      As[k][m] = (m0 + m < M && k0 + k < I)
-                     ? Xl[(long)(m0 + m) * I + (k0 + k)]
-                     : T(0);
-    }
-    // stage B: Bs[k][o] = W[o0+o][k0+k]
-    for (int t = tid; t < BN * BK; t += 256) {
-      const int o = t / BK, k = t % BK;
-      Bs[k][o] = (o0 + o < O && k0 + k < I)
-                     ? W[(long)(o0 + o) * I + (k0 + k)]
-                     : T(0);
-    }
-    __syncthreads();
+  const int nstages = (I + FK - 1) / FK;
+  // ---- load stage 0 into registers
+  {
+    const int k0 = 0;
 #pragma unroll
-    for (int kk = 0; kk < BK; kk += 4) {
-      const int ka = kk + (lane >> 4);
-#pragma unroll
-      for (int fm = 0; fm < 2; ++fm) {
-        const T a = As[ka][wm + fm * 16 + (lane & 15)];
-#pragma unroll
-        for (int fn = 0; fn < 2; ++fn) {
-          const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
-          acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      const int k = k0 + 2 * skp;
+      ra[q] = vec2{0, 0};
+      rb[q] = vec2{0, 0};
+      if (m0 + m < M) {
+        if (k + 1 < I) {
+          ra[q] = *reinterpret_cast<const vec2*>(
+              &Xl[(long)(m0 + m) * I + k]);
+        } else if (k < I) {
+          ra[q].x = Xl[(long)(m0 + m) * I + k];
+        }
+      }
+      if (o0 + m < O) {
+        if (k + 1 < I) {
+          rb[q] = *reinterpret_cast<const vec2*>(
+              &W[(long)(o0 + m) * I + k]);
+        } else if (k < I) {
+          rb[q].x = W[(long)(o0 + m) * I + k];
         }
       }
     }
-    __syncthreads();
+  }
+
+  for (int s = 0; s < nstages; ++s) {
+    __syncthreads();  // previous compute done: LDS free
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      As[2 * skp][m] = ra[q].x;
+      As[2 * skp + 1][m] = ra[q].y;
+      Bs[2 * skp][m] = rb[q].x;
+      Bs[2 * skp + 1][m] = rb[q].y;
+    }
+    if (s + 1 < nstages) {  // issue next stage's fetch NOW (T14)
+      const int k0 = (s + 1) * FK;
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int m = sm + q * 16;
+        const int k = k0 + 2 * skp;
+        ra[q] = vec2{0, 0};
+        rb[q] = vec2{0, 0};
+        if (m0 + m < M) {
+          if (k + 1 < I) {
+            ra[q] = *reinterpret_cast<const vec2*>(
+                &Xl[(long)(m0 + m) * I + k]);
+          } else if (k < I) {
+            ra[q].x = Xl[(long)(m0 + m) * I + k];
+          }
+        }
+        if (o0 + m < O) {
+          if (k + 1 < I) {
+            rb[q] = *reinterpret_cast<const vec2*>(
+                &W[(long)(o0 + m) * I + k]);
+          } else if (k < I) {
+            rb[q].x = W[(long)(o0 + m) * I + k];
+          }
+        }
+      }
+    }
+    __syncthreads();  // LDS image of stage s visible
+    const int klim = min(FK, I - s * FK);
+#pragma unroll
+    for (int kk = 0; kk < FK; kk += 4) {
+      if (kk < klim) {
+        const int ka = kk + (lane >> 4);
+#pragma unroll
+        for (int fm = 0; fm < 2; ++fm) {
+          const T a = As[ka][wm + fm * 16 + (lane & 15)];
+#pragma unroll
+          for (int fn = 0; fn < 2; ++fn) {
+            const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
+            acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+          }
+        }
+      }
+    }
   }
 
   // epilogue: bias + activation, coalesced per-fragment store
@@ -315,6 +384,136 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
         if (o < O && i < I) {
           atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
                     acc[fm][fn][r]);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// TN full-I variant: dW[O, I] = dZ[M,O]^T @ X[M,I] with one block
+// covering a 64-o x FULL-I tile (I <= 448, I%16 == 0, O%64 == 0).
+//
+// Rationale (round-2 rework of mfma_dw_k, which measured 2.5x over its
+// traffic roofline): with 64x64 tiles the dZ panel is re-fetched once
+// per i-tile (4x for I=256), and the single-buffered LDS stage
+// serializes HBM against MFMA.  Full-I tiles fetch dZ and X from HBM
+// exactly once.  Both operands of this contraction (k = m) are
+// m-row-major in memory, so the MFMA fragment reads (16 consecutive
+// o / i within a row) coalesce DIRECTLY from global: no LDS image, no
+// barriers; the CU's L1 serves the (x4 i-wave, x2 o-wave) intra-block
+// re-reads.  8 independent accumulators per wave cover the f64 MFMA
+// dependent latency; 2 waves/SIMD + unrolled k keep ~50 loads in
+// flight per SIMD against the ~900-cycle HBM latency.
+// db is accumulated by the i0==0 wave pair from its own a-loads
+// (each (m, o) element passes through exactly one (lane, k-class)).
+template <typename T, int NIMAX>
+__global__ __launch_bounds__(512) void mfma_dw_direct_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O, int nchunk) {
+  using MF = mfma_t<T>;
+  using acc_t = typename MF::acc_t;
+
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const T* Gl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+
+  // 4-aligned chunk bounds so only the LAST chunk has a ragged tail
+  int mc = (M + nchunk - 1) / nchunk;
+  mc = (mc + 3) & ~3;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+  if (mlo >= mhi) return;
+
+  const int o0 = blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;            // 8 waves: (wo, wi) = (wid&1, wid>>2)? no:
+  const int wo = (wid & 1) * 32;       // 2 o-waves x 4 i-waves
+  const int wi = wid >> 1;
+  const int fi_per = (I / 16 + 3) / 4; // i-fragments per wave (<= 7)
+  const int i0 = wi * fi_per * 16;
+  const int ni = min(fi_per, (I - i0) / 16);  // frags this wave owns
+
+  const int lo = lane & 15;            // fragment row/col
+  const int lk = lane >> 4;            // k sub-step
+  acc_t acc[2][NIMAX] = {};
+  T db0 = T(0), db1 = T(0);
+  const bool bias_wave = (wi == 0) && (b_off >= 0);
+
+  const T* __restrict__ ga0 = Gl + o0 + wo + lo;
+  const T* __restrict__ ga1 = ga0 + 16;
+  const T* __restrict__ gb = Xl + i0 + lo;
+
+  int k = mlo;
+  if (ni > 0) {
+    for (; k + 4 <= mhi; k += 4) {
+      const long ka = k + lk;
+      const T a0 = ga0[ka * O];
+      const T a1 = ga1[ka * O];
+      if (bias_wave) { db0 += a0; db1 += a1; }
+#pragma unroll
+      for (int fi = 0; fi < NIMAX; ++fi) {
+        if (fi < ni) {
+          const T b = gb[ka * I + fi * 16];
+          acc[0][fi] = MF::mma(a0, b, acc[0][fi]);
+          acc[1][fi] = MF::mma(a1, b, acc[1][fi]);
+        }
+      }
+    }
+    if (k < mhi) {  // ragged tail (< 4 rows): zero-padded operands
+      const long ka = k + lk;
+      const bool ok = ka < mhi;
+      const T a0 = ok ? ga0[ka * O] : T(0);
+      const T a1 = ok ? ga1[ka * O] : T(0);
+      if (bias_wave) { db0 += a0; db1 += a1; }
+#pragma unroll
+      for (int fi = 0; fi < NIMAX; ++fi) {
+        if (fi < ni) {
+          const T b = ok ? gb[ka * I + fi * 16] : T(0);
+          acc[0][fi] = MF::mma(a0, b, acc[0][fi]);
+          acc[1][fi] = MF::mma(a1, b, acc[1][fi]);
+        }
+      }
+    }
+  } else if (bias_wave) {
+    for (; k + 4 <= mhi; k += 4) {
+      const long ka = k + lk;
+      db0 += ga0[ka * O];
+      db1 += ga1[ka * O];
+    }
+    if (k < mhi && k + lk < mhi) {
+      db0 += ga0[(long)(k + lk) * O];
+      db1 += ga1[(long)(k + lk) * O];
+    }
+  }
+
+  if (bias_wave) {
+    // fold the 4 k-classes of each column: lanes lo, lo+16, lo+32, lo+48
+#pragma unroll
+    for (int off = 32; off >= 16; off >>= 1) {
+      db0 += __shfl_down(db0, off, WAVE);
+      db1 += __shfl_down(db1, off, WAVE);
+    }
+    if (lane < 16) {
+      atomicAdd(&gstack[l * n + b_off + o0 + wo + lo], db0);
+      atomicAdd(&gstack[l * n + b_off + o0 + wo + 16 + lo], db1);
+    }
+  }
+
+#pragma unroll
+  for (int fo = 0; fo < 2; ++fo) {
+#pragma unroll
+    for (int fi = 0; fi < NIMAX; ++fi) {
+      if (fi < ni) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int o = o0 + wo + fo * 16 + MF::acc_row(lane, r);
+          const int i = i0 + fi * 16 + lo;
+          atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
+                    acc[fo][fi][r]);
         }
       }
     }

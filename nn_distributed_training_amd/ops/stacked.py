@@ -195,12 +195,37 @@ class _OnlineWindowSampler:
         self.B = batch
         self.device = device
         self.epoch_cb = epoch_cb
-        # device-resident generators: window shuffles never touch host
-        self.gens = [
-            torch.Generator(device=device).manual_seed(seed * 100003 + i)
-            for i in range(len(datasets))
-        ]
-        self.pools = [None] * len(datasets)
+        self.seed = seed
+        # On GPU the per-window shuffle is a keyed Feistel bijection in
+        # ONE kernel (ext.feistel_perm) — a device torch.randperm is a
+        # rocprim radix-sort + merge chain that cost ~245 us/round
+        # amortized (profiles/dens_final_topk.txt). Keyed by
+        # (seed, node, refill#): identical streams no matter which rank
+        # hosts the node. CPU keeps generator randperm.
+        self._use_feistel = device.type == "cuda"
+        if not self._use_feistel:
+            self.gens = [
+                torch.Generator(device=device).manual_seed(
+                    seed * 100003 + i
+                )
+                for i in range(len(datasets))
+            ]
+        self.refills = [0] * len(datasets)
+        # ONE [L, cap] pool tensor: when every node's window has the
+        # same size (the standard config — same lidar, same
+        # num_scans_in_window) the positions stay synchronized and
+        # next_ref is ZERO copies and zero kernels: the gather kernels
+        # index the pool directly via (stride=cap, off=pos). Host-side
+        # per-node op chains here were 83% of the density round
+        # (BENCH r2a timing_breakdown: next_batch 1.83 of 2.18 ms).
+        cap = max(
+            ds.window_bounds[1] - ds.window_bounds[0] for ds in datasets
+        )
+        self.cap = cap
+        self.pools = torch.empty(
+            len(datasets), cap, dtype=torch.long, device=device
+        )
+        self.wins = [0] * len(datasets)
         self.pos = [0] * len(datasets)
         self.buf = torch.empty(
             len(datasets), batch, dtype=torch.long, device=device
@@ -218,24 +243,58 @@ class _OnlineWindowSampler:
         # golden path's host list shuffle was ~30 ms per 50k-sample
         # window and dominated the round (profiles/ density pass 2)
         lb, ub = ds.window_bounds
-        perm = torch.randperm(
-            ub - lb, generator=self.gens[li], device=self.device
-        )
-        self.pools[li] = perm + lb
+        win = ub - lb
+        if win > self.cap:  # dynamic window growth: re-alloc the pool
+            self.cap = win
+            self.pools = torch.empty(
+                len(self.dss), win, dtype=torch.long, device=self.device
+            )
+            for lj in range(len(self.dss)):
+                if lj != li and self.wins[lj] > 0:
+                    self._regen(lj)
+        self.wins[li] = win
+        self._regen(li)
+        self.refills[li] += 1
         self.pos[li] = 0
 
+    def _regen(self, li):
+        win = self.wins[li]
+        lb = self.dss[li].window_bounds[0]
+        row = self.pools[li]
+        if self._use_feistel:
+            key = (self.seed * 100003 + li) * 2654435761 + \
+                self.refills[li]
+            get_ext().feistel_perm(row, win, lb,
+                                   key & 0x7FFFFFFFFFFFFFFF)
+        else:
+            gen = self.gens[li]
+            row[:win] = torch.randperm(
+                win, generator=gen, device=self.device
+            ) + lb
+
     def next_ref(self):
+        B = self.B
+        # fast path: synchronized equal windows, batch fits — no copies
+        p0, w0 = self.pos[0], self.wins[0]
+        if (p0 + B <= w0
+                and all(p == p0 for p in self.pos)
+                and all(w == w0 for w in self.wins)):
+            for li in range(len(self.dss)):
+                self.pos[li] = p0 + B
+            return self.pools, self.cap, p0
         for li in range(len(self.dss)):
             p = self.pos[li]
-            pool = self.pools[li]
-            if p + self.B <= pool.numel():
-                self.buf[li] = pool[p : p + self.B]
-                self.pos[li] = p + self.B
+            win = self.wins[li]
+            if p + B <= win:
+                self.buf[li] = self.pools[li, p : p + B]
+                self.pos[li] = p + B
             else:
-                head = pool[p:]
+                head = self.pools[li, p:win].clone()
                 self._refill(li)
-                take = self.B - head.numel()
-                self.buf[li] = torch.cat([head, self.pools[li][:take]])
+                take = B - head.numel()
+                self.buf[li] = torch.cat(
+                    [head, self.pools[li, :take]]
+                )
                 self.pos[li] = take
         return self.buf, self.B, 0
 

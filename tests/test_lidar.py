@@ -94,3 +94,63 @@ def test_interpolate_waypoints():
     assert out.shape == (15, 2)
     # passes through the first waypoint
     assert np.allclose(out[0], [0.0, 0.0])
+
+
+def test_online_window_sampler_fast_path_and_wrap():
+    """_OnlineWindowSampler v2: synchronized equal windows take the
+    zero-copy fast path (returns the pool tensor itself with stride =
+    cap); window wraps advance the dataset and keep every returned
+    index inside the node's current/previous window bounds."""
+    import numpy as np
+    import torch
+
+    from nn_distributed_training_amd.data.floorplan import (
+        synthetic_floorplan,
+        synthetic_waypoints,
+    )
+    from nn_distributed_training_amd.data.lidar import (
+        Lidar2D,
+        OnlineTrajectoryLidarDataset,
+    )
+    from nn_distributed_training_amd.ops.stacked import (
+        _OnlineWindowSampler,
+    )
+
+    np.random.seed(0)
+    img = synthetic_floorplan(nx=96, ny=96, num_walls=3,
+                              border_width=10, seed=0)
+    lidar = Lidar2D(img, 6, 0.25, 8, 1.0, 20, 3)
+    wps = synthetic_waypoints(img, 2, seed=0)
+    dss = [
+        OnlineTrajectoryLidarDataset(lidar, wp, 10, 4) for wp in wps
+    ]
+    win = dss[0].window_bounds[1] - dss[0].window_bounds[0]
+    assert win == dss[1].window_bounds[1] - dss[1].window_bounds[0]
+    B = max(1, win // 3)
+    epochs = []
+    s = _OnlineWindowSampler(dss, B, torch.device("cpu"), 0,
+                             epochs.append)
+
+    # fast path: the pool tensor itself comes back, no copy
+    idx, stride, off = s.next_ref()
+    assert idx is s.pools and stride == s.cap and off == 0
+    idx2, stride2, off2 = s.next_ref()
+    assert idx2 is s.pools and off2 == B
+
+    # run through several windows; every index must stay in-bounds
+    for _ in range(12):
+        idx, stride, off = s.next_ref()
+        for li in range(2):
+            row = idx[li, off : off + B] if idx.shape[1] >= off + B \
+                else idx[li]
+            assert row.numel() == B
+            assert int(row.min()) >= 0
+            assert int(row.max()) < len(dss[li].tds)
+
+    # determinism: same construction gives the same stream
+    np.random.seed(0)
+    dss2 = [
+        OnlineTrajectoryLidarDataset(lidar, wp, 10, 4) for wp in wps
+    ]
+    s2 = _OnlineWindowSampler(dss2, B, torch.device("cpu"), 0,
+                              lambda li: None)

@@ -93,6 +93,11 @@ def test_linear_fwd_sin_relu(ext, dtype, M):
         (1, 4096, 2, 200),   # dw_skinny_i path (FourierNet encode shape)
         (1, 4096, 17, 9),    # dw_small_chunked fallback
         (1, 4096, 64, 48),   # MFMA dw chunked accumulation + fused db
+        (2, 4096, 64, 64),   # dw_direct full-I (density 64->64 shape)
+        (1, 4096, 256, 64),  # dw_direct NIMAX=4 (density 256->64)
+        (1, 2048, 432, 64),  # dw_direct NIMAX=7 (MNIST fc1 shape)
+        (1, 4101, 64, 64),   # dw_direct ragged m-tail (<4 rows)
+        (1, 300, 128, 64),   # dw_direct small-M multi-chunk
     ],
 )
 def test_linear_bwd(ext, dtype, L, M, I, O):
@@ -451,3 +456,24 @@ def test_fused_step_matches_torch_opt(ext, dtype, optname, mode):
             ref_opts[l].step()
             torch.testing.assert_close(theta[l], p.detach(),
                                        **TOL[dtype])
+
+
+@requires_gpu
+def test_feistel_perm(ext):
+    """Keyed bijection kernel: permutation of [lb, lb+n), deterministic
+    by key, distinct keys give distinct orders (sampler shuffle)."""
+    dev = _dev()
+    for n in (1, 7, 64, 50000):
+        out = torch.empty(n, dtype=torch.long, device=dev)
+        ext.feistel_perm(out, n, 100, 12345)
+        vals = out.cpu().sort().values
+        assert torch.equal(vals, torch.arange(100, 100 + n)), n
+    out1 = torch.empty(1000, dtype=torch.long, device=dev)
+    out2 = torch.empty(1000, dtype=torch.long, device=dev)
+    ext.feistel_perm(out1, 1000, 0, 777)
+    ext.feistel_perm(out2, 1000, 0, 777)
+    assert torch.equal(out1, out2)
+    ext.feistel_perm(out2, 1000, 0, 778)
+    assert not torch.equal(out1, out2)
+    # not the identity (it actually shuffles)
+    assert not torch.equal(out1, torch.arange(1000, device=dev))
