@@ -378,29 +378,37 @@ void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
                    c10::optional<torch::Tensor> Yb,
                    c10::optional<torch::Tensor> Zb,
                    long act_below, double scale_below,
-                   long w_off, long M, long I, long O) {
+                   long w_off, long M, long I, long O,
+                   c10::optional<torch::Tensor> Xb2,
+                   long wb_off, long bb_off, long Ib) {
   CHECK_DEV(dZ); CHECK_DEV(dX);
   const long L = theta.size(0), n = theta.size(1);
-  const bool use_mfma = (M >= 128 && I >= 16 && O >= 8);
+  const bool use_mfma =
+      (M >= 128 && I >= 16 && O >= 8 && I % 2 == 0 && O % 2 == 0);
   DISPATCH_FT(dZ, {
     auto ybp = Yb.has_value() ? Yb->data_ptr<scalar_t>() : nullptr;
     auto zbp = Zb.has_value() ? Zb->data_ptr<scalar_t>() : nullptr;
+    auto xb2p = Xb2.has_value() ? Xb2->data_ptr<scalar_t>() : nullptr;
     TORCH_CHECK(act_below == 0 || ybp != nullptr,
                 "act_below needs Yb");
+    TORCH_CHECK(xb2p == nullptr || Ib <= 4,
+                "z-recompute supports below-layer in_dim <= 4");
     if (use_mfma) {
       dim3 grid((I + 63) / 64, (M + 63) / 64, L);
       hipLaunchKernelGGL(gmfma::mfma_dx_k<scalar_t>,
           grid, dim3(256), 0, cur_stream(),
           dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           dX.data_ptr<scalar_t>(), ybp, zbp, (int)act_below,
-          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O);
+          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O,
+          xb2p, wb_off, bb_off, (int)Ib);
     } else {
       dim3 grid((I + 15) / 16, (M + 15) / 16, L);
       hipLaunchKernelGGL(gemm::linear_bwd_dx_k<scalar_t>,
           grid, dim3(16, 16), 0, cur_stream(),
           dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           dX.data_ptr<scalar_t>(), ybp, zbp, (int)act_below,
-          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O);
+          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O,
+          xb2p, wb_off, bb_off, (int)Ib);
     }
   });
   HIP_CHECK_LAST();
@@ -418,23 +426,44 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
     if (M >= 256 && O % 64 == 0 && I % 16 == 0 && I <= 448 && O <= 512) {
       // full-I tiles: dZ and X each fetched from HBM exactly once;
       // direct-global fragment reads (both operands m-row-major), no
-      // LDS/barriers — see gemm_mfma.hip mfma_dw_direct_k rationale
+      // LDS/barriers — see gemm_mfma.hip mfma_dw_direct_k rationale.
+      // Partial tiles go to per-chunk SLABS + a reduce kernel (plain
+      // stores; the nchunk-way atomic burst measured ~2.8x roofline);
+      // NDTA_DW_ATOMIC=1 selects the atomic epilogue for A/B runs.
+      static const bool force_atomic = []() {
+        const char* e = getenv("NDTA_DW_ATOMIC");
+        return e && e[0] == '1';
+      }();
       const long otiles = O / 64;
       long nchunk = std::max<long>(1, 512 / std::max<long>(1, otiles * L));
       nchunk = std::min<long>(nchunk, (M + 63) / 64);
       dim3 grid(1, otiles, L * nchunk);
+      torch::Tensor parts;
+      scalar_t* pp = nullptr;
+      if (!force_atomic) {
+        parts = torch::empty({L * nchunk * O * I}, gstack.options());
+        pp = parts.data_ptr<scalar_t>();
+      }
       if (I <= 256) {
         hipLaunchKernelGGL((gmfma::mfma_dw_direct_k<scalar_t, 4>),
             grid, dim3(512), 0, cur_stream(),
             dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
-            gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+            gstack.data_ptr<scalar_t>(), pp, n, w_off, b_off,
             (int)M, (int)I, (int)O, (int)nchunk);
       } else {
         hipLaunchKernelGGL((gmfma::mfma_dw_direct_k<scalar_t, 7>),
             grid, dim3(512), 0, cur_stream(),
             dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
-            gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+            gstack.data_ptr<scalar_t>(), pp, n, w_off, b_off,
             (int)M, (int)I, (int)O, (int)nchunk);
+      }
+      if (pp != nullptr) {
+        const long tile = O * I;
+        dim3 rgrid(grid_1d(tile), 1, L);
+        hipLaunchKernelGGL(gmfma::dw_reduce_parts_k<scalar_t>,
+            rgrid, dim3(256), 0, cur_stream(),
+            pp, gstack.data_ptr<scalar_t>(), n, w_off, tile,
+            (int)nchunk);
       }
     } else if (M >= 256 && I >= 16 && O >= 16) {
       // fill the chip: tiles * L * nchunk ≈ 2048 blocks
@@ -677,6 +706,152 @@ void regression_bwd(torch::Tensor yhat, torch::Tensor tgt,
   HIP_CHECK_LAST();
 }
 
+// ----------------------------------------------------------- chains --
+// One pybind call per forward / backward pass instead of one per
+// layer-op: host profiling (BENCH r2b timing_breakdown) measured
+// ~10 us of python+pybind overhead PER ext call, making the host the
+// bottleneck of the MNIST round (~23 calls x 10 us vs ~115 us of GPU
+// work). These chains launch exactly the kernels the python loops in
+// ops/stacked.py forward()/backward() launched, in the same order.
+//
+// spec: CPU int64 [nl, 7] rows = (kind, w_off, b_off, in_dim,
+//       out_dim, act, ksize); kind 0 = linear, 1 = conv_pool;
+//       act = ACT_* id (5 = logsoftmax -> linear 'none' + row kernel).
+// scales: CPU float64 [nl] activation scales.
+void fwd_chain(torch::Tensor spec, torch::Tensor scales,
+               c10::optional<torch::Tensor> X_all,
+               c10::optional<torch::Tensor> idx, long idx_stride,
+               long idx_off,
+               torch::Tensor xb, torch::Tensor theta,
+               std::vector<torch::Tensor> acts,
+               std::vector<c10::optional<torch::Tensor>> zs,
+               std::vector<c10::optional<torch::Tensor>> idxs,
+               c10::optional<torch::Tensor> logp, long M,
+               bool train_skip_logp) {
+  TORCH_CHECK(spec.device().is_cpu() && scales.device().is_cpu(),
+              "spec/scales must be CPU tensors");
+  const auto sp = spec.accessor<long, 2>();
+  const auto sc = scales.accessor<double, 1>();
+  const long nl = spec.size(0);
+  if (X_all.has_value()) {
+    gather_batch(*X_all, *idx, xb, idx_stride, idx_off);
+  }
+  torch::Tensor cur = xb;
+  for (long li = 0; li < nl; ++li) {
+    const long kind = sp[li][0], w_off = sp[li][1], b_off = sp[li][2];
+    const long in_dim = sp[li][3], out_dim = sp[li][4];
+    const long act = sp[li][5], ksize = sp[li][6];
+    torch::Tensor out = acts[li];
+    if (kind == 1) {
+      conv_pool_fwd(cur, theta, out, *idxs[li], w_off, b_off, M,
+                    out_dim, ksize, in_dim);
+      cur = out;
+    } else if (act == ACT_LOGSOFTMAX) {
+      linear_fwd(cur, theta, out, c10::nullopt, w_off, b_off, M,
+                 in_dim, out_dim, ACT_NONE, 1.0);
+      if (!train_skip_logp) {
+        logsoftmax(out, *logp, out_dim);
+        cur = *logp;
+      } else {
+        cur = out;
+      }
+    } else {
+      linear_fwd(cur, theta, out, zs[li], w_off, b_off, M, in_dim,
+                 out_dim, act, sc[li]);
+      cur = out;
+    }
+  }
+}
+
+// Backward chain: loss head + layer loop, mirroring
+// ops/stacked.py StackedEngine.backward.
+// loss_kind: 0 = fused NLL (classification), 1 = BCE+sigmoid,
+//            2 = MSE, 3 = L1.
+void bwd_chain(torch::Tensor spec, torch::Tensor scales,
+               torch::Tensor xb, torch::Tensor theta,
+               torch::Tensor grad,
+               std::vector<torch::Tensor> acts,
+               std::vector<c10::optional<torch::Tensor>> zs,
+               std::vector<torch::Tensor> dzs,
+               std::vector<c10::optional<torch::Tensor>> idxs,
+               long loss_kind,
+               c10::optional<torch::Tensor> Y_all,
+               c10::optional<torch::Tensor> idx,
+               c10::optional<torch::Tensor> graph_offs, long pit,
+               long idx_stride, long idx_off,
+               c10::optional<torch::Tensor> yb,
+               c10::optional<torch::Tensor> loss, double loss_scale,
+               long M, long zero_mode) {
+  TORCH_CHECK(spec.device().is_cpu() && scales.device().is_cpu(),
+              "spec/scales must be CPU tensors");
+  const auto sp = spec.accessor<long, 2>();
+  const auto sc = scales.accessor<double, 1>();
+  const long nl = spec.size(0);
+  if (zero_mode == 1) {
+    grad.zero_();
+  } else if (zero_mode == 2) {
+    // conv slices only (kind == 1 rows)
+    for (long li = 0; li < nl; ++li) {
+      if (sp[li][0] == 1) {
+        const long cnt =
+            sp[li][4] * sp[li][6] * sp[li][6] + sp[li][4];
+        grad.index({torch::indexing::Slice(),
+                    torch::indexing::Slice(sp[li][1],
+                                           sp[li][1] + cnt)})
+            .zero_();
+      }
+    }
+  }
+  if (loss.has_value()) loss->zero_();
+
+  const long last = nl - 1;
+  torch::Tensor dz = dzs[last];
+  if (loss_kind == 0) {
+    nll_fused(acts[last], *Y_all, *idx, dz, loss, graph_offs, pit,
+              idx_stride, idx_off, sp[last][4], M, loss_scale);
+  } else if (loss_kind == 1) {
+    bce_bwd(acts[last], *yb, dz, loss, M, loss_scale);
+  } else {
+    auto dy = torch::empty_like(dz);
+    regression_bwd(acts[last], *yb, dy, loss, M, loss_scale,
+                   loss_kind == 2 ? 0 : 1);
+    act_grad(dy, acts[last], c10::nullopt, dz, sp[last][5], sc[last]);
+  }
+
+  for (long li = last; li >= 0; --li) {
+    const long kind = sp[li][0], w_off = sp[li][1], b_off = sp[li][2];
+    const long in_dim = sp[li][3], out_dim = sp[li][4];
+    torch::Tensor below = li > 0 ? acts[li - 1] : xb;
+    dz = dzs[li];
+    if (kind == 1) {
+      conv_pool_bwd(dz, *idxs[li], below, grad, w_off, b_off, M,
+                    out_dim, sp[li][6], in_dim);
+      continue;  // conv is the first layer: no dX
+    }
+    linear_bwd_dw(dz, below, grad, w_off, b_off, M, in_dim, out_dim);
+    if (li > 0) {
+      const long act_b_raw = sp[li - 1][5];
+      const long act_b =
+          (act_b_raw == ACT_NONE || act_b_raw == ACT_LOGSOFTMAX)
+              ? 0 : act_b_raw;
+      c10::optional<torch::Tensor> xb2;
+      long wb_off = 0, bb_off = 0, ib = 0;
+      if (act_b == ACT_SIN_RELU && !zs[li - 1].has_value()) {
+        xb2 = li - 1 > 0 ? acts[li - 2] : xb;
+        wb_off = sp[li - 1][1];
+        bb_off = sp[li - 1][2];
+        ib = sp[li - 1][3];
+      }
+      linear_bwd_dx(
+          dz, theta, dzs[li - 1],
+          act_b ? c10::optional<torch::Tensor>(acts[li - 1])
+                : c10::nullopt,
+          zs[li - 1], act_b, sc[li - 1], w_off, M, in_dim, out_dim,
+          xb2, wb_off, bb_off, ib);
+    }
+  }
+}
+
 // keyed bijection of [0, n) written into `out` (int64, contiguous):
 // the online-density sampler's shuffle (ops/stacked.py
 // _OnlineWindowSampler) — one launch instead of a device randperm's
@@ -724,4 +899,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bce_bwd", &bce_bwd);
   mod.def("regression_bwd", &regression_bwd);
   mod.def("feistel_perm", &feistel_perm);
+  mod.def("fwd_chain", &fwd_chain);
+  mod.def("bwd_chain", &bwd_chain);
 }

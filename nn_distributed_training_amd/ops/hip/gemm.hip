@@ -110,13 +110,16 @@ __global__ void act_grad_k(
 }
 
 // dX[M,I] = dZ[M,O] @ W[O,I], below-layer activation bwd fused
+// (Xb2 != null: recompute z_below from the below layer's tiny input
+// instead of reading Zb — see gemm_mfma.hip mfma_dx_k)
 template <typename T>
 __global__ void linear_bwd_dx_k(
     const T* __restrict__ dZ, const T* __restrict__ theta,
     T* __restrict__ dX,
     const T* __restrict__ Yb, const T* __restrict__ Zb,
     int act_below, T scale_below,
-    long n, long w_off, int M, int I, int O) {
+    long n, long w_off, int M, int I, int O,
+    const T* __restrict__ Xb2, long wb_off, long bb_off, int Ib) {
   __shared__ T gs[TILE][TILE + 1];
   __shared__ T ws[TILE][TILE + 1];
   const long l = blockIdx.z;
@@ -146,8 +149,21 @@ __global__ void linear_bwd_dx_k(
   if (m < M && i < I) {
     const long off = l * (long)M * I + (long)m * I + i;
     if (act_below != ACT_NONE) {
-      acc *= act_bwd(act_below, Zb ? Zb[off] : T(0), Yb[off],
-                     scale_below);
+      T z = T(0);
+      if (Xb2 != nullptr) {
+        const T* Wb = theta + l * n + wb_off;
+        z = theta[l * n + bb_off + i];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          if (j < Ib) {
+            z += Xb2[(long)(l * (long)M + m) * Ib + j]
+                 * Wb[(long)i * Ib + j];
+          }
+        }
+      } else if (Zb != nullptr) {
+        z = Zb[off];
+      }
+      acc *= act_bwd(act_below, z, Yb ? Yb[off] : T(0), scale_below);
     }
     dX[off] = acc;
   }

@@ -215,17 +215,28 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
 // BELOW layer's activation backward fused into the epilogue
 // (dX *= act'(z_below, y_below)): the separate memory-bound act_grad
 // pass was ~14% of the density round.
+//
+// Round-2: same T14 write-after-barrier register pipeline + BK=32 +
+// vec2 staging as mfma_fwd_k, and an optional RECOMPUTE mode for the
+// below layer's pre-activation: when the below layer has a tiny input
+// (the FourierNet encode, Ib = 2), z_below is recomputed from
+// Xb2 @ W_below^T + b_below in the epilogue (4 flops + L1-resident
+// reads) instead of reading a full [M, I] Zb tensor back from HBM
+// (328 MB/call in the density config).
 template <typename T>
 __global__ __launch_bounds__(256) void mfma_dx_k(
     const T* __restrict__ dZ, const T* __restrict__ theta,
     T* __restrict__ dX,
     const T* __restrict__ Yb, const T* __restrict__ Zb,  // may be null
     int act_below, T scale_below,
-    long n, long w_off, int M, int I, int O) {
+    long n, long w_off, int M, int I, int O,
+    const T* __restrict__ Xb2,  // below layer INPUT (recompute mode)
+    long wb_off, long bb_off, int Ib) {
   using MF = mfma_t<T>;
   using acc_t = typename MF::acc_t;
-  __shared__ T As[BK][BM + 1];   // As[k=o][m] = dZ[m][o]
-  __shared__ T Bs[BK][BN + 1];   // Bs[k=o][i] = W[o][i]
+  typedef T vec2 __attribute__((ext_vector_type(2)));
+  __shared__ T As[FK][BM + 1];   // As[k=o][m] = dZ[m][o]
+  __shared__ T Bs[FK][BN + 1];   // Bs[k=o][i] = W[o][i]
 
   const long l = blockIdx.z;
   const T* Gl = dZ + l * (long)M * O;
@@ -239,24 +250,103 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
   const int wm = (wid >> 1) * 32;
   const int wn = (wid & 1) * 32;
 
+  const int sm = tid / 16;   // staged row (m for A, o-row for B image)
+  const int skp = tid % 16;  // k-pair within the row
   acc_t acc[2][2] = {};
+  vec2 ra[4];
 
-  for (int k0 = 0; k0 < O; k0 += BK) {
-    for (int t = tid; t < BM * BK; t += 256) {
-      const int m = t / BK, k = t % BK;
-      As[k][m] = (m0 + m < M && k0 + k < O)
-                     ? Gl[(long)(m0 + m) * O + (k0 + k)]
-                     : T(0);
+  // A: As[k][m] = dZ[m0+m][k]  (k = contraction over O, row-major in
+  // dZ so pairs are contiguous); B: Bs[k][i] = W[k][i0+i] (W rows
+  // contiguous in i).
+  const int nstages = (O + FK - 1) / FK;
+  // stage A via vec2 over k (pairs contiguous in dZ rows); stage B
+  // with vec2 over i (pairs contiguous in W rows): thread covers W
+  // rows bo + q*8 (q < 4 -> 32 k-rows) at i-pair bip.
+  const int bo = tid / 32;    // W row (o) of pair 0 for B staging
+  const int bip = tid % 32;   // i-pair within the row
+  vec2 rb2[4];                // 4 rows per thread over the 32-k stage
+
+  // ---- load stage 0
+  {
+    const int k0 = 0;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      const int k = k0 + 2 * skp;
+      ra[q] = vec2{0, 0};
+      if (m0 + m < M) {
+        if (k + 1 < O) {
+          ra[q] = *reinterpret_cast<const vec2*>(
+              &Gl[(long)(m0 + m) * O + k]);
+        } else if (k < O) {
+          ra[q].x = Gl[(long)(m0 + m) * O + k];
+        }
+      }
     }
-    for (int t = tid; t < BN * BK; t += 256) {
-      const int i = t % BN, k = t / BN;
-      Bs[k][i] = (k0 + k < O && i0 + i < I)
-                     ? W[(long)(k0 + k) * I + (i0 + i)]
-                     : T(0);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int o = k0 + bo + q * 8;
+      const int i = i0 + 2 * bip;
+      rb2[q] = vec2{0, 0};
+      if (o < O) {
+        if (i + 1 < I) {
+          rb2[q] = *reinterpret_cast<const vec2*>(
+              &W[(long)o * I + i]);
+        } else if (i < I) {
+          rb2[q].x = W[(long)o * I + i];
+        }
+      }
+    }
+  }
+
+  for (int s = 0; s < nstages; ++s) {
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      As[2 * skp][m] = ra[q].x;
+      As[2 * skp + 1][m] = ra[q].y;
+    }
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int kb = bo + q * 8;
+      Bs[kb][2 * bip] = rb2[q].x;
+      Bs[kb][2 * bip + 1] = rb2[q].y;
+    }
+    if (s + 1 < nstages) {
+      const int k0 = (s + 1) * FK;
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int m = sm + q * 16;
+        const int k = k0 + 2 * skp;
+        ra[q] = vec2{0, 0};
+        if (m0 + m < M) {
+          if (k + 1 < O) {
+            ra[q] = *reinterpret_cast<const vec2*>(
+                &Gl[(long)(m0 + m) * O + k]);
+          } else if (k < O) {
+            ra[q].x = Gl[(long)(m0 + m) * O + k];
+          }
+        }
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int o = k0 + bo + q * 8;
+        const int i = i0 + 2 * bip;
+        rb2[q] = vec2{0, 0};
+        if (o < O) {
+          if (i + 1 < I) {
+            rb2[q] = *reinterpret_cast<const vec2*>(
+                &W[(long)o * I + i]);
+          } else if (i < I) {
+            rb2[q].x = W[(long)o * I + i];
+          }
+        }
+      }
     }
     __syncthreads();
 #pragma unroll
-    for (int kk = 0; kk < BK; kk += 4) {
+    for (int kk = 0; kk < FK; kk += 4) {
       const int ka = kk + (lane >> 4);
 #pragma unroll
       for (int fm = 0; fm < 2; ++fm) {
@@ -268,9 +358,10 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
         }
       }
     }
-    __syncthreads();
   }
 
+  const T* Wb = theta + l * n + wb_off;
+  const T* bb = theta + l * n + bb_off;
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
@@ -283,7 +374,20 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
           const long off = l * (long)M * I + (long)m * I + i;
           T v = acc[fm][fn][r];
           if (act_below != ACT_NONE) {
-            v *= act_bwd(act_below, Zb ? Zb[off] : T(0), Yb[off],
+            T z = T(0);
+            if (Xb2 != nullptr) {  // recompute z_below (tiny Ib)
+              z = bb[i];
+#pragma unroll
+              for (int j = 0; j < 4; ++j) {
+                if (j < Ib) {
+                  z += Xb2[(long)(l * (long)M + m) * Ib + j]
+                       * Wb[(long)i * Ib + j];
+                }
+              }
+            } else if (Zb != nullptr) {
+              z = Zb[off];
+            }
+            v *= act_bwd(act_below, z, Yb ? Yb[off] : T(0),
                          scale_below);
           }
           dX[off] = v;
@@ -407,10 +511,17 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
 // flight per SIMD against the ~900-cycle HBM latency.
 // db is accumulated by the i0==0 wave pair from its own a-loads
 // (each (m, o) element passes through exactly one (lane, k-class)).
+// parts != null: each block stores its [64, I] partial tile to
+// parts[(l*nchunk + chunk)*O*I + (o0+o)*I + i] with PLAIN coalesced
+// stores instead of nchunk-way atomicAdd contention on gstack (the
+// atomic burst measured ~2.8x the kernel's traffic roofline —
+// profiles r2b); dw_reduce_parts_k folds the chunks afterwards.
+// db stays atomic either way (64 adds/block).
 template <typename T, int NIMAX>
 __global__ __launch_bounds__(512) void mfma_dw_direct_k(
     const T* __restrict__ dZ, const T* __restrict__ X,
-    T* __restrict__ gstack, long n, long w_off, long b_off,
+    T* __restrict__ gstack, T* __restrict__ parts,
+    long n, long w_off, long b_off,
     int M, int I, int O, int nchunk) {
   using MF = mfma_t<T>;
   using acc_t = typename MF::acc_t;
@@ -449,32 +560,60 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
 
   int k = mlo;
   if (ni > 0) {
-    for (; k + 4 <= mhi; k += 4) {
+    // software-pipelined: the NEXT k-step's operands load while the
+    // current step's MFMAs issue (one step of lookahead keeps ~2x the
+    // loads in flight against the ~900-cycle HBM latency)
+    bool have = k + 4 <= mhi;
+    T a0 = T(0), a1 = T(0), b[NIMAX] = {};
+    if (have) {
       const long ka = k + lk;
-      const T a0 = ga0[ka * O];
-      const T a1 = ga1[ka * O];
+      a0 = ga0[ka * O];
+      a1 = ga1[ka * O];
+#pragma unroll
+      for (int fi = 0; fi < NIMAX; ++fi) {
+        if (fi < ni) b[fi] = gb[ka * I + fi * 16];
+      }
+    }
+    while (have) {
+      const int kn = k + 4;
+      const bool haven = kn + 4 <= mhi;
+      T a0n = T(0), a1n = T(0), bn[NIMAX] = {};
+      if (haven) {
+        const long kan = kn + lk;
+        a0n = ga0[kan * O];
+        a1n = ga1[kan * O];
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) {
+          if (fi < ni) bn[fi] = gb[kan * I + fi * 16];
+        }
+      }
       if (bias_wave) { db0 += a0; db1 += a1; }
 #pragma unroll
       for (int fi = 0; fi < NIMAX; ++fi) {
         if (fi < ni) {
-          const T b = gb[ka * I + fi * 16];
-          acc[0][fi] = MF::mma(a0, b, acc[0][fi]);
-          acc[1][fi] = MF::mma(a1, b, acc[1][fi]);
+          acc[0][fi] = MF::mma(a0, b[fi], acc[0][fi]);
+          acc[1][fi] = MF::mma(a1, b[fi], acc[1][fi]);
         }
       }
+      a0 = a0n;
+      a1 = a1n;
+#pragma unroll
+      for (int fi = 0; fi < NIMAX; ++fi) b[fi] = bn[fi];
+      k = kn;
+      have = haven;
     }
     if (k < mhi) {  // ragged tail (< 4 rows): zero-padded operands
       const long ka = k + lk;
       const bool ok = ka < mhi;
-      const T a0 = ok ? ga0[ka * O] : T(0);
-      const T a1 = ok ? ga1[ka * O] : T(0);
-      if (bias_wave) { db0 += a0; db1 += a1; }
+      const T ta0 = ok ? ga0[ka * O] : T(0);
+      const T ta1 = ok ? ga1[ka * O] : T(0);
+      if (bias_wave) { db0 += ta0; db1 += ta1; }
 #pragma unroll
       for (int fi = 0; fi < NIMAX; ++fi) {
         if (fi < ni) {
-          const T b = ok ? gb[ka * I + fi * 16] : T(0);
-          acc[0][fi] = MF::mma(a0, b, acc[0][fi]);
-          acc[1][fi] = MF::mma(a1, b, acc[1][fi]);
+          const T tb = ok ? gb[ka * I + fi * 16] : T(0);
+          acc[0][fi] = MF::mma(ta0, tb, acc[0][fi]);
+          acc[1][fi] = MF::mma(ta1, tb, acc[1][fi]);
         }
       }
     }
@@ -503,20 +642,54 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
     }
   }
 
+  if (parts != nullptr) {
+    T* slab = parts + ((long)blockIdx.z) * (long)O * I;
 #pragma unroll
-  for (int fo = 0; fo < 2; ++fo) {
+    for (int fo = 0; fo < 2; ++fo) {
 #pragma unroll
-    for (int fi = 0; fi < NIMAX; ++fi) {
-      if (fi < ni) {
+      for (int fi = 0; fi < NIMAX; ++fi) {
+        if (fi < ni) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int o = o0 + wo + fo * 16 + MF::acc_row(lane, r);
-          const int i = i0 + fi * 16 + lo;
-          atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
-                    acc[fo][fi][r]);
+          for (int r = 0; r < 4; ++r) {
+            const int o = o0 + wo + fo * 16 + MF::acc_row(lane, r);
+            const int i = i0 + fi * 16 + lo;
+            slab[(long)o * I + i] = acc[fo][fi][r];
+          }
         }
       }
     }
+  } else {
+#pragma unroll
+    for (int fo = 0; fo < 2; ++fo) {
+#pragma unroll
+      for (int fi = 0; fi < NIMAX; ++fi) {
+        if (fi < ni) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int o = o0 + wo + fo * 16 + MF::acc_row(lane, r);
+            const int i = i0 + fi * 16 + lo;
+            atomicAdd(&gstack[l * n + w_off + (long)o * I + i],
+                      acc[fo][fi][r]);
+          }
+        }
+      }
+    }
+  }
+}
+
+// folds dw_direct's per-chunk slabs into the grad stack (WRITE —
+// the stack's w slice needs no pre-zero on this path)
+template <typename T>
+__global__ void dw_reduce_parts_k(
+    const T* __restrict__ parts, T* __restrict__ gstack, long n,
+    long w_off, long tile, int nchunk) {
+  const long l = blockIdx.z;
+  const T* base = parts + l * (long)nchunk * tile;
+  for (long e = blockIdx.x * (long)blockDim.x + threadIdx.x; e < tile;
+       e += (long)gridDim.x * blockDim.x) {
+    T sum = T(0);
+    for (int c = 0; c < nchunk; ++c) sum += base[c * tile + e];
+    gstack[l * n + w_off + e] = sum;
   }
 }
 

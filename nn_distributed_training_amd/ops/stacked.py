@@ -333,6 +333,14 @@ class StackedEngine:
         self._zero_plan = None
         self._bufs = None
         self._loss_kind = type(problem.base_loss).__name__  # NLLLoss etc.
+        # spec tables for the C++ fwd/bwd chains (ONE pybind call per
+        # pass instead of one per layer-op: ~10 us of host overhead per
+        # ext call made the host the MNIST round bottleneck — see
+        # ext.hip fwd_chain). NDTA_PY_CHAIN=1 selects the python
+        # per-op loops (debug / A-B reference).
+        self._use_chain = os.environ.get("NDTA_PY_CHAIN", "0") != "1"
+        self._spec_t = None
+        self._scales_t = None
         # device-resident train-loss EMA (the reference's tloss tracker,
         # problems/dist_online_dense_problem.py:129-137, without a
         # host sync per round)
@@ -408,9 +416,13 @@ class StackedEngine:
         acts, zs, dzs, idxs = [], [], [], []
         for layer in self.spec.layers:
             acts.append(mk(layer.out_elems))
+            # sin_relu needs z in backward, but for a tiny in_dim the
+            # dx epilogue RECOMPUTES it (4 flops) instead of storing /
+            # re-reading an [L*M, out] tensor (kernel z-recompute mode)
             zs.append(
                 mk(layer.out_elems)
-                if (train and layer.activation == "sin_relu")
+                if (train and layer.activation == "sin_relu"
+                    and layer.in_dim > 4)
                 else None
             )
             dzs.append(mk(layer.out_elems) if train else None)
@@ -518,6 +530,21 @@ class StackedEngine:
         return xb, yb
 
     # ------------------------------------------------------------------
+    def _chain_meta(self):
+        if self._spec_t is None:
+            rows, scales = [], []
+            for layer in self.spec.layers:
+                rows.append([
+                    1 if layer.kind == "conv_pool" else 0,
+                    layer.w_off, layer.b_off, layer.in_dim,
+                    layer.out_dim, ACT_IDS[layer.activation],
+                    getattr(layer, "kernel_size", 0) or 0,
+                ])
+                scales.append(float(getattr(layer, "scale", 1.0)))
+            self._spec_t = torch.tensor(rows, dtype=torch.long)
+            self._scales_t = torch.tensor(scales, dtype=torch.float64)
+        return self._spec_t, self._scales_t
+
     def forward(self, xb, bufs=None, train_skip_logp=False):
         if self._bufs is None:
             self._bufs = self._alloc_bufs()
@@ -525,6 +552,17 @@ class StackedEngine:
         ext = self.ext
         cur = xb
         M = bufs["M"]
+        if self._use_chain:
+            spec_t, scales_t = self._chain_meta()
+            ext.fwd_chain(
+                spec_t, scales_t, None, None, 0, 0, xb, self.theta,
+                bufs["acts"], bufs["zs"], bufs["idxs"], bufs["logp"],
+                M, train_skip_logp,
+            )
+            last = self.spec.layers[-1]
+            if last.activation == "logsoftmax" and not train_skip_logp:
+                return bufs["logp"]
+            return bufs["acts"][-1]
         for li, layer in enumerate(self.spec.layers):
             out = bufs["acts"][li]
             if layer.kind == "conv_pool":
@@ -586,6 +624,43 @@ class StackedEngine:
                       and layer.out_dim >= 16) or M > 2048:
                     whole = True
             self._zero_plan = ("whole",) if whole else ("slices", slices)
+        if self._use_chain:
+            zero_mode = 1 if self._zero_plan[0] == "whole" else 2
+            loss_buf = bufs["loss"] if want_loss else None
+            last = layers[-1]
+            if self.classification:
+                idx_t = (
+                    self.sampler.stream
+                    if hasattr(self.sampler, "stream")
+                    else self.sampler.buf
+                )
+                stride = (
+                    self.sampler.S
+                    if hasattr(self.sampler, "S") else self.B
+                )
+                loss_kind = 0
+                yb_arg = None
+                idx_off = 0 if graph_offs is not None else self._last_off
+            else:
+                idx_t, stride, idx_off = None, 0, 0
+                if (self._loss_kind == "BCELoss"
+                        and last.activation == "sigmoid"):
+                    loss_kind = 1
+                elif self._loss_kind == "MSELoss":
+                    loss_kind = 2
+                else:
+                    loss_kind = 3
+                yb_arg = yb
+            spec_t, scales_t = self._chain_meta()
+            ext.bwd_chain(
+                spec_t, scales_t, xb, self.theta, self.grad,
+                bufs["acts"], bufs["zs"], bufs["dzs"], bufs["idxs"],
+                loss_kind, self.Y_all if self.classification else None,
+                idx_t, graph_offs, pit, stride, idx_off, yb_arg,
+                loss_buf, loss_scale, M, zero_mode,
+            )
+            return loss_buf if want_loss else None
+
         if self._zero_plan[0] == "whole":
             self.grad.zero_()
         else:
@@ -649,7 +724,12 @@ class StackedEngine:
             )
             if li > 0:
                 # dX with the below layer's activation bwd fused into
-                # the epilogue (no separate act_grad pass)
+                # the epilogue (no separate act_grad pass). For a
+                # sin_relu below layer with tiny in_dim (FourierNet
+                # encode), z_below is RECOMPUTED from the below input
+                # in the epilogue — no [M, I] Zb tensor is stored in
+                # forward or read back here (2x 328 MB/round saved in
+                # the density config).
                 dz_below = bufs["dzs"][li - 1]
                 lb = layers[li - 1]
                 act_b = (
@@ -657,11 +737,18 @@ class StackedEngine:
                     if lb.activation not in ("none", "logsoftmax")
                     else 0
                 )
+                xb2, wb_off, bb_off, ib = None, 0, 0, 0
+                if (act_b == ACT_IDS["sin_relu"]
+                        and bufs["zs"][li - 1] is None):
+                    xb2 = bufs["acts"][li - 2] if li - 1 > 0 else xb
+                    wb_off, bb_off = lb.w_off, lb.b_off
+                    ib = lb.in_dim
                 ext.linear_bwd_dx(
                     dz, self.theta, dz_below,
                     bufs["acts"][li - 1] if act_b else None,
                     bufs["zs"][li - 1], act_b, lb.scale,
                     layer.w_off, M, layer.in_dim, layer.out_dim,
+                    xb2, wb_off, bb_off, ib,
                 )
         return bufs["loss"] if want_loss else None
 

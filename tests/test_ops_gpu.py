@@ -124,7 +124,8 @@ def test_linear_bwd(ext, dtype, L, M, I, O):
     dZ = torch.empty_like(dY)
     ext.act_grad(dY, Y, None, dZ, 1, 1.0)
     dX = torch.empty(L * M, I, dtype=dtype, device=dev)
-    ext.linear_bwd_dx(dZ, theta, dX, None, None, 0, 1.0, 0, M, I, O)
+    ext.linear_bwd_dx(dZ, theta, dX, None, None, 0, 1.0, 0, M, I, O,
+                      None, 0, 0, 0)
     gstack = torch.zeros_like(theta)
     ext.linear_bwd_dw(dZ, X.detach(), gstack, 0, I * O, M, I, O)
 
@@ -477,3 +478,33 @@ def test_feistel_perm(ext):
     assert not torch.equal(out1, out2)
     # not the identity (it actually shuffles)
     assert not torch.equal(out1, torch.arange(1000, device=dev))
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+@pytest.mark.parametrize("M", [40, 2048])  # VALU vs MFMA dx paths
+def test_linear_bwd_dx_sin_relu_recompute(ext, dtype, M):
+    """dx with the encode layer's z RECOMPUTED from its tiny input
+    (Xb2 @ Wb^T + bb) must match dx with the explicitly stored Zb."""
+    torch.manual_seed(5)
+    L, Ib, I, O = 2, 2, 24, 16  # below: 2->24 sin_relu; above: 24->16
+    dev = _dev()
+    scale = 0.05
+    nb = Ib * I + I           # below layer params (W1 [I, Ib], b1 [I])
+    n = nb + I * O + O        # + above layer
+    theta = torch.randn(L, n, dtype=dtype, device=dev)
+    Xb2 = torch.randn(L * M, Ib, dtype=dtype, device=dev)
+
+    # below layer forward (explicit z for the reference path)
+    Yb = torch.empty(L * M, I, dtype=dtype, device=dev)
+    Zb = torch.empty_like(Yb)
+    ext.linear_fwd(Xb2, theta, Yb, Zb, 0, Ib * I, M, Ib, I, 2, scale)
+
+    dZ = torch.randn(L * M, O, dtype=dtype, device=dev)
+    ref = torch.empty(L * M, I, dtype=dtype, device=dev)
+    ext.linear_bwd_dx(dZ, theta, ref, Yb, Zb, 2, scale, nb, M, I, O,
+                      None, 0, 0, 0)
+    out = torch.empty_like(ref)
+    ext.linear_bwd_dx(dZ, theta, out, Yb, None, 2, scale, nb, M, I, O,
+                      Xb2, 0, Ib * I, Ib)
+    torch.testing.assert_close(out, ref, **TOL[dtype])
