@@ -393,9 +393,21 @@ void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
                 "act_below needs Yb");
     TORCH_CHECK(xb2p == nullptr || Ib <= 4,
                 "z-recompute supports below-layer in_dim <= 4");
-    if (use_mfma) {
+    static const bool dx_new = []() {
+      const char* e = getenv("NDTA_DX_NEW");
+      return !(e && e[0] == '0');
+    }();
+    if (use_mfma && dx_new) {
       dim3 grid((I + 63) / 64, (M + 63) / 64, L);
       hipLaunchKernelGGL(gmfma::mfma_dx_k<scalar_t>,
+          grid, dim3(256), 0, cur_stream(),
+          dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          dX.data_ptr<scalar_t>(), ybp, zbp, (int)act_below,
+          (scalar_t)scale_below, n, w_off, (int)M, (int)I, (int)O,
+          xb2p, wb_off, bb_off, (int)Ib);
+    } else if (use_mfma) {
+      dim3 grid((I + 63) / 64, (M + 63) / 64, L);
+      hipLaunchKernelGGL(gmfma::mfma_dx16_k<scalar_t>,
           grid, dim3(256), 0, cur_stream(),
           dZ.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           dX.data_ptr<scalar_t>(), ybp, zbp, (int)act_below,
@@ -706,6 +718,33 @@ void regression_bwd(torch::Tensor yhat, torch::Tensor tgt,
   HIP_CHECK_LAST();
 }
 
+// one-launch fc block: fc1+fc2 fwd, NLL, full fc backward (see
+// fused_mnist.hip fc_block_k). Caller pre-zeroes the grad stack.
+void fc_block(torch::Tensor x0, torch::Tensor theta,
+              torch::Tensor Y_all, torch::Tensor idx, long idx_stride,
+              long idx_off, torch::Tensor grad, torch::Tensor dx0,
+              c10::optional<torch::Tensor> loss, long w1_off,
+              long b1_off, long w2_off, long b2_off, long M, long I,
+              long H, long C, double loss_scale) {
+  CHECK_DEV(x0); CHECK_DEV(theta); CHECK_DEV(grad); CHECK_DEV(dx0);
+  const long L = theta.size(0), n = theta.size(1);
+  const long maxlen = Y_all.size(1);
+  TORCH_CHECK(H <= 64 && C <= 16, "fc_block: H <= 64, C <= 16");
+  const long mtiles = (M + fmnist::FC_RT - 1) / fmnist::FC_RT;
+  DISPATCH_FT(x0, {
+    hipLaunchKernelGGL(fmnist::fc_block_k<scalar_t>,
+        dim3(mtiles, 1, L), dim3(256), 0, cur_stream(),
+        x0.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+        Y_all.data_ptr<long>(), idx.data_ptr<long>(), idx_stride,
+        idx_off, maxlen, grad.data_ptr<scalar_t>(),
+        dx0.data_ptr<scalar_t>(),
+        loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr,
+        n, w1_off, b1_off, w2_off, b2_off, (int)M, (int)I, (int)H,
+        (int)C, (scalar_t)loss_scale);
+  });
+  HIP_CHECK_LAST();
+}
+
 // ----------------------------------------------------------- chains --
 // One pybind call per forward / backward pass instead of one per
 // layer-op: host profiling (BENCH r2b timing_breakdown) measured
@@ -899,6 +938,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bce_bwd", &bce_bwd);
   mod.def("regression_bwd", &regression_bwd);
   mod.def("feistel_perm", &feistel_perm);
+  mod.def("fc_block", &fc_block);
   mod.def("fwd_chain", &fwd_chain);
   mod.def("bwd_chain", &bwd_chain);
 }

@@ -511,6 +511,109 @@ __global__ __launch_bounds__(256) void mfma_dw_k(
 // flight per SIMD against the ~900-cycle HBM latency.
 // db is accumulated by the i0==0 wave pair from its own a-loads
 // (each (m, o) element passes through exactly one (lane, k-class)).
+// ---------------------------------------------------------------------
+// BK=16 single-buffered dx variant (round-1 structure) kept for A/B:
+// the T14 rework above REGRESSED dx at O=64 (199 -> 228 us avg,
+// profiles/dens_r2d_topk.txt) — only 2 stages leave no pipeline to
+// fill. NDTA_DX_NEW=0 selects this kernel.
+// NN: dX[M,I] = dZ[M,O] @ W_l[O,I]   (contraction over O), with the
+// BELOW layer's activation backward fused into the epilogue
+// (dX *= act'(z_below, y_below)): the separate memory-bound act_grad
+// pass was ~14% of the density round.
+template <typename T>
+__global__ __launch_bounds__(256) void mfma_dx16_k(
+    const T* __restrict__ dZ, const T* __restrict__ theta,
+    T* __restrict__ dX,
+    const T* __restrict__ Yb, const T* __restrict__ Zb,  // may be null
+    int act_below, T scale_below,
+    long n, long w_off, int M, int I, int O,
+    const T* __restrict__ Xb2, long wb_off, long bb_off, int Ib) {
+  using MF = mfma_t<T>;
+  using acc_t = typename MF::acc_t;
+  __shared__ T As[BK][BM + 1];   // As[k=o][m] = dZ[m][o]
+  __shared__ T Bs[BK][BN + 1];   // Bs[k=o][i] = W[o][i]
+
+  const long l = blockIdx.z;
+  const T* Gl = dZ + l * (long)M * O;
+  const T* W = theta + l * n + w_off;
+
+  const int m0 = blockIdx.y * BM;
+  const int i0 = blockIdx.x * BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int wm = (wid >> 1) * 32;
+  const int wn = (wid & 1) * 32;
+
+  acc_t acc[2][2] = {};
+
+  for (int k0 = 0; k0 < O; k0 += BK) {
+    for (int t = tid; t < BM * BK; t += 256) {
+      const int m = t / BK, k = t % BK;
+      As[k][m] = (m0 + m < M && k0 + k < O)
+                     ? Gl[(long)(m0 + m) * O + (k0 + k)]
+                     : T(0);
+    }
+    for (int t = tid; t < BN * BK; t += 256) {
+      const int i = t % BN, k = t / BN;
+      Bs[k][i] = (k0 + k < O && i0 + i < I)
+                     ? W[(long)(k0 + k) * I + (i0 + i)]
+                     : T(0);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 4) {
+      const int ka = kk + (lane >> 4);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        const T a = As[ka][wm + fm * 16 + (lane & 15)];
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
+          acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
+        const int i = i0 + wn + fn * 16 + (lane & 15);
+        if (m < M && i < I) {
+          const long off = l * (long)M * I + (long)m * I + i;
+          T v = acc[fm][fn][r];
+          if (act_below != ACT_NONE) {
+            T z = T(0);
+            if (Xb2 != nullptr) {
+              const T* Wb = theta + l * n + wb_off;
+              z = theta[l * n + bb_off + i];
+#pragma unroll
+              for (int j = 0; j < 4; ++j) {
+                if (j < Ib) {
+                  z += Xb2[(long)(l * (long)M + m) * Ib + j]
+                       * Wb[(long)i * Ib + j];
+                }
+              }
+            } else if (Zb != nullptr) {
+              z = Zb[off];
+            }
+            v *= act_bwd(act_below, z, Yb ? Yb[off] : T(0),
+                         scale_below);
+          }
+          dX[off] = v;
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
 // parts != null: each block stores its [64, I] partial tile to
 // parts[(l*nchunk + chunk)*O*I + (o0+o)*I + i] with PLAIN coalesced
 // stores instead of nchunk-way atomicAdd contention on gstack (the

@@ -601,6 +601,71 @@ class StackedEngine:
         return cur
 
     # ------------------------------------------------------------------
+    def fc_block_applicable(self):
+        """One-launch fused fc block (fused_mnist.hip fc_block_k):
+        conv -> [fc1+fc2 fwd + NLL + full fc bwd] -> conv bwd.
+        Replaces 7 floor-bound small kernels per primal iteration
+        (~65 us -> ~15 us of GPU time at the MNIST bench shapes)."""
+        if os.environ.get("NDTA_FC_BLOCK", "1") == "0":
+            return False
+        ls = self.spec.layers
+        return (
+            self.device.type == "cuda"
+            and self.classification
+            and len(ls) == 3
+            and ls[0].kind == "conv_pool"
+            and ls[1].kind == "linear"
+            and ls[2].kind == "linear"
+            and ls[1].activation == "relu"
+            and ls[2].activation == "logsoftmax"
+            and ls[1].out_dim <= 64
+            and ls[2].out_dim <= 16
+        )
+
+    def fwd_bwd(self, xb, yb, loss_scale=1.0, want_loss=False,
+                graph_offs=None, pit=0):
+        """Forward + backward for one primal iteration; takes the fused
+        fc-block path when the model/loss shape allows it."""
+        if graph_offs is None and self.fc_block_applicable():
+            bufs = self._bufs
+            conv, fc1, fc2 = self.spec.layers
+            M = self.B
+            ext = self.ext
+            ext.conv_pool_fwd(
+                xb, self.theta, bufs["acts"][0], bufs["idxs"][0],
+                conv.w_off, conv.b_off, M, conv.out_dim,
+                conv.kernel_size, conv.in_dim,
+            )
+            # fc grads accumulate atomically: zero the whole stack
+            # (~1.8 MB — same cost as the conv-slice fill it replaces)
+            self.grad.zero_()
+            loss_buf = None
+            if want_loss:
+                bufs["loss"].zero_()
+                loss_buf = bufs["loss"]
+            idx_t = self.sampler.stream if hasattr(
+                self.sampler, "stream") else self.sampler.buf
+            stride = self.sampler.S if hasattr(
+                self.sampler, "S") else self.B
+            ext.fc_block(
+                bufs["acts"][0], self.theta, self.Y_all, idx_t,
+                stride, self._last_off, self.grad, bufs["dzs"][0],
+                loss_buf, fc1.w_off, fc1.b_off, fc2.w_off, fc2.b_off,
+                M, fc1.in_dim, fc1.out_dim, fc2.out_dim, loss_scale,
+            )
+            ext.conv_pool_bwd(
+                bufs["dzs"][0], bufs["idxs"][0], xb, self.grad,
+                conv.w_off, conv.b_off, M, conv.out_dim,
+                conv.kernel_size, conv.in_dim,
+            )
+            return loss_buf if want_loss else None
+        self.forward(xb, train_skip_logp=True)
+        return self.backward(
+            xb, yb, loss_scale=loss_scale, want_loss=want_loss,
+            graph_offs=graph_offs, pit=pit,
+        )
+
+    # ------------------------------------------------------------------
     def backward(self, xb, yb, loss_scale=1.0, want_loss=False,
                  graph_offs=None, pit=0):
         bufs = self._bufs
@@ -1001,8 +1066,7 @@ class DiNNOStackedDriver:
                     eng.X_all, eng.sampler.stream, xb, self._offs_dev,
                     pit, eng.sampler.S,
                 )
-                eng.forward(xb, train_skip_logp=True)
-                eng.backward(
+                eng.fwd_bwd(
                     xb, None, graph_offs=self._offs_dev, pit=pit
                 )
                 grad_t, nparts = eng.grad, 1
@@ -1116,10 +1180,8 @@ class DiNNOStackedDriver:
             else:
                 with _timer("next_batch"):
                     xb, yb = eng.next_batch()
-                with _timer("forward"):
-                    eng.forward(xb, train_skip_logp=True)
-                with _timer("backward"):
-                    lb = eng.backward(xb, yb, want_loss=wl)
+                with _timer("fwd_bwd"):
+                    lb = eng.fwd_bwd(xb, yb, want_loss=wl)
                 grad_t, nparts = eng.grad, 1
             if wl and lb is not None:
                 eng.update_tloss(lb)
@@ -1201,8 +1263,7 @@ class DSGDStackedDriver:
             eng.reduce_fused_grad()
         else:
             xb, yb = eng.next_batch()
-            eng.forward(xb, train_skip_logp=True)
-            lb = eng.backward(xb, yb, want_loss=want_tl)
+            lb = eng.fwd_bwd(xb, yb, want_loss=want_tl)
         if want_tl and lb is not None:
             eng.update_tloss(lb)
         ext.axpy(eng.theta, eng.grad, -self.alph)
@@ -1242,8 +1303,7 @@ class DSGTStackedDriver:
                 eng.reduce_fused_grad()
             else:
                 xb, yb = eng.next_batch()
-                eng.forward(xb, train_skip_logp=True)
-                eng.backward(xb, yb)
+                eng.fwd_bwd(xb, yb)
             self.y.copy_(eng.grad)
             self.g.copy_(eng.grad)
 
@@ -1299,10 +1359,8 @@ class DSGTStackedDriver:
         else:
             with _timer("next_batch"):
                 xb, yb = eng.next_batch()
-            with _timer("forward"):
-                eng.forward(xb, train_skip_logp=True)
-            with _timer("backward"):
-                lb = eng.backward(xb, yb, want_loss=want_tl)
+            with _timer("fwd_bwd"):
+                lb = eng.fwd_bwd(xb, yb, want_loss=want_tl)
         if want_tl and lb is not None:
             eng.update_tloss(lb)
         with _timer("y_update"):

@@ -508,3 +508,61 @@ def test_linear_bwd_dx_sin_relu_recompute(ext, dtype, M):
     ext.linear_bwd_dx(dZ, theta, out, Yb, None, 2, scale, nb, M, I, O,
                       Xb2, 0, Ib * I, Ib)
     torch.testing.assert_close(out, ref, **TOL[dtype])
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+@pytest.mark.parametrize("M,I,H,C", [(64, 432, 64, 10),  # MNIST shape
+                                     (40, 50, 48, 7)])   # ragged all
+def test_fc_block_matches_autograd(ext, dtype, M, I, H, C):
+    """One-launch fused fc block (fc1+fc2 fwd, NLL, full fc backward)
+    vs torch autograd on the same two-layer head."""
+    torch.manual_seed(6)
+    L = 2
+    dev = _dev()
+    n = H * I + H + C * H + C
+    w1_off, b1_off = 0, H * I
+    w2_off, b2_off = H * I + H, H * I + H + C * H
+    theta = torch.randn(L, n, dtype=dtype, device=dev) * 0.3
+    x0 = torch.randn(L * M, I, dtype=dtype, device=dev)
+    maxlen = M + 5
+    Y_all = torch.randint(0, C, (L, maxlen), device=dev)
+    # identity-ish index stream with an offset
+    off = 3
+    idx = torch.arange(maxlen, device=dev).repeat(L, 1)
+    idx = (idx + 1) % maxlen  # non-trivial mapping
+    loss_scale = 0.7
+
+    grad = torch.zeros_like(theta)
+    dx0 = torch.empty_like(x0)
+    loss = torch.zeros(L, dtype=dtype, device=dev)
+    ext.fc_block(x0, theta, Y_all, idx, maxlen, off, grad, dx0, loss,
+                 w1_off, b1_off, w2_off, b2_off, M, I, H, C,
+                 loss_scale)
+
+    tol = TOL[dtype] if dtype == torch.float64 else dict(
+        rtol=1e-3, atol=1e-4)
+    for l in range(L):
+        W1 = theta[l, :H * I].reshape(H, I).detach().requires_grad_()
+        b1 = theta[l, b1_off:b1_off + H].detach().requires_grad_()
+        W2 = theta[l, w2_off:w2_off + C * H].reshape(C, H) \
+            .detach().requires_grad_()
+        b2 = theta[l, b2_off:].detach().requires_grad_()
+        xl = x0[l * M:(l + 1) * M].detach().requires_grad_()
+        tgt = Y_all[l][idx[l, off:off + M]]
+        y1 = torch.relu(xl @ W1.T + b1)
+        z2 = y1 @ W2.T + b2
+        ref_loss = torch.nn.functional.nll_loss(
+            torch.log_softmax(z2, dim=1), tgt)
+        (ref_loss * loss_scale).backward()
+        torch.testing.assert_close(loss[l], ref_loss.detach(), **tol)
+        torch.testing.assert_close(dx0[l * M:(l + 1) * M], xl.grad,
+                                   **tol)
+        torch.testing.assert_close(
+            grad[l, :H * I].reshape(H, I), W1.grad, **tol)
+        torch.testing.assert_close(grad[l, b1_off:b1_off + H],
+                                   b1.grad, **tol)
+        torch.testing.assert_close(
+            grad[l, w2_off:w2_off + C * H].reshape(C, H), W2.grad,
+            **tol)
+        torch.testing.assert_close(grad[l, b2_off:], b2.grad, **tol)
