@@ -392,19 +392,37 @@ __global__ __launch_bounds__(256) void fc_block_k(
   // P1: fc1 forward on MFMA — each wave owns one 16-wide o-fragment
   acc_t a1 = {};
   const int o0w = wid * 16;
+  const bool fullm = (m0 + RT) <= M;
   for (int k0 = 0; k0 < I; k0 += 32) {
-    for (int t = tid; t < RT * 32; t += 256) {
-      const int m = t / 32, k = t % 32;  // consecutive k: coalesced
-      As[k * (RT + 1) + m] =
-          (m0 + m < M && k0 + k < I)
-              ? x0[(long)(l * (long)M + m0 + m) * I + (k0 + k)]
-              : T(0);
-    }
-    for (int t = tid; t < 64 * 32; t += 256) {
-      const int o = t / 32, k = t % 32;
-      Bs[k * 65 + o] = (o < H && k0 + k < I)
-                           ? W1[(long)o * I + (k0 + k)]
-                           : T(0);
+    // guard-free loads on full tiles/stages: a runtime condition on
+    // each load of an unrolled chain forces hipcc into per-element
+    // branch + vmcnt(0) waits (cdna_hip_programming.md §5 trap 4c) —
+    // the first fc_block build measured 97 us/launch from exactly that
+    const bool fullk = (k0 + 32) <= I;
+    if (fullm && fullk && H == 64) {
+      for (int t = tid; t < RT * 32; t += 256) {
+        const int m = t / 32, k = t % 32;
+        As[k * (RT + 1) + m] =
+            x0[(long)(l * (long)M + m0 + m) * I + (k0 + k)];
+      }
+      for (int t = tid; t < 64 * 32; t += 256) {
+        const int o = t / 32, k = t % 32;
+        Bs[k * 65 + o] = W1[(long)o * I + (k0 + k)];
+      }
+    } else {
+      for (int t = tid; t < RT * 32; t += 256) {
+        const int m = t / 32, k = t % 32;
+        As[k * (RT + 1) + m] =
+            (m0 + m < M && k0 + k < I)
+                ? x0[(long)(l * (long)M + m0 + m) * I + (k0 + k)]
+                : T(0);
+      }
+      for (int t = tid; t < 64 * 32; t += 256) {
+        const int o = t / 32, k = t % 32;
+        Bs[k * 65 + o] = (o < H && k0 + k < I)
+                             ? W1[(long)o * I + (k0 + k)]
+                             : T(0);
+      }
     }
     __syncthreads();
 #pragma unroll
@@ -498,23 +516,42 @@ __global__ __launch_bounds__(256) void fc_block_k(
   __syncthreads();
 
   // P4: per 64-wide i-chunk — fc1 dW (K = RT, x0 read direct from
-  // global, coalesced) and dX0 (K = H, W1 read direct, coalesced)
+  // global, coalesced) and dX0 (K = H, W1 read direct, coalesced).
+  // Full chunks take the GUARD-FREE loops (trap 4c, see P1); only the
+  // ragged last chunk / last m-tile pay the guarded versions.
+  // dX0 carries the conv layer's relu' mask (the layered path fused it
+  // in linear_bwd_dx; conv_pool_bwd expects dZ pre-masked).
   for (int i0 = 0; i0 < I; i0 += 64) {
+    const bool fulli = (i0 + 64) <= I;
     // dW1[o = wid*16 .., i0..i0+64): A[o][k=m] = dz1, B[k=m][i] = x0
     acc_t aw[4] = {};
     if (o0w < H) {
+      if (fullm && fulli) {
 #pragma unroll
-      for (int kk = 0; kk < RT; kk += 4) {
-        const int ka = kk + lk;
-        const bool rok = (m0 + ka) < M;
-        const T a = dz1s[ka * 65 + o0w + lo];
+        for (int kk = 0; kk < RT; kk += 4) {
+          const int ka = kk + lk;
+          const T a = dz1s[ka * 65 + o0w + lo];
 #pragma unroll
-        for (int fi = 0; fi < 4; ++fi) {
-          const int i = i0 + fi * 16 + lo;
-          const T b = (rok && i < I)
-                          ? x0[(long)(l * (long)M + m0 + ka) * I + i]
-                          : T(0);
-          aw[fi] = MF::mma(a, b, aw[fi]);
+          for (int fi = 0; fi < 4; ++fi) {
+            const T b = x0[(long)(l * (long)M + m0 + ka) * I + i0 +
+                           fi * 16 + lo];
+            aw[fi] = MF::mma(a, b, aw[fi]);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int kk = 0; kk < RT; kk += 4) {
+          const int ka = kk + lk;
+          const bool rok = (m0 + ka) < M;
+          const T a = dz1s[ka * 65 + o0w + lo];
+#pragma unroll
+          for (int fi = 0; fi < 4; ++fi) {
+            const int i = i0 + fi * 16 + lo;
+            const T b = (rok && i < I)
+                            ? x0[(long)(l * (long)M + m0 + ka) * I + i]
+                            : T(0);
+            aw[fi] = MF::mma(a, b, aw[fi]);
+          }
         }
       }
 #pragma unroll
@@ -533,13 +570,24 @@ __global__ __launch_bounds__(256) void fc_block_k(
     // dX0[m, i0 + wid*16 + ..): A[m][k=o] = dz1, B[k=o][i] = W1
     acc_t ax = {};
     const int ix = i0 + wid * 16;
+    if (H == 64 && ix + 16 <= I) {
 #pragma unroll
-    for (int kk = 0; kk < 64; kk += 4) {
-      const int ka = kk + lk;
-      if (ka < H) {
+      for (int kk = 0; kk < 64; kk += 4) {
+        const int ka = kk + lk;
         const T a = dz1s[lo * 65 + ka];
-        const T b = (ix + lo < I) ? W1[(long)ka * I + ix + lo] : T(0);
+        const T b = W1[(long)ka * I + ix + lo];
         ax = MF::mma(a, b, ax);
+      }
+    } else {
+#pragma unroll
+      for (int kk = 0; kk < 64; kk += 4) {
+        const int ka = kk + lk;
+        if (ka < H) {
+          const T a = dz1s[lo * 65 + ka];
+          const T b = (ix + lo < I) ? W1[(long)ka * I + ix + lo]
+                                    : T(0);
+          ax = MF::mma(a, b, ax);
+        }
       }
     }
 #pragma unroll
@@ -547,7 +595,9 @@ __global__ __launch_bounds__(256) void fc_block_k(
       const int m = m0 + MF::acc_row(lane, r);
       const int i = ix + lo;
       if (m < M && i < I) {
-        dx0[(long)(l * (long)M + m) * I + i] = ax[r];
+        const long off = (long)(l * (long)M + m) * I + i;
+        // conv relu' mask: x0 IS the conv block's relu output
+        dx0[off] = x0[off] > T(0) ? ax[r] : T(0);
       }
     }
   }

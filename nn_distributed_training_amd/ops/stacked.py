@@ -614,6 +614,8 @@ class StackedEngine:
             and self.classification
             and len(ls) == 3
             and ls[0].kind == "conv_pool"
+            # dX0 carries the conv block's relu' mask inside fc_block_k
+            and getattr(ls[0], "activation", "relu") == "relu"
             and ls[1].kind == "linear"
             and ls[2].kind == "linear"
             and ls[1].activation == "relu"
