@@ -723,10 +723,12 @@ void regression_bwd(torch::Tensor yhat, torch::Tensor tgt,
 void fc_block(torch::Tensor x0, torch::Tensor theta,
               torch::Tensor Y_all, torch::Tensor idx, long idx_stride,
               long idx_off, torch::Tensor grad, torch::Tensor dx0,
+              torch::Tensor dz1g,
               c10::optional<torch::Tensor> loss, long w1_off,
               long b1_off, long w2_off, long b2_off, long M, long I,
-              long H, long C, double loss_scale) {
+              long H, long C, double loss_scale, bool mask_dx0) {
   CHECK_DEV(x0); CHECK_DEV(theta); CHECK_DEV(grad); CHECK_DEV(dx0);
+  CHECK_DEV(dz1g);
   const long L = theta.size(0), n = theta.size(1);
   const long maxlen = Y_all.size(1);
   TORCH_CHECK(H <= 64 && C <= 16, "fc_block: H <= 64, C <= 16");
@@ -737,12 +739,15 @@ void fc_block(torch::Tensor x0, torch::Tensor theta,
         x0.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
         Y_all.data_ptr<long>(), idx.data_ptr<long>(), idx_stride,
         idx_off, maxlen, grad.data_ptr<scalar_t>(),
-        dx0.data_ptr<scalar_t>(),
+        dx0.data_ptr<scalar_t>(), dz1g.data_ptr<scalar_t>(),
         loss.has_value() ? loss->data_ptr<scalar_t>() : nullptr,
         n, w1_off, b1_off, w2_off, b2_off, (int)M, (int)I, (int)H,
-        (int)C, (scalar_t)loss_scale);
+        (int)C, (scalar_t)loss_scale, mask_dx0 ? 1 : 0);
   });
   HIP_CHECK_LAST();
+  // fc1 dW/db from the exported dz1 (store-path kernel: overwrite,
+  // no atomics)
+  linear_bwd_dw(dz1g, x0, grad, w1_off, b1_off, M, I, H);
 }
 
 // ----------------------------------------------------------- chains --

@@ -355,9 +355,13 @@ __global__ __launch_bounds__(256) void fc_block_k(
     long idx_stride, long idx_off, long maxlen,
     T* __restrict__ grad,            // [L, n]
     T* __restrict__ dx0,             // [L*M, I]
+    T* __restrict__ dz1g,            // [L*M, H] dz1 out (fc1 dW/db
+                                     // run in linear_bwd_dw: the
+                                     // in-kernel atomic dW1 was ~884K
+                                     // f64 atomics/launch)
     T* __restrict__ loss,            // nullable [L]
     long n, long w1_off, long b1_off, long w2_off, long b2_off,
-    int M, int I, int H, int C, T loss_scale) {
+    int M, int I, int H, int C, T loss_scale, int mask_dx0) {
   using MF = gmfma::mfma_t<T>;
   using acc_t = typename MF::acc_t;
   constexpr int RT = FC_RT;
@@ -491,7 +495,11 @@ __global__ __launch_bounds__(256) void fc_block_k(
     for (int c = 0; c < C; ++c) {
       s += dz2s[m * 17 + c] * w2s[c * H + h];
     }
-    dz1s[m * 65 + h] = y1s[m * 65 + h] > T(0) ? s : T(0);
+    const T v = y1s[m * 65 + h] > T(0) ? s : T(0);
+    dz1s[m * 65 + h] = v;
+    if (m0 + m < M) {
+      dz1g[(long)(l * (long)M + m0 + m) * H + h] = v;
+    }
   }
   __syncthreads();
   for (int t = tid; t < C * H; t += 256) {
@@ -507,12 +515,6 @@ __global__ __launch_bounds__(256) void fc_block_k(
     for (int m = 0; m < RT; ++m) s += dz2s[m * 17 + tid];
     atomicAdd(&grad[l * n + b2_off + tid], s);
   }
-  if (tid >= 64 && tid < 64 + H) {
-    const int h = tid - 64;
-    T s = T(0);
-    for (int m = 0; m < RT; ++m) s += dz1s[m * 65 + h];
-    atomicAdd(&grad[l * n + b1_off + h], s);
-  }
   __syncthreads();
 
   // P4: per 64-wide i-chunk — fc1 dW (K = RT, x0 read direct from
@@ -522,51 +524,6 @@ __global__ __launch_bounds__(256) void fc_block_k(
   // dX0 carries the conv layer's relu' mask (the layered path fused it
   // in linear_bwd_dx; conv_pool_bwd expects dZ pre-masked).
   for (int i0 = 0; i0 < I; i0 += 64) {
-    const bool fulli = (i0 + 64) <= I;
-    // dW1[o = wid*16 .., i0..i0+64): A[o][k=m] = dz1, B[k=m][i] = x0
-    acc_t aw[4] = {};
-    if (o0w < H) {
-      if (fullm && fulli) {
-#pragma unroll
-        for (int kk = 0; kk < RT; kk += 4) {
-          const int ka = kk + lk;
-          const T a = dz1s[ka * 65 + o0w + lo];
-#pragma unroll
-          for (int fi = 0; fi < 4; ++fi) {
-            const T b = x0[(long)(l * (long)M + m0 + ka) * I + i0 +
-                           fi * 16 + lo];
-            aw[fi] = MF::mma(a, b, aw[fi]);
-          }
-        }
-      } else {
-#pragma unroll
-        for (int kk = 0; kk < RT; kk += 4) {
-          const int ka = kk + lk;
-          const bool rok = (m0 + ka) < M;
-          const T a = dz1s[ka * 65 + o0w + lo];
-#pragma unroll
-          for (int fi = 0; fi < 4; ++fi) {
-            const int i = i0 + fi * 16 + lo;
-            const T b = (rok && i < I)
-                            ? x0[(long)(l * (long)M + m0 + ka) * I + i]
-                            : T(0);
-            aw[fi] = MF::mma(a, b, aw[fi]);
-          }
-        }
-      }
-#pragma unroll
-      for (int fi = 0; fi < 4; ++fi) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int o = o0w + MF::acc_row(lane, r);
-          const int i = i0 + fi * 16 + lo;
-          if (o < H && i < I) {
-            atomicAdd(&grad[l * n + w1_off + (long)o * I + i],
-                      aw[fi][r]);
-          }
-        }
-      }
-    }
     // dX0[m, i0 + wid*16 + ..): A[m][k=o] = dz1, B[k=o][i] = W1
     acc_t ax = {};
     const int ix = i0 + wid * 16;
@@ -597,7 +554,7 @@ __global__ __launch_bounds__(256) void fc_block_k(
       if (m < M && i < I) {
         const long off = (long)(l * (long)M + m) * I + i;
         // conv relu' mask: x0 IS the conv block's relu output
-        dx0[off] = x0[off] > T(0) ? ax[r] : T(0);
+        dx0[off] = (!mask_dx0 || x0[off] > T(0)) ? ax[r] : T(0);
       }
     }
   }

@@ -652,8 +652,10 @@ class StackedEngine:
             ext.fc_block(
                 bufs["acts"][0], self.theta, self.Y_all, idx_t,
                 stride, self._last_off, self.grad, bufs["dzs"][0],
+                bufs["dzs"][1],
                 loss_buf, fc1.w_off, fc1.b_off, fc2.w_off, fc2.b_off,
                 M, fc1.in_dim, fc1.out_dim, fc2.out_dim, loss_scale,
+                True,
             )
             ext.conv_pool_bwd(
                 bufs["dzs"][0], bufs["idxs"][0], xb, self.grad,

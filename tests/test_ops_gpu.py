@@ -535,10 +535,12 @@ def test_fc_block_matches_autograd(ext, dtype, M, I, H, C):
 
     grad = torch.zeros_like(theta)
     dx0 = torch.empty_like(x0)
+    dz1g = torch.empty(L * M, H, dtype=dtype, device=dev)
     loss = torch.zeros(L, dtype=dtype, device=dev)
-    ext.fc_block(x0, theta, Y_all, idx, maxlen, off, grad, dx0, loss,
-                 w1_off, b1_off, w2_off, b2_off, M, I, H, C,
-                 loss_scale)
+    # mask_dx0=False: x0 here is a plain input, not a relu output
+    ext.fc_block(x0, theta, Y_all, idx, maxlen, off, grad, dx0, dz1g,
+                 loss, w1_off, b1_off, w2_off, b2_off, M, I, H, C,
+                 loss_scale, False)
 
     tol = TOL[dtype] if dtype == torch.float64 else dict(
         rtol=1e-3, atol=1e-4)
