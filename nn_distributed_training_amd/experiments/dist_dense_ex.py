@@ -124,6 +124,14 @@ def experiment(yaml_pth: str):
             solo_confs, exp_conf, output_dir,
         )
 
+    cent_conf = exp_conf.get("centralized_training",
+                             {"train_centralized": False})
+    if cent_conf.get("train_centralized", False) and rank == 0:
+        _train_centralized(
+            base_model, base_loss, train_subsets, val_set, device,
+            cent_conf, exp_conf, output_dir,
+        )
+
     for prob_key, prob_conf in conf_dict["problem_configs"].items():
         opt_conf = prob_conf["optimizer_config"]
         prob = DistDensityProblem(
@@ -170,6 +178,47 @@ def _train_solo_all(N, base_model, base_loss, train_subsets, val_set,
     if exp_conf["writeout"]:
         torch.save(
             solo_results, os.path.join(output_dir, "solo_results.pt")
+        )
+
+
+def _train_centralized(base_model, base_loss, train_subsets, val_set,
+                       device, cent_conf, exp_conf, output_dir):
+    """Pooled-data upper-bound baseline (reference
+    centralized/online_density.ipynb cell 4: the same FourierNet
+    trained centrally on the union of all nodes' scans), with a
+    per-epoch validation-loss curve."""
+    print("Performing centralized (pooled-data) training ...")
+    model = copy.deepcopy(base_model).to(device)
+    pooled = torch.utils.data.ConcatDataset(list(train_subsets))
+    loader = torch.utils.data.DataLoader(
+        pooled, cent_conf["train_batch_size"], shuffle=True
+    )
+    opt = torch.optim.Adam(model.parameters(), lr=cent_conf["lr"])
+    curves = {"validation_loss": [], "epoch": []}
+    for ep in range(cent_conf["epochs"]):
+        for locs, dens in loader:
+            opt.zero_grad()
+            yh = model.forward(locs.to(device))
+            base_loss(torch.squeeze(yh), dens.to(device)).backward()
+            opt.step()
+        with torch.no_grad():
+            vloss = 0.0
+            vloader = torch.utils.data.DataLoader(
+                val_set, cent_conf["val_batch_size"]
+            )
+            for locs, dens in vloader:
+                yh = model.forward(locs.to(device))
+                vloss += base_loss(
+                    torch.squeeze(yh), dens.to(device)
+                ).item()
+        curves["epoch"].append(ep)
+        curves["validation_loss"].append(vloss)
+        if cent_conf.get("verbose", False):
+            print(f"Centralized epoch {ep} - Val Loss = {vloss:.4f}")
+    if exp_conf["writeout"]:
+        torch.save(
+            curves,
+            os.path.join(output_dir, "centralized_results.pt"),
         )
 
 

@@ -68,6 +68,63 @@ def train_solo(model, loss, train_set, val_set, device, conf):
     }
 
 
+def train_centralized(model, loss, train_subsets, val_set, device,
+                      conf):
+    """Pooled-data upper-bound baseline: ONE model trained on the
+    union of every node's shard, validation tracked per epoch.
+
+    This reproduces the reference's centralized notebooks (the 0.985
+    top-1 'centralized' line every figure in
+    visualization/mnist_four.ipynb cells 1-5 is anchored to) — the
+    reference has no script for it, only notebook cells; here it is a
+    config-driven part of the experiment (`centralized_training:`).
+    """
+    pooled = torch.utils.data.ConcatDataset(list(train_subsets))
+    trainloader = torch.utils.data.DataLoader(
+        pooled, conf["train_batch_size"], shuffle=True
+    )
+    valloader = torch.utils.data.DataLoader(
+        val_set, conf["val_batch_size"]
+    )
+    model = model.to(device)
+    opts = {
+        "adam": torch.optim.Adam,
+        "sgd": torch.optim.SGD,
+        "adamw": torch.optim.AdamW,
+    }
+    if conf["optimizer"] not in opts:
+        raise NameError("Unknown centralized optimizer.")
+    opt = opts[conf["optimizer"]](model.parameters(), lr=conf["lr"])
+
+    curves = {"validation_loss": [], "validation_accuracy": [],
+              "epoch": []}
+    for ep in range(conf["epochs"]):
+        for x, y in trainloader:
+            opt.zero_grad()
+            out = model.forward(x.to(device))
+            loss(out, y.to(device)).backward()
+            opt.step()
+        with torch.no_grad():
+            val_loss, correct = 0.0, 0
+            for x, y in valloader:
+                x, y = x.to(device), y.to(device)
+                out = model.forward(x)
+                val_loss += loss(out, y).item()
+                pred = out.argmax(dim=1, keepdim=True)
+                correct += pred.eq(y.view_as(pred)).sum().item()
+            nval = len(valloader.dataset)
+        curves["epoch"].append(ep)
+        curves["validation_loss"].append(val_loss / nval)
+        curves["validation_accuracy"].append(correct / nval)
+        if conf.get("verbose", False):
+            print(
+                "Centralized epoch {} - Validation Acc = {:.4f}".format(
+                    ep, curves["validation_accuracy"][-1]
+                )
+            )
+    return curves
+
+
 def experiment(yaml_pth: str):
     with open(yaml_pth) as f:
         conf_dict = yaml.safe_load(f)
@@ -128,6 +185,22 @@ def experiment(yaml_pth: str):
         if exp_conf["writeout"]:
             torch.save(
                 solo_results, os.path.join(output_dir, "solo_results.pt")
+            )
+
+    # pooled-data centralized baseline (upper-bound curve; reference
+    # produced it in centralized/*.ipynb notebooks only)
+    cent_conf = exp_conf.get("centralized_training",
+                             {"train_centralized": False})
+    if cent_conf.get("train_centralized", False) and rank == 0:
+        print("Performing centralized (pooled-data) training ...")
+        cent_results = train_centralized(
+            copy.deepcopy(base_model), base_loss, train_subsets,
+            val_set, device, cent_conf,
+        )
+        if exp_conf["writeout"]:
+            torch.save(
+                cent_results,
+                os.path.join(output_dir, "centralized_results.pt"),
             )
 
     # per-(problem, optimizer) loop

@@ -110,6 +110,18 @@ def experiment(yaml_pth: str):
             for wp in waypoint_sets
         ]
 
+    cent_conf = exp_conf.get("centralized_training",
+                             {"train_centralized": False})
+    if cent_conf.get("train_centralized", False) and rank == 0:
+        from .dist_dense_ex import _train_centralized
+
+        # pooled = every node's FULL trajectory (the reference's
+        # centralized/online_density.ipynb cell 4 trains on all data)
+        _train_centralized(
+            base_model, base_loss, train_subsets, val_set, device,
+            cent_conf, exp_conf, output_dir,
+        )
+
     for prob_key, prob_conf in conf_dict["problem_configs"].items():
         opt_conf = prob_conf["optimizer_config"]
         # fresh sliding windows per problem run

@@ -393,41 +393,32 @@ __global__ __launch_bounds__(256) void fc_block_k(
   for (int t = tid; t < C * H; t += 256) w2s[t] = th[w2_off + t];
   for (int t = tid; t < C; t += 256) w2s[C * H + t] = th[b2_off + t];
 
-  // P1: fc1 forward on MFMA — each wave owns one 16-wide o-fragment
+  // P1: fc1 forward on MFMA — each wave owns one 16-wide o-fragment.
+  // T14 write-after-barrier register pipeline + vec2 loads: stage s+1
+  // streams from HBM while stage s runs its MFMAs (W1's 221 KB/block
+  // stream was the serial chain that left the first builds at
+  // 97 -> 67 us); guard-free loads on full tiles/stages (trap 4c).
+  typedef T vec2 __attribute__((ext_vector_type(2)));
   acc_t a1 = {};
   const int o0w = wid * 16;
   const bool fullm = (m0 + RT) <= M;
-  for (int k0 = 0; k0 < I; k0 += 32) {
-    // guard-free loads on full tiles/stages: a runtime condition on
-    // each load of an unrolled chain forces hipcc into per-element
-    // branch + vmcnt(0) waits (cdna_hip_programming.md §5 trap 4c) —
-    // the first fc_block build measured 97 us/launch from exactly that
-    const bool fullk = (k0 + 32) <= I;
-    if (fullm && fullk && H == 64) {
-      for (int t = tid; t < RT * 32; t += 256) {
-        const int m = t / 32, k = t % 32;
-        As[k * (RT + 1) + m] =
-            x0[(long)(l * (long)M + m0 + m) * I + (k0 + k)];
-      }
-      for (int t = tid; t < 64 * 32; t += 256) {
-        const int o = t / 32, k = t % 32;
-        Bs[k * 65 + o] = W1[(long)o * I + (k0 + k)];
-      }
-    } else {
-      for (int t = tid; t < RT * 32; t += 256) {
-        const int m = t / 32, k = t % 32;
-        As[k * (RT + 1) + m] =
-            (m0 + m < M && k0 + k < I)
-                ? x0[(long)(l * (long)M + m0 + m) * I + (k0 + k)]
-                : T(0);
-      }
-      for (int t = tid; t < 64 * 32; t += 256) {
-        const int o = t / 32, k = t % 32;
-        Bs[k * 65 + o] = (o < H && k0 + k < I)
-                             ? W1[(long)o * I + (k0 + k)]
-                             : T(0);
-      }
+  const int am = tid / 16, akp = tid % 16;   // A pair: row am, kpair
+  vec2 rA, rB[4];
+  const int nst = (I + 31) / 32;
+
+#define FC_LOAD_STAGE(k0v)                                               {                                                                        const int k0_ = (k0v);                                                 const bool fullk_ = (k0_ + 32) <= I;                                   if (fullm && fullk_ && H == 64) { rA = *reinterpret_cast<const vec2*>(&x0[(long)(l * (long)M + m0 + am) * I + k0_ + 2 * akp]); _Pragma("unroll") for (int q = 0; q < 4; ++q) { rB[q] = *reinterpret_cast<const vec2*>(&W1[(long)(am + q * 16) * I + k0_ + 2 * akp]); } } else {                                                                 rA = vec2{0, 0};                                                       const int k_ = k0_ + 2 * akp;                                          if (m0 + am < M) {                                                       if (k_ + 1 < I) {                                                        rA = *reinterpret_cast<const vec2*>(                                       &x0[(long)(l * (long)M + m0 + am) * I + k_]);                    } else if (k_ < I) {                                                     rA.x = x0[(long)(l * (long)M + m0 + am) * I + k_];                   }                                                                    }                                                                      _Pragma("unroll") for (int q = 0; q < 4; ++q) {                          const int o = am + q * 16;                                             rB[q] = vec2{0, 0};                                                    if (o < H) {                                                             if (k_ + 1 < I) {                                                        rB[q] = *reinterpret_cast<const vec2*>(                                    &W1[(long)o * I + k_]);                                          } else if (k_ < I) {                                                     rB[q].x = W1[(long)o * I + k_];                                      }                                                                    }                                                                    }                                                                    }                                                                    }
+
+  FC_LOAD_STAGE(0)
+  for (int st = 0; st < nst; ++st) {
+    __syncthreads();
+    As[(2 * akp) * (RT + 1) + am] = rA.x;
+    As[(2 * akp + 1) * (RT + 1) + am] = rA.y;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      Bs[(2 * akp) * 65 + am + q * 16] = rB[q].x;
+      Bs[(2 * akp + 1) * 65 + am + q * 16] = rB[q].y;
     }
+    if (st + 1 < nst) FC_LOAD_STAGE((st + 1) * 32)
     __syncthreads();
 #pragma unroll
     for (int kk = 0; kk < 32; kk += 4) {
@@ -436,8 +427,8 @@ __global__ __launch_bounds__(256) void fc_block_k(
       const T b = Bs[ka * 65 + o0w + lo];
       a1 = MF::mma(a, b, a1);
     }
-    __syncthreads();
   }
+#undef FC_LOAD_STAGE
   // y1 = relu(z1 + b1) into LDS (never stored to HBM)
   if (o0w < H) {
 #pragma unroll
@@ -501,6 +492,12 @@ __global__ __launch_bounds__(256) void fc_block_k(
       dz1g[(long)(l * (long)M + m0 + m) * H + h] = v;
     }
   }
+  if (H < 64) {  // zero the pad columns the dX0 MFMA sweeps over
+    for (int t = tid; t < RT * (64 - H); t += 256) {
+      const int m = t / (64 - H), h = H + t % (64 - H);
+      dz1s[m * 65 + h] = T(0);
+    }
+  }
   __syncthreads();
   for (int t = tid; t < C * H; t += 256) {
     const int c = t / H, h = t % H;
@@ -523,39 +520,49 @@ __global__ __launch_bounds__(256) void fc_block_k(
   // ragged last chunk / last m-tile pay the guarded versions.
   // dX0 carries the conv layer's relu' mask (the layered path fused it
   // in linear_bwd_dx; conv_pool_bwd expects dZ pre-masked).
-  for (int i0 = 0; i0 < I; i0 += 64) {
-    // dX0[m, i0 + wid*16 + ..): A[m][k=o] = dz1, B[k=o][i] = W1
-    acc_t ax = {};
-    const int ix = i0 + wid * 16;
-    if (H == 64 && ix + 16 <= I) {
+  // dX0[m, i]: A[m][k=o] = dz1 (LDS), B[k=o][i] = W1 (global,
+  // coalesced). W1's 16-value column block for chunk c+1 prefetches
+  // into registers while chunk c's MFMAs issue (dz1s pad columns are
+  // zeroed above, so the full 64-deep sweep is guard-free).
+  {
+    T w1v[16], w1n[16];
+    const auto ldw1 = [&](int ixv, T* dst) {
+      if (H == 64 && ixv + 16 <= I) {  // guard-free (trap 4c)
 #pragma unroll
-      for (int kk = 0; kk < 64; kk += 4) {
-        const int ka = kk + lk;
-        const T a = dz1s[lo * 65 + ka];
-        const T b = W1[(long)ka * I + ix + lo];
-        ax = MF::mma(a, b, ax);
-      }
-    } else {
+        for (int t = 0; t < 16; ++t) {
+          dst[t] = W1[(long)(4 * t + lk) * I + ixv + lo];
+        }
+      } else {
 #pragma unroll
-      for (int kk = 0; kk < 64; kk += 4) {
-        const int ka = kk + lk;
-        if (ka < H) {
-          const T a = dz1s[lo * 65 + ka];
-          const T b = (ix + lo < I) ? W1[(long)ka * I + ix + lo]
-                                    : T(0);
-          ax = MF::mma(a, b, ax);
+        for (int t = 0; t < 16; ++t) {
+          dst[t] = ((4 * t + lk) < H && ixv + lo < I)
+                       ? W1[(long)(4 * t + lk) * I + ixv + lo]
+                       : T(0);
         }
       }
-    }
+    };
+    ldw1(wid * 16, w1v);
+    for (int i0 = 0; i0 < I; i0 += 64) {
+      const int ix = i0 + wid * 16;
+      if (i0 + 64 < I) ldw1(i0 + 64 + wid * 16, w1n);
+      acc_t ax = {};
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int m = m0 + MF::acc_row(lane, r);
-      const int i = ix + lo;
-      if (m < M && i < I) {
-        const long off = (long)(l * (long)M + m) * I + i;
-        // conv relu' mask: x0 IS the conv block's relu output
-        dx0[off] = (!mask_dx0 || x0[off] > T(0)) ? ax[r] : T(0);
+      for (int t = 0; t < 16; ++t) {
+        const T a = dz1s[lo * 65 + 4 * t + lk];
+        ax = MF::mma(a, w1v[t], ax);
       }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + MF::acc_row(lane, r);
+        const int i = ix + lo;
+        if (m < M && i < I) {
+          const long off = (long)(l * (long)M + m) * I + i;
+          // conv relu' mask: x0 IS the conv block's relu output
+          dx0[off] = (!mask_dx0 || x0[off] > T(0)) ? ax[r] : T(0);
+        }
+      }
+#pragma unroll
+      for (int t = 0; t < 16; ++t) w1v[t] = w1n[t];
     }
   }
   (void)rowval;

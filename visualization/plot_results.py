@@ -43,7 +43,13 @@ def _band(ax, xs, series, label):
 
 def plot_run_dir(run_dir: str, out_dir: str, eval_every: int = 20):
     os.makedirs(out_dir, exist_ok=True)
-    results = sorted(glob.glob(os.path.join(run_dir, "*_results.pt")))
+    results = sorted(
+        rp
+        for rp in glob.glob(os.path.join(run_dir, "*_results.pt"))
+        # baselines have scalar-per-epoch curves, not per-node rows
+        if os.path.basename(rp) not in ("solo_results.pt",
+                                        "centralized_results.pt")
+    )
     if not results:
         print(f"no *_results.pt under {run_dir}")
         return
@@ -95,6 +101,53 @@ def plot_run_dir(run_dir: str, out_dir: str, eval_every: int = 20):
         fig.tight_layout()
         fig.savefig(os.path.join(out_dir, "consensus.png"), dpi=130)
     plt.close(fig)
+
+    # mnist_four-style comparability figure: centralized (pooled-data
+    # upper bound) + per-node solo baselines + all decentralized
+    # curves on one accuracy axis (reference
+    # visualization/mnist_four.ipynb cells 1-5)
+    cent_path = os.path.join(run_dir, "centralized_results.pt")
+    solo_path = os.path.join(run_dir, "solo_results.pt")
+    if os.path.exists(cent_path) or os.path.exists(solo_path):
+        for metric, fname, ylabel in (
+            ("top1_accuracy", "four_accuracy.png", "top-1 accuracy"),
+            ("validation_loss", "four_val_loss.png",
+             "validation loss"),
+        ):
+            fig, ax = plt.subplots(figsize=(6.5, 4))
+            drew = False
+            for rp in results:
+                name = os.path.basename(rp).replace("_results.pt", "")
+                res = torch.load(rp, weights_only=False)
+                if metric not in res or not res[metric]:
+                    continue
+                xs = np.arange(len(res[metric])) * eval_every
+                _band(ax, xs, res[metric], name)
+                drew = True
+            key = ("validation_accuracy"
+                   if metric == "top1_accuracy" else metric)
+            if os.path.exists(cent_path):
+                cent = torch.load(cent_path, weights_only=False)
+                if key in cent and cent[key]:
+                    ax.axhline(cent[key][-1], color="k", ls="--",
+                               label="centralized (pooled)")
+                    drew = True
+            if os.path.exists(solo_path):
+                solo = torch.load(solo_path, weights_only=False)
+                vals = [
+                    v[key] for v in solo.values() if key in v
+                ]
+                if vals:
+                    ax.axhline(np.mean(vals), color="gray", ls=":",
+                               label="solo mean")
+                    drew = True
+            if drew:
+                ax.set_xlabel("communication rounds")
+                ax.set_ylabel(ylabel)
+                ax.legend(fontsize=8)
+                fig.tight_layout()
+                fig.savefig(os.path.join(out_dir, fname), dpi=130)
+            plt.close(fig)
 
     summary = os.path.join(run_dir, "scaling_summary.pt")
     if os.path.exists(summary):
