@@ -28,7 +28,9 @@ def test_env_api():
     env = SimpleTagEnv(num_predators=3, num_obstacles=2, seed=0)
     obs = env.reset()
     assert obs.shape == (3, env.obs_dim)
-    assert env.obs_dim == 2 + 2 + 4 + 4 + 2 + 2
+    # modified-scenario adversary obs: NO obstacle entries (reference
+    # simple_tag.py:135-151 drops entity_pos from the concatenate)
+    assert env.obs_dim == 2 + 2 + 2 * 2 + 2 + 2
     a = np.zeros((3, 5))
     obs2, rews, done, info = env.step(a)
     assert obs2.shape == obs.shape
@@ -163,3 +165,135 @@ def test_eval_policy_roundtrip(tmp_path):
     actors = load_actors(str(path), env, hidden=(16, 16))
     rews, _ = eval_episodes(actors, env, episodes=2, max_steps=20)
     assert len(rews) == 2 and all(np.isfinite(rews))
+
+
+# ---------------------------------------------------------------------
+# MPE simple_tag parity properties: each formula
+# checked against the reference scenario's math
+# (RL/pettingzoo/mpe/scenarios/simple_tag.py and _mpe_utils/core.py).
+
+
+def test_mpe_obstacles_fixed_layout():
+    """Reference modification pins the obstacle layout
+    (simple_tag.py:51-53); the env must use the same positions."""
+    from nn_distributed_training_amd.rl.envs import OBSTACLE_POS
+
+    env = SimpleTagEnv(num_predators=3, num_obstacles=8, seed=0)
+    np.testing.assert_allclose(env.obst_pos, OBSTACLE_POS)
+    np.testing.assert_allclose(env.obst_pos[0], [-1.2, -0.6])
+
+
+def test_mpe_adversary_reward_shared_and_shaped():
+    """simple_tag.py:117-132: every adversary's reward is
+    sum_adv(-0.1 * dist(prey, adv)) + 10 per colliding (prey, adv)
+    pair — identical across adversaries."""
+    env = SimpleTagEnv(num_predators=3, num_obstacles=0, seed=0)
+    env.reset()
+    env.pred_pos = np.array([[0.0, 0.0], [0.5, 0.0], [0.0, 0.5]])
+    env.prey_pos = np.array([0.1, 0.0])
+    rews, d = env._rewards()
+    expect = -0.1 * (0.1 + 0.4 + np.hypot(0.1, 0.5))
+    # pred 0 at distance 0.1 < 0.125 collides: +10 once, to EVERY adv
+    expect += 10.0
+    np.testing.assert_allclose(rews, expect, rtol=1e-12)
+    assert np.all(rews == rews[0])
+
+
+def test_mpe_collision_force_soft_margin():
+    """core.py get_collision_force: penetration =
+    logaddexp(0, -(dist-dist_min)/k)*k, f = contact_force *
+    delta/dist * penetration."""
+    from nn_distributed_training_amd.rl.envs import (
+        CONTACT_FORCE,
+        CONTACT_MARGIN,
+        mpe_collision_force,
+    )
+
+    delta = np.array([0.05, 0.0])
+    dist = 0.05
+    dist_min = 0.125  # pred + prey
+    f = mpe_collision_force(delta, dist, dist_min)
+    k = CONTACT_MARGIN
+    pen = np.logaddexp(0, -(dist - dist_min) / k) * k
+    np.testing.assert_allclose(
+        f, CONTACT_FORCE * delta / dist * pen, rtol=1e-12
+    )
+    # far apart: force ~ 0
+    f2 = mpe_collision_force(np.array([1.0, 0.0]), 1.0, 0.125)
+    assert np.linalg.norm(f2) < 1e-6
+
+
+def test_mpe_predators_collide_with_each_other():
+    """MPE applies contact forces between ALL colliding entity pairs,
+    including predator-predator (core.py apply_environment_force)."""
+    env = SimpleTagEnv(num_predators=2, num_obstacles=0, seed=0)
+    env.reset()
+    env.pred_pos = np.array([[0.0, 0.0], [0.1, 0.0]])  # overlapping
+    env.pred_vel[:] = 0.0
+    env.prey_pos = np.array([5.0, 5.0])  # far away
+    env.prey_vel[:] = 0.0
+    env.step(np.zeros((2, 5)))
+    assert env.pred_vel[0, 0] < 0 and env.pred_vel[1, 0] > 0
+
+
+def test_mpe_integrator_form():
+    """core.py integrate_state: vel = vel*(1-damping) + f*dt, speed
+    clamp at max_speed, pos += vel*dt."""
+    from nn_distributed_training_amd.rl.envs import DAMPING, DT
+
+    env = SimpleTagEnv(num_predators=1, num_obstacles=0, seed=0)
+    env.reset()
+    env.pred_pos = np.array([[0.0, 0.0]])
+    env.pred_vel = np.array([[0.2, 0.0]])
+    env.prey_pos = np.array([10.0, 10.0])
+    env.prey_vel[:] = 0.0
+    a = np.zeros((1, 5))
+    a[0, 1] = 0.5  # +x force channel
+    env.step(a)
+    v_expect = 0.2 * (1 - DAMPING) + 0.5 * env.pred_accel * DT
+    np.testing.assert_allclose(env.pred_vel[0, 0], v_expect,
+                               rtol=1e-9)
+    np.testing.assert_allclose(env.pred_pos[0, 0], v_expect * DT,
+                               rtol=1e-9)
+    # speed clamp
+    env.pred_vel = np.array([[5.0, 0.0]])
+    env.step(np.zeros((1, 5)))
+    assert np.linalg.norm(env.pred_vel[0]) <= env.pred_max_speed + 1e-9
+
+
+def test_mpe_prey_heuristic_matches_reference_form():
+    """RL/dist_rl/dist_ppo.py:79-126: flee nearest adversary with the
+    force normalized by max |component| (inf-norm), outward channel
+    zeroed at the +-1.2 boundary."""
+    env = SimpleTagEnv(num_predators=2, num_obstacles=0, seed=0)
+    env.reset()
+    env.prey_pos = np.array([0.0, 0.0])
+    env.pred_pos = np.array([[0.4, 0.2], [2.0, 2.0]])
+    a = env._prey_heuristic_action()
+    # nearest = pred0 at rel (+0.4, +0.2); force = -rel / 0.4
+    np.testing.assert_allclose(a[2], 1.0, rtol=1e-12)   # -x channel
+    np.testing.assert_allclose(a[4], 0.5, rtol=1e-12)   # -y channel
+    assert a[1] == 0.0 and a[3] == 0.0
+    # boundary cutoff: at x <= -1.2 the -x channel is cleared
+    env.prey_pos = np.array([-1.25, 0.0])
+    env.pred_pos = np.array([[0.0, 0.0], [2.0, 2.0]])
+    a = env._prey_heuristic_action()
+    assert a[2] == 0.0
+
+
+def test_mpe_prey_reward_bound_penalty():
+    """simple_tag.py:91-114 good-agent reward: -10 per touching
+    adversary and the piecewise boundary penalty."""
+    env = SimpleTagEnv(num_predators=1, num_obstacles=0, seed=0)
+    env.reset()
+    env.pred_pos = np.array([[5.0, 5.0]])
+    env.prey_pos = np.array([0.95, 0.0])
+    np.testing.assert_allclose(env.prey_reward(), -(0.95 - 0.9) * 10,
+                               rtol=1e-9)
+    env.prey_pos = np.array([1.5, 0.0])
+    np.testing.assert_allclose(env.prey_reward(),
+                               -min(np.exp(2 * 1.5 - 2), 10),
+                               rtol=1e-9)
+    env.prey_pos = np.array([0.0, 0.0])
+    env.pred_pos = np.array([[0.05, 0.0]])  # touching
+    np.testing.assert_allclose(env.prey_reward(), -10.0, rtol=1e-9)
