@@ -168,33 +168,37 @@ __global__ void reduce_parts_k(const T* __restrict__ parts,
 // dinno.py:57-72).
 template <typename T, int MODE, bool WITH_PENALTY>
 __global__ void fused_step_k(
-    T* __restrict__ theta, const T* __restrict__ grad,
+    T* __restrict__ theta, T* __restrict__ grad,
     const T* __restrict__ dual,   // null unless WITH_PENALTY
     const T* __restrict__ s,      // null unless WITH_PENALTY
     const int* __restrict__ deg,  // [L], null unless WITH_PENALTY
     T* __restrict__ m, T* __restrict__ v,  // Adam state (null for SGD)
     T rho, T lr, T beta1, T beta2, T eps, T wd, T bc1, T bc2,
-    int first_step, int nparts, long n, long L) {
+    int first_step, int nparts, long n, long L, int zero_grad) {
   const long total = L * n;
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
     const long l = t / n;
     const long e = t - l * n;
+    T g;
+    if (nparts > 1) {  // per-tile slabs from the fused train step
+      g = T(0);
+      for (int p = 0; p < nparts; ++p) {
+        const long gi = (l * nparts + p) * n + e;
+        g += grad[gi];
+        if (zero_grad) grad[gi] = T(0);  // next iter's atomics land
+                                         // on a clean slate for free
+      }
+    } else {
+      g = grad[t];
+      if (zero_grad) grad[t] = T(0);
+    }
     // isolated node under a dynamic graph: the golden engine skips the
     // whole primal update (the reference crashes on torch.stack of an
     // empty neighbor list, so "frozen" is this framework's defined
     // behavior) — freeze theta AND the Adam moments to match exactly
     if (WITH_PENALTY && deg[l] == 0) continue;
     T th = theta[t];
-    T g;
-    if (nparts > 1) {  // per-tile slabs from the fused train step
-      g = T(0);
-      for (int p = 0; p < nparts; ++p) {
-        g += grad[(l * nparts + p) * n + e];
-      }
-    } else {
-      g = grad[t];
-    }
     if (WITH_PENALTY) {
       g += dual[t] + T(2) * rho * (T(deg[l]) * th - s[t]);
     }
@@ -365,13 +369,16 @@ __global__ void gather_targets_dev_k(
   }
 }
 
-// Plain axpy: theta -= alpha * grad  (DSGD local step)
+// Plain axpy: theta -= alpha * grad  (DSGD local step); zero_grad
+// clears grad after consuming it (next iteration's atomic kernels
+// then need no separate fill launch)
 template <typename T>
-__global__ void axpy_k(T* __restrict__ x, const T* __restrict__ g,
-                       T alpha, long total) {
+__global__ void axpy_k(T* __restrict__ x, T* __restrict__ g,
+                       T alpha, long total, int zero_grad) {
   for (long t = blockIdx.x * (long)BLOCK + threadIdx.x; t < total;
        t += (long)gridDim.x * BLOCK) {
     x[t] += alpha * g[t];
+    if (zero_grad) g[t] = T(0);
   }
 }
 

@@ -150,7 +150,7 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
                 c10::optional<torch::Tensor> v,
                 double rho, double lr, double beta1, double beta2,
                 double eps, double wd, long step_t, long mode,
-                bool first_step, long nparts) {
+                bool first_step, long nparts, bool zero_grad) {
   CHECK_DEV(theta); CHECK_DEV(grad);
   const long L = theta.size(0), n = theta.size(1);
   TORCH_CHECK(grad.numel() == L * nparts * n, "grad/nparts mismatch");
@@ -173,7 +173,7 @@ void fused_step(torch::Tensor theta, torch::Tensor grad,
           v.has_value() ? v->data_ptr<scalar_t>() : nullptr,
           (scalar_t)rho, (scalar_t)lr, (scalar_t)beta1,
           (scalar_t)beta2, (scalar_t)eps, (scalar_t)wd, bc1, bc2,
-          first_step ? 1 : 0, (int)nparts, n, L);
+          first_step ? 1 : 0, (int)nparts, n, L, zero_grad ? 1 : 0);
     };
     using c0 = std::integral_constant<int, 0>;
     using c1 = std::integral_constant<int, 1>;
@@ -304,13 +304,14 @@ void reduce_parts(torch::Tensor parts, torch::Tensor out,
   HIP_CHECK_LAST();
 }
 
-void axpy(torch::Tensor x, torch::Tensor g, double alpha) {
+void axpy(torch::Tensor x, torch::Tensor g, double alpha,
+          bool zero_grad) {
   CHECK_DEV(x);
   DISPATCH_FT(x, {
     hipLaunchKernelGGL(ew::axpy_k<scalar_t>,
         dim3(grid_1d(x.numel())), dim3(ew::BLOCK), 0, cur_stream(),
         x.data_ptr<scalar_t>(), g.data_ptr<scalar_t>(),
-        (scalar_t)alpha, x.numel());
+        (scalar_t)alpha, x.numel(), zero_grad ? 1 : 0);
   });
   HIP_CHECK_LAST();
 }

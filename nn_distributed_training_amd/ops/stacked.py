@@ -333,6 +333,9 @@ class StackedEngine:
         self._zero_plan = None
         self._bufs = None
         self._loss_kind = type(problem.base_loss).__name__  # NLLLoss etc.
+        # True while self.grad is all zeros (init, or after a step
+        # kernel that cleared it); lets fwd_bwd skip the fill launch
+        self.grad_is_zero = True
         # spec tables for the C++ fwd/bwd chains (ONE pybind call per
         # pass instead of one per layer-op: ~10 us of host overhead per
         # ext call made the host the MNIST round bottleneck — see
@@ -638,9 +641,12 @@ class StackedEngine:
                 conv.w_off, conv.b_off, M, conv.out_dim,
                 conv.kernel_size, conv.in_dim,
             )
-            # fc grads accumulate atomically: zero the whole stack
-            # (~1.8 MB — same cost as the conv-slice fill it replaces)
-            self.grad.zero_()
+            # fc grads accumulate atomically: the stack must be all
+            # zeros here. The step kernels clear it as they consume it
+            # (zero_grad), so the fill launch is usually skipped.
+            if not self.grad_is_zero:
+                self.grad.zero_()
+            self.grad_is_zero = False
             loss_buf = None
             if want_loss:
                 bufs["loss"].zero_()
@@ -1191,13 +1197,17 @@ class DiNNOStackedDriver:
                 eng.update_tloss(lb)
             with _timer("fused_step"):
                 self.step_t += 1
+                zg = nparts == 1 and grad_t is eng.grad
                 ext.fused_step(
                     eng.theta, grad_t, self.duals, self.s, deg,
                     None if self.mode == 2 else self.m,
                     None if self.mode == 2 else self.v,
                     self.rho, lr, 0.9, 0.999, 1e-8, self.wd,
                     self.step_t, self.mode, self.step_t == 1, nparts,
+                    zg,
                 )
+                if zg:
+                    eng.grad_is_zero = True
 
     def _opt_state(self):
         return {
@@ -1270,7 +1280,8 @@ class DSGDStackedDriver:
             lb = eng.fwd_bwd(xb, yb, want_loss=want_tl)
         if want_tl and lb is not None:
             eng.update_tloss(lb)
-        ext.axpy(eng.theta, eng.grad, -self.alph)
+        ext.axpy(eng.theta, eng.grad, -self.alph, True)
+        eng.grad_is_zero = True
 
     def _opt_state(self):
         return {"alph": self.alph}

@@ -442,6 +442,7 @@ def test_fused_step_matches_torch_opt(ext, dtype, optname, mode):
             theta, pred_grad, duals, s, deg,
             None if mode == 2 else m, None if mode == 2 else v,
             rho, lr, 0.9, 0.999, 1e-8, wd, step, mode, step == 1, 1,
+            False,
         )
         for l in range(L):
             p = ref_params[l]
