@@ -217,3 +217,79 @@ def test_online_density_two_rank_matches_single(tmp_path):
             pieces[i] = torch.from_numpy(stack[li])
     dist_stack = torch.stack([pieces[i] for i in range(4)])
     torch.testing.assert_close(dist_stack, golden, rtol=0, atol=0)
+
+
+# ---------------------------------------------------------------------
+# Rank-packing stress: the scaling-study shape —
+# 32 logical nodes packed over 8 ranks — with DSGT's double-width
+# (p, y) bundles, on a random graph (uneven cross-rank edge pattern).
+
+
+def _run_training_32(alg, world_check=None):
+    torch.set_default_dtype(torch.float64)
+    torch.manual_seed(42)
+    N = 32
+    graph = nx.erdos_renyi_graph(N, 0.15, seed=3)
+    while not nx.is_connected(graph):
+        graph = nx.erdos_renyi_graph(N, 0.15, seed=4)
+    train = SyntheticMNIST(320, seed=0)
+    val = SyntheticMNIST(64, seed=1)
+    subsets = split_train_set(train, N, "hetero_sorted")
+    base_model = MNISTConvNet(2, 5, 16)
+    conf = _prob_conf(alg)
+    conf["train_batch_size"] = 4
+    conf["optimizer_config"]["outer_iterations"] = 2
+    if alg == "dinno":
+        conf["optimizer_config"]["primal_iterations"] = 1
+    pr = DistMNISTProblem(
+        graph, base_model, torch.nn.NLLLoss(), subsets, val,
+        torch.device("cpu"), conf,
+    )
+    if world_check is not None:
+        assert pr.comm.world == world_check
+    dopt = build_optimizer(pr, torch.device("cpu"),
+                           conf["optimizer_config"])
+    dopt.train()
+    return pr.local_params_stack(), pr.local_nodes
+
+
+def _worker32(rank, world, alg, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        stack, nodes = _run_training_32(alg, world_check=world)
+        with open(os.path.join(out_dir, f"rank{rank}.pkl"), "wb") as f:
+            pickle.dump((list(nodes), stack.numpy()), f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("alg", ["dsgt", "dinno"])
+def test_32_nodes_8_ranks_matches_single_process(alg, tmp_path):
+    """32 logical nodes packed 4-per-rank over 8 gloo ranks (the
+    dist_mnist_scaling shape the driver's 8-GPU SCALE run uses) must be
+    BITWISE identical to the single-process run. DSGT doubles the
+    per-edge payload ((p, y) bundles)."""
+    golden, nodes = _run_training_32(alg)
+    assert list(nodes) == list(range(32))
+
+    port = 29641 if alg == "dsgt" else 29643
+    mp.start_processes(
+        _worker32, args=(8, alg, port, str(tmp_path)), nprocs=8,
+        join=True, start_method="spawn",
+    )
+    seen = set()
+    for rank in range(8):
+        with open(tmp_path / f"rank{rank}.pkl", "rb") as f:
+            local_nodes, stack = pickle.load(f)
+        assert len(local_nodes) == 4  # 32/8 contiguous packing
+        for li, i in enumerate(local_nodes):
+            seen.add(i)
+            torch.testing.assert_close(
+                torch.from_numpy(stack[li]), golden[i],
+                rtol=0, atol=0,
+            )
+    assert seen == set(range(32))
