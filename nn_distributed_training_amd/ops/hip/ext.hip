@@ -457,7 +457,14 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
         parts = torch::empty({L * nchunk * O * I}, gstack.options());
         pp = parts.data_ptr<scalar_t>();
       }
-      if (I <= 256) {
+      if (I <= 64) {
+        // fi_per == 1: NIMAX=1 keeps the load loops guard-free
+        hipLaunchKernelGGL((gmfma::mfma_dw_direct_k<scalar_t, 1>),
+            grid, dim3(512), 0, cur_stream(),
+            dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+            gstack.data_ptr<scalar_t>(), pp, n, w_off, b_off,
+            (int)M, (int)I, (int)O, (int)nchunk);
+      } else if (I <= 256) {
         hipLaunchKernelGGL((gmfma::mfma_dw_direct_k<scalar_t, 4>),
             grid, dim3(512), 0, cur_stream(),
             dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),

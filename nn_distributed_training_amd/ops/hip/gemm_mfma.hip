@@ -107,46 +107,22 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
   vec2 ra[4], rb[4];
 
   const int nstages = (I + FK - 1) / FK;
-  // ---- load stage 0 into registers
-  {
-    const int k0 = 0;
+  // Full tiles/stages load guard-free: runtime bounds-guards on each
+  // load of an unrolled chain force per-element branch + vmcnt(0)
+  // waits that serialize the prefetch (trap 4c)
+  const bool full_mo = (m0 + BM) <= M && (o0 + BN) <= O;
+  const auto load_stage = [&](int k0) {
+    if (full_mo && k0 + FK <= I) {
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int m = sm + q * 16;
-      const int k = k0 + 2 * skp;
-      ra[q] = vec2{0, 0};
-      rb[q] = vec2{0, 0};
-      if (m0 + m < M) {
-        if (k + 1 < I) {
-          ra[q] = *reinterpret_cast<const vec2*>(
-              &Xl[(long)(m0 + m) * I + k]);
-        } else if (k < I) {
-          ra[q].x = Xl[(long)(m0 + m) * I + k];
-        }
+      for (int q = 0; q < 4; ++q) {
+        const int m = sm + q * 16;
+        const int k = k0 + 2 * skp;
+        ra[q] = *reinterpret_cast<const vec2*>(
+            &Xl[(long)(m0 + m) * I + k]);
+        rb[q] = *reinterpret_cast<const vec2*>(
+            &W[(long)(o0 + m) * I + k]);
       }
-      if (o0 + m < O) {
-        if (k + 1 < I) {
-          rb[q] = *reinterpret_cast<const vec2*>(
-              &W[(long)(o0 + m) * I + k]);
-        } else if (k < I) {
-          rb[q].x = W[(long)(o0 + m) * I + k];
-        }
-      }
-    }
-  }
-
-  for (int s = 0; s < nstages; ++s) {
-    __syncthreads();  // previous compute done: LDS free
-#pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int m = sm + q * 16;
-      As[2 * skp][m] = ra[q].x;
-      As[2 * skp + 1][m] = ra[q].y;
-      Bs[2 * skp][m] = rb[q].x;
-      Bs[2 * skp + 1][m] = rb[q].y;
-    }
-    if (s + 1 < nstages) {  // issue next stage's fetch NOW (T14)
-      const int k0 = (s + 1) * FK;
+    } else {
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         const int m = sm + q * 16;
@@ -170,6 +146,22 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
           }
         }
       }
+    }
+  };
+  load_stage(0);
+
+  for (int s = 0; s < nstages; ++s) {
+    __syncthreads();  // previous compute done: LDS free
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      As[2 * skp][m] = ra[q].x;
+      As[2 * skp + 1][m] = ra[q].y;
+      Bs[2 * skp][m] = rb[q].x;
+      Bs[2 * skp + 1][m] = rb[q].y;
+    }
+    if (s + 1 < nstages) {  // issue next stage's fetch NOW (T14)
+      load_stage((s + 1) * FK);
     }
     __syncthreads();  // LDS image of stage s visible
     const int klim = min(FK, I - s * FK);
@@ -266,55 +258,21 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
   const int bip = tid % 32;   // i-pair within the row
   vec2 rb2[4];                // 4 rows per thread over the 32-k stage
 
-  // ---- load stage 0
-  {
-    const int k0 = 0;
+  // full tiles/stages load guard-free (trap 4c, see mfma_fwd_k)
+  const bool full_mi = (m0 + BM) <= M && (i0 + BN) <= I;
+  const auto load_stage = [&](int k0) {
+    if (full_mi && k0 + FK <= O) {
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int m = sm + q * 16;
-      const int k = k0 + 2 * skp;
-      ra[q] = vec2{0, 0};
-      if (m0 + m < M) {
-        if (k + 1 < O) {
-          ra[q] = *reinterpret_cast<const vec2*>(
-              &Gl[(long)(m0 + m) * O + k]);
-        } else if (k < O) {
-          ra[q].x = Gl[(long)(m0 + m) * O + k];
-        }
+      for (int q = 0; q < 4; ++q) {
+        ra[q] = *reinterpret_cast<const vec2*>(
+            &Gl[(long)(m0 + sm + q * 16) * O + k0 + 2 * skp]);
       }
-    }
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int o = k0 + bo + q * 8;
-      const int i = i0 + 2 * bip;
-      rb2[q] = vec2{0, 0};
-      if (o < O) {
-        if (i + 1 < I) {
-          rb2[q] = *reinterpret_cast<const vec2*>(
-              &W[(long)o * I + i]);
-        } else if (i < I) {
-          rb2[q].x = W[(long)o * I + i];
-        }
+      for (int q = 0; q < 4; ++q) {
+        rb2[q] = *reinterpret_cast<const vec2*>(
+            &W[(long)(k0 + bo + q * 8) * I + i0 + 2 * bip]);
       }
-    }
-  }
-
-  for (int s = 0; s < nstages; ++s) {
-    __syncthreads();
-#pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int m = sm + q * 16;
-      As[2 * skp][m] = ra[q].x;
-      As[2 * skp + 1][m] = ra[q].y;
-    }
-#pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      const int kb = bo + q * 8;
-      Bs[kb][2 * bip] = rb2[q].x;
-      Bs[kb][2 * bip + 1] = rb2[q].y;
-    }
-    if (s + 1 < nstages) {
-      const int k0 = (s + 1) * FK;
+    } else {
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         const int m = sm + q * 16;
@@ -343,6 +301,26 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
           }
         }
       }
+    }
+  };
+  load_stage(0);
+
+  for (int s = 0; s < nstages; ++s) {
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      As[2 * skp][m] = ra[q].x;
+      As[2 * skp + 1][m] = ra[q].y;
+    }
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int kb = bo + q * 8;
+      Bs[kb][2 * bip] = rb2[q].x;
+      Bs[kb][2 * bip + 1] = rb2[q].y;
+    }
+    if (s + 1 < nstages) {
+      load_stage((s + 1) * FK);
     }
     __syncthreads();
 #pragma unroll
@@ -668,7 +646,15 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
     // loads in flight against the ~900-cycle HBM latency)
     bool have = k + 4 <= mhi;
     T a0 = T(0), a1 = T(0), b[NIMAX] = {};
-    if (have) {
+    if (have && ni == NIMAX) {
+      const long ka = k + lk;
+      a0 = ga0[ka * O];
+      a1 = ga1[ka * O];
+#pragma unroll
+      for (int fi = 0; fi < NIMAX; ++fi) {
+        b[fi] = gb[ka * I + fi * 16];
+      }
+    } else if (have) {
       const long ka = k + lk;
       a0 = ga0[ka * O];
       a1 = ga1[ka * O];
@@ -677,33 +663,66 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
         if (fi < ni) b[fi] = gb[ka * I + fi * 16];
       }
     }
-    while (have) {
-      const int kn = k + 4;
-      const bool haven = kn + 4 <= mhi;
-      T a0n = T(0), a1n = T(0), bn[NIMAX] = {};
-      if (haven) {
-        const long kan = kn + lk;
-        a0n = ga0[kan * O];
-        a1n = ga1[kan * O];
+    // guard-free body when this wave owns ALL NIMAX fragments (the
+    // I%64==0 shapes, i.e. every density layer): a runtime `fi < ni`
+    // inside the unrolled load chain forces per-element branch +
+    // vmcnt(0) waits (trap 4c — the same bug cost fc_block 3x)
+    if (ni == NIMAX) {
+      while (have) {
+        const int kn = k + 4;
+        const bool haven = kn + 4 <= mhi;
+        T a0n = T(0), a1n = T(0), bn[NIMAX] = {};
+        if (haven) {
+          const long kan = kn + lk;
+          a0n = ga0[kan * O];
+          a1n = ga1[kan * O];
+#pragma unroll
+          for (int fi = 0; fi < NIMAX; ++fi) {
+            bn[fi] = gb[kan * I + fi * 16];
+          }
+        }
+        if (bias_wave) { db0 += a0; db1 += a1; }
 #pragma unroll
         for (int fi = 0; fi < NIMAX; ++fi) {
-          if (fi < ni) bn[fi] = gb[kan * I + fi * 16];
-        }
-      }
-      if (bias_wave) { db0 += a0; db1 += a1; }
-#pragma unroll
-      for (int fi = 0; fi < NIMAX; ++fi) {
-        if (fi < ni) {
           acc[0][fi] = MF::mma(a0, b[fi], acc[0][fi]);
           acc[1][fi] = MF::mma(a1, b[fi], acc[1][fi]);
         }
-      }
-      a0 = a0n;
-      a1 = a1n;
+        a0 = a0n;
+        a1 = a1n;
 #pragma unroll
-      for (int fi = 0; fi < NIMAX; ++fi) b[fi] = bn[fi];
-      k = kn;
-      have = haven;
+        for (int fi = 0; fi < NIMAX; ++fi) b[fi] = bn[fi];
+        k = kn;
+        have = haven;
+      }
+    } else {
+      while (have) {
+        const int kn = k + 4;
+        const bool haven = kn + 4 <= mhi;
+        T a0n = T(0), a1n = T(0), bn[NIMAX] = {};
+        if (haven) {
+          const long kan = kn + lk;
+          a0n = ga0[kan * O];
+          a1n = ga1[kan * O];
+#pragma unroll
+          for (int fi = 0; fi < NIMAX; ++fi) {
+            if (fi < ni) bn[fi] = gb[kan * I + fi * 16];
+          }
+        }
+        if (bias_wave) { db0 += a0; db1 += a1; }
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) {
+          if (fi < ni) {
+            acc[0][fi] = MF::mma(a0, b[fi], acc[0][fi]);
+            acc[1][fi] = MF::mma(a1, b[fi], acc[1][fi]);
+          }
+        }
+        a0 = a0n;
+        a1 = a1n;
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) b[fi] = bn[fi];
+        k = kn;
+        have = haven;
+      }
     }
     if (k < mhi) {  // ragged tail (< 4 rows): zero-padded operands
       const long ka = k + lk;
