@@ -182,7 +182,30 @@ __global__ __launch_bounds__(256) void mfma_fwd_k(
     }
   }
 
-  // epilogue: bias + activation, coalesced per-fragment store
+  // epilogue: bias + activation, coalesced per-fragment store.
+  // Full tiles store guard-free (per-element bounds guards serialize
+  // the store chain — trap 4c).
+  if (full_mo && Z == nullptr) {
+    T bv[2];
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      bv[fn] = bias[o0 + wn + fn * 16 + (lane & 15)];
+    }
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
+          const int o = o0 + wn + fn * 16 + (lane & 15);
+          const long off = l * (long)M * O + (long)m * O + o;
+          Y[off] = act_fwd(act, acc[fm][fn][r] + bv[fn], scale);
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
@@ -340,6 +363,79 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
 
   const T* Wb = theta + l * n + wb_off;
   const T* bb = theta + l * n + bb_off;
+  if (full_mi && act_below == ACT_RELU) {
+    // guard-free fast path for the relu-below layers: prefetch the 16
+    // Yb mask values unconditionally, then compute + store (per-
+    // element bounds guards serialized this epilogue — trap 4c)
+    T yv[2][2][4];
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
+          const int i = i0 + wn + fn * 16 + (lane & 15);
+          yv[fm][fn][r] = Yb[l * (long)M * I + (long)m * I + i];
+        }
+      }
+    }
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
+          const int i = i0 + wn + fn * 16 + (lane & 15);
+          const long off = l * (long)M * I + (long)m * I + i;
+          dX[off] = yv[fm][fn][r] > T(0) ? acc[fm][fn][r] : T(0);
+        }
+      }
+    }
+    return;
+  }
+  if (full_mi && act_below == ACT_SIN_RELU && Xb2 != nullptr
+      && Ib == 2) {
+    // encode-below fast path: z recomputed from the 2-wide input,
+    // all loads issued before the sin/cos block
+    T x0v[2][4], x1v[2][4];
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
+        const long xoff = (long)(l * (long)M + m) * 2;
+        x0v[fm][r] = Xb2[xoff];
+        x1v[fm][r] = Xb2[xoff + 1];
+      }
+    }
+    T w0v[2], w1v[2], bv[2];
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      const int i = i0 + wn + fn * 16 + (lane & 15);
+      w0v[fn] = Wb[(long)i * 2];
+      w1v[fn] = Wb[(long)i * 2 + 1];
+      bv[fn] = bb[i];
+    }
+#pragma unroll
+    for (int fm = 0; fm < 2; ++fm) {
+#pragma unroll
+      for (int fn = 0; fn < 2; ++fn) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = m0 + wm + fm * 16 + MF::acc_row(lane, r);
+          const int i = i0 + wn + fn * 16 + (lane & 15);
+          const long off = l * (long)M * I + (long)m * I + i;
+          const T z = bv[fn] + x0v[fm][r] * w0v[fn]
+                      + x1v[fm][r] * w1v[fn];
+          dX[off] = acc[fm][fn][r]
+                    * act_bwd(ACT_SIN_RELU, z, T(0), scale_below);
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int fm = 0; fm < 2; ++fm) {
 #pragma unroll
@@ -663,37 +759,62 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
         if (fi < ni) b[fi] = gb[ka * I + fi * 16];
       }
     }
-    // guard-free body when this wave owns ALL NIMAX fragments (the
-    // I%64==0 shapes, i.e. every density layer): a runtime `fi < ni`
-    // inside the unrolled load chain forces per-element branch +
-    // vmcnt(0) waits (trap 4c — the same bug cost fc_block 3x)
+    // guard-free DEPTH-4 pipeline when this wave owns ALL NIMAX
+    // fragments (the I%64==0 shapes, i.e. every density layer): one
+    // step of lookahead only covers ~512 MFMA-issue cycles against the
+    // ~900-cycle HBM latency, so four k-steps of operands stay in
+    // flight; the drain loop keeps the refill guard-free (trap 4c).
     if (ni == NIMAX) {
-      while (have) {
-        const int kn = k + 4;
-        const bool haven = kn + 4 <= mhi;
-        T a0n = T(0), a1n = T(0), bn[NIMAX] = {};
-        if (haven) {
-          const long kan = kn + lk;
-          a0n = ga0[kan * O];
-          a1n = ga1[kan * O];
-#pragma unroll
-          for (int fi = 0; fi < NIMAX; ++fi) {
-            bn[fi] = gb[kan * I + fi * 16];
-          }
-        }
-        if (bias_wave) { db0 += a0; db1 += a1; }
+      const int nfull = (mhi - k) / 4;
+      // depth = latency(~900cyc) / MFMA-cover-per-step: NIMAX=1 steps
+      // cover only 2x64 issue cycles -> depth 8; NIMAX>=4 cover
+      // 8x64 -> depth 2. Powers of two keep the slot mask cheap.
+      constexpr int D = (NIMAX == 1) ? 8 : 2;
+      T a0p[D], a1p[D], bp[D][NIMAX];
+      const int lead = nfull < D ? nfull : D;
+      for (int d = 0; d < lead; ++d) {
+        const long kd = k + 4 * d + lk;
+        a0p[d] = ga0[kd * O];
+        a1p[d] = ga1[kd * O];
 #pragma unroll
         for (int fi = 0; fi < NIMAX; ++fi) {
-          acc[0][fi] = MF::mma(a0, b[fi], acc[0][fi]);
-          acc[1][fi] = MF::mma(a1, b[fi], acc[1][fi]);
+          bp[d][fi] = gb[kd * I + fi * 16];
         }
-        a0 = a0n;
-        a1 = a1n;
-#pragma unroll
-        for (int fi = 0; fi < NIMAX; ++fi) b[fi] = bn[fi];
-        k = kn;
-        have = haven;
       }
+      int step = 0;
+      const int body = nfull > D ? nfull - D : 0;
+      for (; step < body; ++step) {
+        const int slot = step & (D - 1);
+        const T a0c = a0p[slot], a1c = a1p[slot];
+        T bc[NIMAX];
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) bc[fi] = bp[slot][fi];
+        const long kf = k + 4 * (step + D) + lk;
+        a0p[slot] = ga0[kf * O];
+        a1p[slot] = ga1[kf * O];
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) {
+          bp[slot][fi] = gb[kf * I + fi * 16];
+        }
+        if (bias_wave) { db0 += a0c; db1 += a1c; }
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) {
+          acc[0][fi] = MF::mma(a0c, bc[fi], acc[0][fi]);
+          acc[1][fi] = MF::mma(a1c, bc[fi], acc[1][fi]);
+        }
+      }
+      for (; step < nfull; ++step) {  // drain (no refill)
+        const int slot = step & (D - 1);
+        if (bias_wave) { db0 += a0p[slot]; db1 += a1p[slot]; }
+#pragma unroll
+        for (int fi = 0; fi < NIMAX; ++fi) {
+          acc[0][fi] = MF::mma(a0p[slot], bp[slot][fi], acc[0][fi]);
+          acc[1][fi] = MF::mma(a1p[slot], bp[slot][fi], acc[1][fi]);
+        }
+      }
+      k += 4 * nfull;
+      have = false;
+      (void)a0; (void)a1; (void)b;
     } else {
       while (have) {
         const int kn = k + 4;
