@@ -766,50 +766,73 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
     // flight; the drain loop keeps the refill guard-free (trap 4c).
     if (ni == NIMAX) {
       const int nfull = (mhi - k) / 4;
-      // depth = latency(~900cyc) / MFMA-cover-per-step: NIMAX=1 steps
-      // cover only 2x64 issue cycles -> depth 8; NIMAX>=4 cover
-      // 8x64 -> depth 2. Powers of two keep the slot mask cheap.
+      // Guard-free software pipeline with COMPILE-TIME slot indices
+      // (a runtime-indexed operand array lives in scratch — rule 20;
+      // the first modulo-slot version ran 1.7x SLOWER). The main loop
+      // processes D steps per iteration so each slot index is the
+      // unrolled d; depth covers the ~900-cycle HBM latency against
+      // each step's MFMA-issue cover (NIMAX=1 steps cover only
+      // 2x64 cycles -> deeper).
       constexpr int D = (NIMAX == 1) ? 8 : 2;
-      T a0p[D], a1p[D], bp[D][NIMAX];
-      const int lead = nfull < D ? nfull : D;
-      for (int d = 0; d < lead; ++d) {
-        const long kd = k + 4 * d + lk;
-        a0p[d] = ga0[kd * O];
-        a1p[d] = ga1[kd * O];
-#pragma unroll
-        for (int fi = 0; fi < NIMAX; ++fi) {
-          bp[d][fi] = gb[kd * I + fi * 16];
-        }
-      }
       int step = 0;
-      const int body = nfull > D ? nfull - D : 0;
-      for (; step < body; ++step) {
-        const int slot = step & (D - 1);
-        const T a0c = a0p[slot], a1c = a1p[slot];
-        T bc[NIMAX];
+      if (nfull >= D) {
+        T a0p[D], a1p[D], bp[D][NIMAX];
 #pragma unroll
-        for (int fi = 0; fi < NIMAX; ++fi) bc[fi] = bp[slot][fi];
-        const long kf = k + 4 * (step + D) + lk;
-        a0p[slot] = ga0[kf * O];
-        a1p[slot] = ga1[kf * O];
+        for (int d = 0; d < D; ++d) {
+          const long kd = k + 4 * d + lk;
+          a0p[d] = ga0[kd * O];
+          a1p[d] = ga1[kd * O];
 #pragma unroll
-        for (int fi = 0; fi < NIMAX; ++fi) {
-          bp[slot][fi] = gb[kf * I + fi * 16];
+          for (int fi = 0; fi < NIMAX; ++fi) {
+            bp[d][fi] = gb[kd * I + fi * 16];
+          }
         }
+        const int body = nfull - D;
+        for (; step + D <= body; step += D) {
+#pragma unroll
+          for (int d = 0; d < D; ++d) {
+            const T a0c = a0p[d], a1c = a1p[d];
+            T bc[NIMAX];
+#pragma unroll
+            for (int fi = 0; fi < NIMAX; ++fi) bc[fi] = bp[d][fi];
+            const long kf = k + 4 * (step + d + D) + lk;
+            a0p[d] = ga0[kf * O];
+            a1p[d] = ga1[kf * O];
+#pragma unroll
+            for (int fi = 0; fi < NIMAX; ++fi) {
+              bp[d][fi] = gb[kf * I + fi * 16];
+            }
+            if (bias_wave) { db0 += a0c; db1 += a1c; }
+#pragma unroll
+            for (int fi = 0; fi < NIMAX; ++fi) {
+              acc[0][fi] = MF::mma(a0c, bc[fi], acc[0][fi]);
+              acc[1][fi] = MF::mma(a1c, bc[fi], acc[1][fi]);
+            }
+          }
+        }
+        // consume the last in-flight slots (steps step..step+D-1 are
+        // resident; anything beyond reloads in the plain tail below)
+#pragma unroll
+        for (int d = 0; d < D; ++d) {
+          if (bias_wave) { db0 += a0p[d]; db1 += a1p[d]; }
+#pragma unroll
+          for (int fi = 0; fi < NIMAX; ++fi) {
+            acc[0][fi] = MF::mma(a0p[d], bp[d][fi], acc[0][fi]);
+            acc[1][fi] = MF::mma(a1p[d], bp[d][fi], acc[1][fi]);
+          }
+        }
+        step += D;
+      }
+      for (; step < nfull; ++step) {  // plain tail (< 2D steps)
+        const long kd = k + 4 * step + lk;
+        const T a0c = ga0[kd * O];
+        const T a1c = ga1[kd * O];
         if (bias_wave) { db0 += a0c; db1 += a1c; }
 #pragma unroll
         for (int fi = 0; fi < NIMAX; ++fi) {
-          acc[0][fi] = MF::mma(a0c, bc[fi], acc[0][fi]);
-          acc[1][fi] = MF::mma(a1c, bc[fi], acc[1][fi]);
-        }
-      }
-      for (; step < nfull; ++step) {  // drain (no refill)
-        const int slot = step & (D - 1);
-        if (bias_wave) { db0 += a0p[slot]; db1 += a1p[slot]; }
-#pragma unroll
-        for (int fi = 0; fi < NIMAX; ++fi) {
-          acc[0][fi] = MF::mma(a0p[slot], bp[slot][fi], acc[0][fi]);
-          acc[1][fi] = MF::mma(a1p[slot], bp[slot][fi], acc[1][fi]);
+          const T bc = gb[kd * I + fi * 16];
+          acc[0][fi] = MF::mma(a0c, bc, acc[0][fi]);
+          acc[1][fi] = MF::mma(a1c, bc, acc[1][fi]);
         }
       }
       k += 4 * nfull;
