@@ -398,7 +398,7 @@ void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
       const char* e = getenv("NDTA_DX_NEW");
       return !(e && e[0] == '0');
     }();
-    if (use_mfma && dx_new) {
+    if (use_mfma && dx_new && O <= 64) {
       dim3 grid((I + 63) / 64, (M + 63) / 64, L);
       hipLaunchKernelGGL(gmfma::mfma_dx_k<scalar_t>,
           grid, dim3(256), 0, cur_stream(),
@@ -499,11 +499,20 @@ void linear_bwd_dw(torch::Tensor dZ, torch::Tensor X,
     } else if (M > 2048 && I <= 4 && O <= 256) {
       long nchunk = std::max<long>(1, 1024 / std::max<long>(1, L));
       nchunk = std::min<long>(nchunk, (M + 255) / 256);
-      hipLaunchKernelGGL((gemm::dw_skinny_i_k<scalar_t, 4>),
-          dim3(1, 1, L * nchunk), dim3(256), 0, cur_stream(),
-          dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
-          gstack.data_ptr<scalar_t>(), n, w_off, b_off,
-          (int)M, (int)I, (int)O, (int)nchunk);
+      if (I == 2 && O == 256) {  // FourierNet encode shape: all
+        // bounds compile-time, unconditional loads (trap 4c)
+        hipLaunchKernelGGL((gemm::dw_skinny_i_exact_k<scalar_t, 2>),
+            dim3(1, 1, L * nchunk), dim3(256), 0, cur_stream(),
+            dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+            gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+            (int)M, (int)I, (int)O, (int)nchunk);
+      } else {
+        hipLaunchKernelGGL((gemm::dw_skinny_i_k<scalar_t, 4>),
+            dim3(1, 1, L * nchunk), dim3(256), 0, cur_stream(),
+            dZ.data_ptr<scalar_t>(), X.data_ptr<scalar_t>(),
+            gstack.data_ptr<scalar_t>(), n, w_off, b_off,
+            (int)M, (int)I, (int)O, (int)nchunk);
+      }
     } else if (M > 2048 && O <= 4 && I <= 256) {
       long nchunk = std::max<long>(1, 1024 / std::max<long>(1, L));
       nchunk = std::min<long>(nchunk, (M + 255) / 256);

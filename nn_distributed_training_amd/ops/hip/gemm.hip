@@ -266,6 +266,67 @@ __global__ void dw_small_chunked_k(
 //   O <= OMAX (density head, dW[1, 64]): lanes own i-columns, X rows
 //     read coalesced, the tiny dZ row broadcast-loaded.
 // M is chunked over blocks; atomic accumulation (grad slice zeroed).
+// Exact-shape fast variant (I == IMAX, O == 4*WAVE): all loop bounds
+// compile-time, loads unconditional — the runtime `i < I` / `o < O`
+// guards in the generic kernel serialized its load pipeline (trap 4c;
+// measured 169 us vs a ~52 us traffic roofline on the encode dW).
+// The m-loop unrolls 4-wide for ILP against the ~900-cycle latency.
+template <typename T, int IMAX>
+__global__ void dw_skinny_i_exact_k(
+    const T* __restrict__ dZ, const T* __restrict__ X,
+    T* __restrict__ gstack, long n, long w_off, long b_off,
+    int M, int I, int O, int nchunk) {
+  const int chunk = blockIdx.z % nchunk;
+  const long l = blockIdx.z / nchunk;
+  const int mc = (M + nchunk - 1) / nchunk;
+  const int mlo = chunk * mc;
+  const int mhi = min(M, mlo + mc);
+  const T* dZl = dZ + l * (long)M * O;
+  const T* Xl = X + l * (long)M * I;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+
+  T acc[4][IMAX + 1] = {};
+  int m = mlo + wid;
+  for (; m + 12 < mhi; m += 16) {  // 4 rows per wave-iteration
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const long mr = m + 4 * u;
+      T xv[IMAX];
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) xv[i] = Xl[mr * I + i];
+#pragma unroll
+      for (int no = 0; no < 4; ++no) {
+        const T g = dZl[mr * O + no * WAVE + lane];
+        acc[no][IMAX] += g;
+#pragma unroll
+        for (int i = 0; i < IMAX; ++i) acc[no][i] += g * xv[i];
+      }
+    }
+  }
+  for (; m < mhi; m += 4) {  // tail rows
+    T xv[IMAX];
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) xv[i] = Xl[(long)m * I + i];
+#pragma unroll
+    for (int no = 0; no < 4; ++no) {
+      const T g = dZl[(long)m * O + no * WAVE + lane];
+      acc[no][IMAX] += g;
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) acc[no][i] += g * xv[i];
+    }
+  }
+#pragma unroll
+  for (int no = 0; no < 4; ++no) {
+    const int o = no * WAVE + lane;
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      atomicAdd(&gstack[l * n + w_off + (long)o * I + i], acc[no][i]);
+    }
+    atomicAdd(&gstack[l * n + b_off + o], acc[no][IMAX]);
+  }
+}
+
 template <typename T, int IMAX>
 __global__ void dw_skinny_i_k(
     const T* __restrict__ dZ, const T* __restrict__ X,

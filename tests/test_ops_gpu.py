@@ -90,7 +90,8 @@ def test_linear_fwd_sin_relu(ext, dtype, M):
         (2, 29, 23, 17),     # VALU path (M < 32)
         (2, 150, 70, 40),    # MFMA fwd/dx/dw with ragged edges
         (2, 4096, 24, 2),    # dw_skinny_o path (O <= 4, big M)
-        (1, 4096, 2, 200),   # dw_skinny_i path (FourierNet encode shape)
+        (1, 4096, 2, 200),   # dw_skinny_i path (generic)
+        (1, 4096, 2, 256),   # dw_skinny_i_exact (encode shape I=2 O=256)
         (1, 4096, 17, 9),    # dw_small_chunked fallback
         (1, 4096, 64, 48),   # MFMA dw chunked accumulation + fused db
         (2, 4096, 64, 64),   # dw_direct full-I (density 64->64 shape)
