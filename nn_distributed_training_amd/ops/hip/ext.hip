@@ -398,7 +398,7 @@ void linear_bwd_dx(torch::Tensor dZ, torch::Tensor theta,
       const char* e = getenv("NDTA_DX_NEW");
       return !(e && e[0] == '0');
     }();
-    if (use_mfma && dx_new && O <= 64) {
+    if (use_mfma && dx_new) {
       dim3 grid((I + 63) / 64, (M + 63) / 64, L);
       hipLaunchKernelGGL(gmfma::mfma_dx_k<scalar_t>,
           grid, dim3(256), 0, cur_stream(),

@@ -247,17 +247,11 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
     long n, long w_off, int M, int I, int O,
     const T* __restrict__ Xb2,  // below layer INPUT (recompute mode)
     long wb_off, long bb_off, int Ib) {
-  // Round-2c structure (O <= 64 — the only shapes dispatched here;
-  // larger O falls back to mfma_dx16_k): the ENTIRE contraction
-  // (K = O) stages once into a single LDS dZ^T image with zero-padded
-  // rows to 64, so the MFMA loop is a fully unrolled, branch-free
-  // 16-step sweep; the W operand reads DIRECT from global (rows are
-  // contiguous in i — coalesced), letting the compiler hoist all 32
-  // independent loads. One barrier per block instead of 2 per stage.
   using MF = mfma_t<T>;
   using acc_t = typename MF::acc_t;
   typedef T vec2 __attribute__((ext_vector_type(2)));
-  __shared__ T As[64][BM + 1];   // As[k=o][m] = dZ[m][o], rows >= O zero
+  __shared__ T As[FK][BM + 1];   // As[k=o][m] = dZ[m][o]
+  __shared__ T Bs[FK][BN + 1];   // Bs[k=o][i] = W[o][i]
 
   const long l = blockIdx.z;
   const T* Gl = dZ + l * (long)M * O;
@@ -270,73 +264,100 @@ __global__ __launch_bounds__(256) void mfma_dx_k(
   const int wid = tid >> 6;
   const int wm = (wid >> 1) * 32;
   const int wn = (wid & 1) * 32;
-  const int lo = lane & 15;
-  const int lk = lane >> 4;
 
-  // stage: pair u = tid + 256*q covers row m = u/32, k-pair u%32
-  const int sm = tid / 32;
-  const int skp = tid % 32;
-  const bool full_mi = (m0 + BM) <= M && (i0 + BN) <= I;
-  if (full_mi && O == 64) {
-#pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      const int m = sm + q * 8;
-      const vec2 v = *reinterpret_cast<const vec2*>(
-          &Gl[(long)(m0 + m) * O + 2 * skp]);
-      As[2 * skp][m] = v.x;
-      As[2 * skp + 1][m] = v.y;
-    }
-  } else {
-#pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      const int m = sm + q * 8;
-      const int k = 2 * skp;
-      vec2 v = vec2{0, 0};
-      if (m0 + m < M) {
-        if (k + 1 < O) {
-          v = *reinterpret_cast<const vec2*>(
-              &Gl[(long)(m0 + m) * O + k]);
-        } else if (k < O) {
-          v.x = Gl[(long)(m0 + m) * O + k];
-        }
-      }
-      As[2 * skp][m] = v.x;
-      As[2 * skp + 1][m] = v.y;
-    }
-  }
-  __syncthreads();
-
+  const int sm = tid / 16;   // staged row (m for A, o-row for B image)
+  const int skp = tid % 16;  // k-pair within the row
   acc_t acc[2][2] = {};
-  if (full_mi && O == 64) {
+  vec2 ra[4];
+
+  // A: As[k][m] = dZ[m0+m][k]  (k = contraction over O, row-major in
+  // dZ so pairs are contiguous); B: Bs[k][i] = W[k][i0+i] (W rows
+  // contiguous in i).
+  const int nstages = (O + FK - 1) / FK;
+  // stage A via vec2 over k (pairs contiguous in dZ rows); stage B
+  // with vec2 over i (pairs contiguous in W rows): thread covers W
+  // rows bo + q*8 (q < 4 -> 32 k-rows) at i-pair bip.
+  const int bo = tid / 32;    // W row (o) of pair 0 for B staging
+  const int bip = tid % 32;   // i-pair within the row
+  vec2 rb2[4];                // 4 rows per thread over the 32-k stage
+
+  // full tiles/stages load guard-free (trap 4c, see mfma_fwd_k)
+  const bool full_mi = (m0 + BM) <= M && (i0 + BN) <= I;
+  const auto load_stage = [&](int k0) {
+    if (full_mi && k0 + FK <= O) {
 #pragma unroll
-    for (int kk = 0; kk < 16; ++kk) {
-      const int ka = 4 * kk + lk;
-      const T a0 = As[ka][wm + lo];
-      const T a1 = As[ka][wm + 16 + lo];
-      const T b0 = W[(long)ka * I + i0 + wn + lo];
-      const T b1 = W[(long)ka * I + i0 + wn + 16 + lo];
-      acc[0][0] = MF::mma(a0, b0, acc[0][0]);
-      acc[0][1] = MF::mma(a0, b1, acc[0][1]);
-      acc[1][0] = MF::mma(a1, b0, acc[1][0]);
-      acc[1][1] = MF::mma(a1, b1, acc[1][1]);
-    }
-  } else {
-    const int nk = (O + 3) / 4;
-    for (int kk = 0; kk < nk; ++kk) {
-      const int ka = 4 * kk + lk;
-      const T a0 = ka < O ? As[ka][wm + lo] : T(0);
-      const T a1 = ka < O ? As[ka][wm + 16 + lo] : T(0);
-      T b0 = T(0), b1 = T(0);
-      if (ka < O) {
-        if (i0 + wn + lo < I) b0 = W[(long)ka * I + i0 + wn + lo];
-        if (i0 + wn + 16 + lo < I) {
-          b1 = W[(long)ka * I + i0 + wn + 16 + lo];
+      for (int q = 0; q < 4; ++q) {
+        ra[q] = *reinterpret_cast<const vec2*>(
+            &Gl[(long)(m0 + sm + q * 16) * O + k0 + 2 * skp]);
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        rb2[q] = *reinterpret_cast<const vec2*>(
+            &W[(long)(k0 + bo + q * 8) * I + i0 + 2 * bip]);
+      }
+    } else {
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int m = sm + q * 16;
+        const int k = k0 + 2 * skp;
+        ra[q] = vec2{0, 0};
+        if (m0 + m < M) {
+          if (k + 1 < O) {
+            ra[q] = *reinterpret_cast<const vec2*>(
+                &Gl[(long)(m0 + m) * O + k]);
+          } else if (k < O) {
+            ra[q].x = Gl[(long)(m0 + m) * O + k];
+          }
         }
       }
-      acc[0][0] = MF::mma(a0, b0, acc[0][0]);
-      acc[0][1] = MF::mma(a0, b1, acc[0][1]);
-      acc[1][0] = MF::mma(a1, b0, acc[1][0]);
-      acc[1][1] = MF::mma(a1, b1, acc[1][1]);
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int o = k0 + bo + q * 8;
+        const int i = i0 + 2 * bip;
+        rb2[q] = vec2{0, 0};
+        if (o < O) {
+          if (i + 1 < I) {
+            rb2[q] = *reinterpret_cast<const vec2*>(
+                &W[(long)o * I + i]);
+          } else if (i < I) {
+            rb2[q].x = W[(long)o * I + i];
+          }
+        }
+      }
+    }
+  };
+  load_stage(0);
+
+  for (int s = 0; s < nstages; ++s) {
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int m = sm + q * 16;
+      As[2 * skp][m] = ra[q].x;
+      As[2 * skp + 1][m] = ra[q].y;
+    }
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int kb = bo + q * 8;
+      Bs[kb][2 * bip] = rb2[q].x;
+      Bs[kb][2 * bip + 1] = rb2[q].y;
+    }
+    if (s + 1 < nstages) {
+      load_stage((s + 1) * FK);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < FK; kk += 4) {
+      const int ka = kk + (lane >> 4);
+#pragma unroll
+      for (int fm = 0; fm < 2; ++fm) {
+        const T a = As[ka][wm + fm * 16 + (lane & 15)];
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const T b = Bs[ka][wn + fn * 16 + (lane & 15)];
+          acc[fm][fn] = MF::mma(a, b, acc[fm][fn]);
+        }
+      }
     }
   }
 
