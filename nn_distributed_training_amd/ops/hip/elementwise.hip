@@ -420,4 +420,75 @@ __global__ void feistel_perm_k(long* __restrict__ out, long n, long lb,
   }
 }
 
+
+// ---------------------------------------------------------------------
+// Consensus-error metric kernels (eval-only; the one op that still ran
+// through torch on the gathered stack — reference
+// problems/dist_mnist_problem.py:152-175): row norms, pairwise
+// normalized L2 distances, normalized mean vector, distances to mean.
+
+template <typename T>
+__global__ void row_norms_k(const T* __restrict__ stack,
+                            T* __restrict__ norms, long n) {
+  __shared__ T scratch[BLOCK / WAVE];
+  const long i = blockIdx.x;
+  T acc = T(0);
+  for (long e = threadIdx.x; e < n; e += BLOCK) {
+    const T v = stack[i * n + e];
+    acc += v * v;
+  }
+  acc = block_reduce_sum<T, BLOCK>(acc, scratch);
+  if (threadIdx.x == 0) norms[i] = ::sqrt(acc);
+}
+
+// D[i, j] = || stack_i/|stack_i| - stack_j/|stack_j| ||_2
+template <typename T>
+__global__ void pairwise_normed_dist_k(
+    const T* __restrict__ stack, const T* __restrict__ norms,
+    T* __restrict__ D, long N, long n) {
+  __shared__ T scratch[BLOCK / WAVE];
+  const long i = blockIdx.x / N;
+  const long j = blockIdx.x % N;
+  const T ri = T(1) / norms[i];
+  const T rj = T(1) / norms[j];
+  T acc = T(0);
+  for (long e = threadIdx.x; e < n; e += BLOCK) {
+    const T d = stack[i * n + e] * ri - stack[j * n + e] * rj;
+    acc += d * d;
+  }
+  acc = block_reduce_sum<T, BLOCK>(acc, scratch);
+  if (threadIdx.x == 0) D[i * N + j] = ::sqrt(acc);
+}
+
+template <typename T>
+__global__ void normed_mean_k(const T* __restrict__ stack,
+                              const T* __restrict__ norms,
+                              T* __restrict__ mean, long N, long n) {
+  for (long e = blockIdx.x * (long)BLOCK + threadIdx.x; e < n;
+       e += (long)gridDim.x * BLOCK) {
+    T acc = T(0);
+    for (long i = 0; i < N; ++i) {
+      acc += stack[i * n + e] / norms[i];
+    }
+    mean[e] = acc / T(N);
+  }
+}
+
+template <typename T>
+__global__ void dist_to_mean_k(const T* __restrict__ stack,
+                               const T* __restrict__ norms,
+                               const T* __restrict__ mean,
+                               T* __restrict__ Dm, long n) {
+  __shared__ T scratch[BLOCK / WAVE];
+  const long i = blockIdx.x;
+  const T ri = T(1) / norms[i];
+  T acc = T(0);
+  for (long e = threadIdx.x; e < n; e += BLOCK) {
+    const T d = stack[i * n + e] * ri - mean[e];
+    acc += d * d;
+  }
+  acc = block_reduce_sum<T, BLOCK>(acc, scratch);
+  if (threadIdx.x == 0) Dm[i] = ::sqrt(acc);
+}
+
 }  // namespace ew

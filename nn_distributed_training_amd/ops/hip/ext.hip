@@ -767,6 +767,38 @@ void fc_block(torch::Tensor x0, torch::Tensor theta,
   linear_bwd_dw(dz1g, x0, grad, w1_off, b1_off, M, I, H);
 }
 
+// consensus-error metric: normalized pairwise distances + distance to
+// the normalized mean (reference dist_mnist_problem.py:152-175) on the
+// all-gathered [N, n] stack — returns (D [N,N], Dm [N,1])
+std::vector<torch::Tensor> consensus_cdist(torch::Tensor stack) {
+  CHECK_DEV(stack);
+  const long N = stack.size(0), n = stack.size(1);
+  auto opts = stack.options();
+  auto norms = torch::empty({N}, opts);
+  auto D = torch::empty({N, N}, opts);
+  auto mean = torch::empty({n}, opts);
+  auto Dm = torch::empty({N, 1}, opts);
+  DISPATCH_FT(stack, {
+    hipLaunchKernelGGL(ew::row_norms_k<scalar_t>,
+        dim3(N), dim3(ew::BLOCK), 0, cur_stream(),
+        stack.data_ptr<scalar_t>(), norms.data_ptr<scalar_t>(), n);
+    hipLaunchKernelGGL(ew::pairwise_normed_dist_k<scalar_t>,
+        dim3(N * N), dim3(ew::BLOCK), 0, cur_stream(),
+        stack.data_ptr<scalar_t>(), norms.data_ptr<scalar_t>(),
+        D.data_ptr<scalar_t>(), N, n);
+    hipLaunchKernelGGL(ew::normed_mean_k<scalar_t>,
+        dim3(grid_1d(n)), dim3(ew::BLOCK), 0, cur_stream(),
+        stack.data_ptr<scalar_t>(), norms.data_ptr<scalar_t>(),
+        mean.data_ptr<scalar_t>(), N, n);
+    hipLaunchKernelGGL(ew::dist_to_mean_k<scalar_t>,
+        dim3(N), dim3(ew::BLOCK), 0, cur_stream(),
+        stack.data_ptr<scalar_t>(), norms.data_ptr<scalar_t>(),
+        mean.data_ptr<scalar_t>(), Dm.data_ptr<scalar_t>(), n);
+  });
+  HIP_CHECK_LAST();
+  return {D, Dm};
+}
+
 // ----------------------------------------------------------- chains --
 // One pybind call per forward / backward pass instead of one per
 // layer-op: host profiling (BENCH r2b timing_breakdown) measured
@@ -960,6 +992,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bce_bwd", &bce_bwd);
   mod.def("regression_bwd", &regression_bwd);
   mod.def("feistel_perm", &feistel_perm);
+  mod.def("consensus_cdist", &consensus_cdist);
   mod.def("fc_block", &fc_block);
   mod.def("fwd_chain", &fwd_chain);
   mod.def("bwd_chain", &bwd_chain);

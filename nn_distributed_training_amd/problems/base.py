@@ -128,9 +128,16 @@ class ProblemBase:
             else:
                 local = self.local_params_stack()
             th_stack = self.comm.all_gather_stack(self.layout, local)
-            th_stack = torch.nn.functional.normalize(
-                th_stack.to(torch.float64), dim=1
-            )
+            th_stack = th_stack.to(torch.float64)
+            if self.stacked is not None and th_stack.is_cuda:
+                # first-party pairwise-distance kernels (the last
+                # eager-torch op in the system; tested vs torch.cdist
+                # in tests/test_ops_gpu.py)
+                d_all, d_mean = self.stacked.ext.consensus_cdist(
+                    th_stack.contiguous()
+                )
+                return d_all.cpu(), d_mean.cpu()
+            th_stack = torch.nn.functional.normalize(th_stack, dim=1)
             distances_all = torch.cdist(th_stack, th_stack)
             th_mean = th_stack.mean(dim=0, keepdim=True)
             distances_mean = torch.cdist(th_stack, th_mean)

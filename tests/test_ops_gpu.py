@@ -570,3 +570,19 @@ def test_fc_block_matches_autograd(ext, dtype, M, I, H, C):
             grad[l, w2_off:w2_off + C * H].reshape(C, H), W2.grad,
             **tol)
         torch.testing.assert_close(grad[l, b2_off:], b2.grad, **tol)
+
+
+@requires_gpu
+@pytest.mark.parametrize("N,n", [(8, 28440), (3, 100), (32, 25601)])
+def test_consensus_cdist_matches_torch(ext, N, n):
+    """First-party consensus-error kernels vs the torch reference
+    (normalize -> cdist -> cdist-to-mean, reference
+    problems/dist_mnist_problem.py:152-175)."""
+    torch.manual_seed(8)
+    stack = torch.randn(N, n, dtype=torch.float64, device=_dev())
+    D, Dm = ext.consensus_cdist(stack)
+    ref = torch.nn.functional.normalize(stack, dim=1)
+    Dref = torch.cdist(ref, ref)
+    Dmref = torch.cdist(ref, ref.mean(dim=0, keepdim=True))
+    torch.testing.assert_close(D, Dref, rtol=1e-9, atol=1e-9)
+    torch.testing.assert_close(Dm, Dmref, rtol=1e-9, atol=1e-9)
