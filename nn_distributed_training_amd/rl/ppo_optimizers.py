@@ -145,11 +145,100 @@ class DiNNOPPO(_PPOBase):
             for i in range(problem.N)
         }
 
+    # ------------------------------------------------------------------
+    def _hip_available(self):
+        if os.environ.get("NDTA_RL_HIP", "1") == "0":
+            return False
+        if self.device.type != "cuda":
+            return False
+        from ..ops import ext_available
+
+        return ext_available()
+
+    def _csr(self):
+        """CSR neighbor tables over all nodes (single-process RL)."""
+        if getattr(self, "_csr_cache", None) is None:
+            offs, idx = [0], []
+            for i in range(self.pr.N):
+                ns = self._neighbors(i)
+                idx.extend(ns)
+                offs.append(len(idx))
+            dev = self.device
+            self._csr_cache = (
+                torch.tensor(offs, dtype=torch.int32, device=dev),
+                torch.tensor(idx, dtype=torch.int32, device=dev),
+                torch.tensor(
+                    [len(self._neighbors(i)) for i in range(self.pr.N)],
+                    dtype=torch.int32, device=dev,
+                ),
+            )
+        return self._csr_cache
+
+    def _step_round_hip(self, it, lr):
+        """DiNNO round math on the stacked CDNA4 kernels (VERDICT r1
+        item 9; reference round math RL/dist_rl/dinnoPPO.py:87-131):
+        the [actor|critic] vectors of all nodes form one [N, n] stack;
+        dual ascent + s-reduction run in ONE dinno_dual_threg launch
+        and each primal step is ONE fused penalty-Adam launch (the
+        analytic consensus-penalty gradient folds into the update).
+        Rollouts and pred-loss autograd stay per-node torch (the env
+        stepping is host-side by construction)."""
+        from ..ops import get_ext
+
+        ext = get_ext()
+        pr = self.pr
+        dt = torch.get_default_dtype()
+        ths = torch.stack(
+            [pr.node_vector(i).detach() for i in range(pr.N)]
+        ).contiguous()
+        if getattr(self, "_duals_t", None) is None:
+            self._duals_t = torch.zeros_like(ths)
+            self._s_t = torch.empty_like(ths)
+            self._m_t = torch.empty_like(ths)
+            self._v_t = torch.empty_like(ths)
+        offs, idx, deg = self._csr()
+        ext.dinno_dual_threg(
+            ths, None, offs, idx, self._duals_t, self._s_t, self.rho
+        )
+        # dict view kept in sync for checkpoints/inspection
+        for i in range(pr.N):
+            self.duals[i] = self._duals_t[i]
+        theta = ths.clone()
+        for pit in range(self.pits):
+            grads = []
+            for i in range(pr.N):
+                for p in pr.node_parameters(i):
+                    p.grad = None
+                pred_loss = pr.local_batch_loss(i)
+                pred_loss.backward()
+                grads.append(
+                    torch.cat(
+                        [
+                            (p.grad if p.grad is not None
+                             else torch.zeros_like(p)).reshape(-1)
+                            for p in pr.node_parameters(i)
+                        ]
+                    )
+                )
+            grad = torch.stack(grads).to(dt).contiguous()
+            # fresh Adam per round (reference recreates the optimizer):
+            # first_step resets the moments, step_t restarts at 1
+            ext.fused_step(
+                theta, grad, self._duals_t, self._s_t, deg,
+                self._m_t, self._v_t,
+                self.rho, lr, 0.9, 0.999, 1e-8, 0.0,
+                pit + 1, 0, pit == 0, 1, False,
+            )
+            for i in range(pr.N):
+                pr.set_node_vector(i, theta[i])
+
     def step_round(self, it):
         pr = self.pr
-        ths = self._snapshot_vectors()
         self.rho *= self.rho_scaling
         lr = float(self.lr[min(it, len(self.lr) - 1)])
+        if self._hip_available():
+            return self._step_round_hip(it, lr)
+        ths = self._snapshot_vectors()
         for i in range(pr.N):
             neighs = self._neighbors(i)
             thj = torch.stack([ths[j] for j in neighs])

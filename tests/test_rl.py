@@ -297,3 +297,51 @@ def test_mpe_prey_reward_bound_penalty():
     env.prey_pos = np.array([0.0, 0.0])
     env.pred_pos = np.array([[0.05, 0.0]])  # touching
     np.testing.assert_allclose(env.prey_reward(), -10.0, rtol=1e-9)
+
+
+@pytest.mark.gpu
+@pytest.mark.skipif(not torch.cuda.is_available(),
+                    reason="needs MI355X")
+def test_dinno_ppo_hip_round_matches_torch(monkeypatch):
+    """DiNNO-PPO round math on the stacked HIP kernels (dual ascent +
+    fused penalty-Adam) vs the per-node torch loop, same fixed rollout
+    (reference RL/dist_rl/dinnoPPO.py:87-131)."""
+    import copy
+
+    from nn_distributed_training_amd.rl.ppo_optimizers import DiNNOPPO
+
+    torch.manual_seed(0)
+    env = SimpleTagEnv(num_predators=3, num_obstacles=1, seed=0,
+                       max_steps=20)
+    conf = {
+        "timesteps_per_batch": 60,
+        "max_timesteps_per_episode": 20,
+        "hidden": (16, 16),
+        "verbose": False,
+        "rho_init": 0.2,
+        "primal_iterations": 3,
+        "expected_iterations": 10,
+        "primal_lr_start": 1e-3,
+        "primal_lr_finish": 1e-4,
+    }
+    graph = nx.wheel_graph(3)
+    pr1 = DistPPOProblem(graph, env, torch.device("cuda"), conf)
+    pr1.rollout()
+    pr2 = copy.deepcopy(pr1)
+
+    monkeypatch.setenv("NDTA_RL_HIP", "0")
+    o1 = DiNNOPPO(pr1, conf)
+    o1.step_round(0)
+    monkeypatch.setenv("NDTA_RL_HIP", "1")
+    o2 = DiNNOPPO(pr2, conf)
+    assert o2._hip_available()
+    o2.step_round(0)
+
+    for i in range(3):
+        torch.testing.assert_close(
+            pr2.node_vector(i), pr1.node_vector(i),
+            rtol=2e-4, atol=2e-5,
+        )
+        torch.testing.assert_close(
+            o2.duals[i], o1.duals[i], rtol=2e-4, atol=2e-5
+        )
