@@ -107,31 +107,58 @@ __global__ void conv_pool_bwd_k(
   for (int i = 0; i < KMAX * KMAX; ++i) dw[i] = T(0);
 
   const int work = (b1 - b0) * P * P;
-  for (int t = threadIdx.x; t < work; t += blockDim.x) {
-    const int b = b0 + t / (P * P);
-    const int py = (t / P) % P;
-    const int px = t % P;
-    const long lb = (long)l * B + b;
-    const long o = lb * npool + f * P * P + py * P + px;
-    const T g = dY[o];
-    if (g == T(0)) continue;
-    const int d = idx[o];
-    const int cy = 2 * py + (d >> 1);
-    const int cx = 2 * px + (d & 1);
-    const T* img = X + lb * IMG * IMG;
-    db += g;
-    // compile-time bounds + KMAX-strided indices: a runtime `ky*K+kx`
-    // subscript makes dw[] dynamically indexed and the compiler spills
-    // the whole accumulator to scratch (guide §5.4 rule 20 — the same
-    // trap dw_skinny_*_k hit); this kernel was 45 us/call that way
+  if (K == KMAX) {
+    // exact-K fast path: no per-element zero-skip branch and no
+    // runtime breaks inside the unrolled GLOBAL-load chain — both
+    // forced per-element branch + vmcnt waits around the 25 img loads
+    // (trap 4c); unconditional taps pipeline (g = 0 contributes 0)
+    for (int t = threadIdx.x; t < work; t += blockDim.x) {
+      const int b = b0 + t / (P * P);
+      const int py = (t / P) % P;
+      const int px = t % P;
+      const long lb = (long)l * B + b;
+      const long o = lb * npool + f * P * P + py * P + px;
+      const T g = dY[o];
+      const int d = idx[o];
+      const int cy = 2 * py + (d >> 1);
+      const int cx = 2 * px + (d & 1);
+      const T* img = X + lb * IMG * IMG;
+      db += g;
 #pragma unroll
-    for (int ky = 0; ky < KMAX; ++ky) {
-      if (ky >= K) break;
-      const T* row = img + (cy + ky) * IMG + cx;
+      for (int ky = 0; ky < KMAX; ++ky) {
+        const T* row = img + (cy + ky) * IMG + cx;
 #pragma unroll
-      for (int kx = 0; kx < KMAX; ++kx) {
-        if (kx >= K) break;
-        dw[ky * KMAX + kx] += g * row[kx];
+        for (int kx = 0; kx < KMAX; ++kx) {
+          dw[ky * KMAX + kx] += g * row[kx];
+        }
+      }
+    }
+  } else {
+    for (int t = threadIdx.x; t < work; t += blockDim.x) {
+      const int b = b0 + t / (P * P);
+      const int py = (t / P) % P;
+      const int px = t % P;
+      const long lb = (long)l * B + b;
+      const long o = lb * npool + f * P * P + py * P + px;
+      const T g = dY[o];
+      if (g == T(0)) continue;
+      const int d = idx[o];
+      const int cy = 2 * py + (d >> 1);
+      const int cx = 2 * px + (d & 1);
+      const T* img = X + lb * IMG * IMG;
+      db += g;
+      // compile-time bounds + KMAX-strided indices: a runtime
+      // `ky*K+kx` subscript makes dw[] dynamically indexed and the
+      // compiler spills the whole accumulator to scratch (rule 20)
+#pragma unroll
+      for (int ky = 0; ky < KMAX; ++ky) {
+        if (ky >= K) break;
+        const T* row = img + (cy + ky) * IMG + cx;
+#pragma unroll
+        for (int kx = 0; kx < KMAX; ++kx) {
+          if (kx >= K) break;
+          dw[ky * KMAX + kx] += g * row[kx];
+        }
       }
     }
   }
