@@ -134,7 +134,7 @@ def test_two_rank_matches_single_process(alg, tmp_path):
 # dynamic-graph multirank coverage: the online density problem's
 # position all-gather + per-round disk-graph rebuild must produce
 # identical schedules (and therefore identical parameters) on 2 ranks
-def _run_online(alg="dsgd"):
+def _run_online(alg="dsgd", comm_radius=500.0, rounds=4):
     import numpy as np
 
     from nn_distributed_training_amd.data.floorplan import (
@@ -168,7 +168,7 @@ def _run_online(alg="dsgd"):
         "problem_name": "odense",
         "train_batch_size": 32,
         "val_batch_size": 64,
-        "comm_radius": 500.0,
+        "comm_radius": comm_radius,
         "dynamic_graph": True,
         "save_models": False,
         "data_seed": 9,
@@ -176,7 +176,8 @@ def _run_online(alg="dsgd"):
         "metrics": ["consensus_error"],
         "metrics_config": {"evaluate_frequency": 1000},
         "optimizer_config": {
-            "alg_name": alg, "outer_iterations": 4, "alpha0": 0.002,
+            "alg_name": alg, "outer_iterations": rounds,
+            "alpha0": 0.002,
             "mu": 0.001, "profile": False,
         },
     }
@@ -189,12 +190,14 @@ def _run_online(alg="dsgd"):
     return pr.local_params_stack(), pr.local_nodes
 
 
-def _online_worker(rank, world, port, out_dir):
+def _online_worker(rank, world, port, out_dir, comm_radius=500.0,
+                   rounds=4):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
-        stack, nodes = _run_online()
+        stack, nodes = _run_online(comm_radius=comm_radius,
+                                   rounds=rounds)
         with open(os.path.join(out_dir, f"od{rank}.pkl"), "wb") as f:
             pickle.dump((list(nodes), stack.numpy()), f)
     finally:
@@ -293,3 +296,43 @@ def test_32_nodes_8_ranks_matches_single_process(alg, tmp_path):
                 rtol=0, atol=0,
             )
     assert seen == set(range(32))
+
+
+def test_online_density_churn_four_ranks(tmp_path):
+    """Dynamic-graph EDGE CHURN at 1 node/rank: a tight comm radius
+    makes the disk graph change shape (and drop nodes to isolation)
+    as the robots move — every edge is cross-rank, so the per-round
+    P2P schedule rebuild is exercised under churn.
+    Still bitwise vs the single-process run."""
+    # verify the graph actually churns at this radius (the robots'
+    # pairwise distances cross 16.0 as the windows advance)
+    import nn_distributed_training_amd.problems.\
+dist_online_dense_problem as _m
+
+    edge_sets = []
+    orig = _m.DistOnlineDensityProblem.update_graph
+
+    def spy(self):
+        orig(self)
+        edge_sets.append(frozenset(map(tuple, self.graph.edges())))
+
+    _m.DistOnlineDensityProblem.update_graph = spy
+    try:
+        golden, nodes = _run_online(comm_radius=16.0, rounds=12)
+    finally:
+        _m.DistOnlineDensityProblem.update_graph = orig
+    assert list(nodes) == list(range(4))
+    assert len(set(edge_sets)) >= 2, "no topology churn at r=16"
+
+    mp.start_processes(
+        _online_worker, args=(4, 29727, str(tmp_path), 16.0, 12),
+        nprocs=4, join=True, start_method="spawn",
+    )
+    pieces = {}
+    for r in range(4):
+        with open(tmp_path / f"od{r}.pkl", "rb") as f:
+            local_nodes, stack = pickle.load(f)
+        for li, i in enumerate(local_nodes):
+            pieces[i] = torch.from_numpy(stack[li])
+    dist_stack = torch.stack([pieces[i] for i in range(4)])
+    torch.testing.assert_close(dist_stack, golden, rtol=0, atol=0)
