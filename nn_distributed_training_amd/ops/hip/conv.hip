@@ -14,11 +14,17 @@ namespace conv {
 // idx: [L*B, F*P*P] argmax position (0..3) for pool backward.
 // Grid: one block per image, 256 threads. KMAX bounds the tap loops at
 // compile time (K<=KMAX checked by the caller) so they fully unroll.
+// src_idx != null: the image rows GATHER straight from the resident
+// dataset (X = X_all [L, maxlen, IMG*IMG], row = src_idx[l*idx_stride
+// + idx_off + b]) — the separate gather_batch launch and the xb
+// buffer round-trip disappear from the fused-fc train path.
 template <typename T, int KMAX>
 __global__ void conv_pool_fwd_k(
     const T* __restrict__ X, const T* __restrict__ theta,
     T* __restrict__ Y, unsigned char* __restrict__ idx,
-    long n, long w_off, long b_off, int B, int F, int K, int IMG) {
+    long n, long w_off, long b_off, int B, int F, int K, int IMG,
+    const long* __restrict__ src_idx, long idx_stride, long idx_off,
+    long maxlen) {
   extern __shared__ __align__(16) unsigned char smem_raw[];
   T* img = reinterpret_cast<T*>(smem_raw);             // [IMG*IMG]
   T* wgt = img + IMG * IMG;                            // [F*K*K + F]
@@ -29,8 +35,11 @@ __global__ void conv_pool_fwd_k(
   const int P = conv_out / 2;
 
   // stage image + this node's conv weights/bias
+  const long src_row =
+      src_idx ? l * maxlen + src_idx[l * idx_stride + idx_off + lb % B]
+              : lb;
   for (int t = threadIdx.x; t < IMG * IMG; t += blockDim.x) {
-    img[t] = X[lb * IMG * IMG + t];
+    img[t] = X[src_row * IMG * IMG + t];
   }
   const T* Wg = theta + l * n + w_off;
   const T* bg = theta + l * n + b_off;
@@ -90,7 +99,9 @@ __global__ void conv_pool_bwd_k(
     const T* __restrict__ dY, const unsigned char* __restrict__ idx,
     const T* __restrict__ X, T* __restrict__ gstack,
     long n, long w_off, long b_off, int B, int F, int K, int IMG,
-    int nchunk) {
+    int nchunk,
+    const long* __restrict__ src_idx, long idx_stride, long idx_off,
+    long maxlen) {
   const int chunk = blockIdx.x % nchunk;
   const int f = (blockIdx.x / nchunk) % F;
   const int l = blockIdx.x / (nchunk * F);
@@ -122,7 +133,10 @@ __global__ void conv_pool_bwd_k(
       const int d = idx[o];
       const int cy = 2 * py + (d >> 1);
       const int cx = 2 * px + (d & 1);
-      const T* img = X + lb * IMG * IMG;
+      const long src_row =
+          src_idx ? l * maxlen + src_idx[l * idx_stride + idx_off + b]
+                  : lb;
+      const T* img = X + src_row * IMG * IMG;
       db += g;
 #pragma unroll
       for (int ky = 0; ky < KMAX; ++ky) {
@@ -145,7 +159,10 @@ __global__ void conv_pool_bwd_k(
       const int d = idx[o];
       const int cy = 2 * py + (d >> 1);
       const int cx = 2 * px + (d & 1);
-      const T* img = X + lb * IMG * IMG;
+      const long src_row =
+          src_idx ? l * maxlen + src_idx[l * idx_stride + idx_off + b]
+                  : lb;
+      const T* img = X + src_row * IMG * IMG;
       db += g;
       // compile-time bounds + KMAX-strided indices: a runtime
       // `ky*K+kx` subscript makes dw[] dynamically indexed and the

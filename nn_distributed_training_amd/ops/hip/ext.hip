@@ -559,13 +559,15 @@ void conv_pool_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
           dim3(L * B), dim3(256), shmem, cur_stream(),
           X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
-          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
+          (const long*)nullptr, 0L, 0L, 0L);
     } else {
       hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 7>),
           dim3(L * B), dim3(256), shmem, cur_stream(),
           X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
-          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG);
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
+          (const long*)nullptr, 0L, 0L, 0L);
     }
   });
   HIP_CHECK_LAST();
@@ -589,13 +591,15 @@ void conv_pool_bwd(torch::Tensor dY, torch::Tensor idx, torch::Tensor X,
           dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
           dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
           X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
-          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk,
+          (const long*)nullptr, 0L, 0L, 0L);
     } else {
       hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
           dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
           dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
           X.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
-          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk);
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk,
+          (const long*)nullptr, 0L, 0L, 0L);
     }
   });
   HIP_CHECK_LAST();
@@ -730,6 +734,71 @@ void regression_bwd(torch::Tensor yhat, torch::Tensor tgt,
           yhat.data_ptr<scalar_t>(), tgt.data_ptr<scalar_t>(),
           dY.data_ptr<scalar_t>(), lptr, yhat.numel(), (int)B,
           (scalar_t)loss_scale);
+    }
+  });
+  HIP_CHECK_LAST();
+}
+
+
+// gather-fused conv wrappers: image rows read straight from the
+// resident dataset via the sampler's index stream (no gather_batch
+// launch, no xb buffer round-trip)
+void conv_pool_fwd_idx(torch::Tensor X_all, torch::Tensor src_idx,
+                       long idx_stride, long idx_off,
+                       torch::Tensor theta, torch::Tensor Y,
+                       torch::Tensor idx, long w_off, long b_off,
+                       long B, long F, long K, long IMG) {
+  CHECK_DEV(X_all); CHECK_DEV(theta); CHECK_DEV(Y);
+  const long L = theta.size(0), n = theta.size(1);
+  const long maxlen = X_all.size(1);
+  TORCH_CHECK(K <= 7, "conv_pool_fwd supports kernel size <= 7");
+  DISPATCH_FT(X_all, {
+    const size_t shmem =
+        (IMG * IMG + F * K * K + F) * sizeof(scalar_t);
+    if (K <= 5) {
+      hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 5>),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
+          X_all.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
+          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen);
+    } else {
+      hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 7>),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
+          X_all.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
+          Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
+          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen);
+    }
+  });
+  HIP_CHECK_LAST();
+}
+
+void conv_pool_bwd_idx(torch::Tensor dY, torch::Tensor idx,
+                       torch::Tensor X_all, torch::Tensor src_idx,
+                       long idx_stride, long idx_off,
+                       torch::Tensor gstack, long w_off, long b_off,
+                       long B, long F, long K, long IMG) {
+  CHECK_DEV(dY); CHECK_DEV(X_all); CHECK_DEV(gstack);
+  const long L = gstack.size(0), n = gstack.size(1);
+  const long maxlen = X_all.size(1);
+  TORCH_CHECK(K <= 7, "conv_pool_bwd supports kernel size <= 7");
+  const int nchunk = (int)std::min<long>(B, 32);
+  DISPATCH_FT(dY, {
+    if (K <= 5) {
+      hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 5>),
+          dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
+          dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          X_all.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk,
+          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen);
+    } else {
+      hipLaunchKernelGGL((conv::conv_pool_bwd_k<scalar_t, 7>),
+          dim3(L * F * nchunk), dim3(256), 0, cur_stream(),
+          dY.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
+          X_all.data_ptr<scalar_t>(), gstack.data_ptr<scalar_t>(),
+          n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG, nchunk,
+          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen);
     }
   });
   HIP_CHECK_LAST();
@@ -985,6 +1054,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("linear_bwd_dw", &linear_bwd_dw);
   mod.def("conv_pool_fwd", &conv_pool_fwd);
   mod.def("conv_pool_bwd", &conv_pool_bwd);
+  mod.def("conv_pool_fwd_idx", &conv_pool_fwd_idx);
+  mod.def("conv_pool_bwd_idx", &conv_pool_bwd_idx);
   mod.def("logsoftmax", &logsoftmax);
   mod.def("nll_bwd", &nll_bwd);
   mod.def("nll_fused", &nll_fused);

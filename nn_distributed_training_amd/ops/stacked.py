@@ -516,20 +516,25 @@ class StackedEngine:
 
     # ------------------------------------------------------------------
     def next_batch(self):
-        """Assemble the next per-node batches with two gather kernels."""
+        """Assemble the next per-node batches with two gather kernels.
+        On the fused-fc path the conv kernels gather the image rows
+        themselves (conv_pool_*_idx): no gather launch, no xb buffer.
+        """
         if self._bufs is None:
             self._bufs = self._alloc_bufs()
         idx, stride, off = self.sampler.next_ref()
+        self._last_off = off
+        if self._has_node0:
+            self.pr.forward_cnt += self.B
+        if self.fc_block_applicable():
+            return None, None
         xb, yb = self._bufs["xb"], self._bufs["yb"]
         self.ext.gather_batch(self.X_all, idx, xb, stride, off)
-        self._last_off = off
         if self.classification:
             # targets stay resident: the fused NLL kernel gathers them
             yb = None
         else:
             self.ext.gather_targets(self.Y_all, idx, yb, stride, off)
-        if self._has_node0:
-            self.pr.forward_cnt += self.B
         return xb, yb
 
     # ------------------------------------------------------------------
@@ -636,8 +641,13 @@ class StackedEngine:
             conv, fc1, fc2 = self.spec.layers
             M = self.B
             ext = self.ext
-            ext.conv_pool_fwd(
-                xb, self.theta, bufs["acts"][0], bufs["idxs"][0],
+            idx_t = self.sampler.stream if hasattr(
+                self.sampler, "stream") else self.sampler.buf
+            stride = self.sampler.S if hasattr(
+                self.sampler, "S") else self.B
+            ext.conv_pool_fwd_idx(
+                self.X_all, idx_t, stride, self._last_off,
+                self.theta, bufs["acts"][0], bufs["idxs"][0],
                 conv.w_off, conv.b_off, M, conv.out_dim,
                 conv.kernel_size, conv.in_dim,
             )
@@ -651,10 +661,6 @@ class StackedEngine:
             if want_loss:
                 bufs["loss"].zero_()
                 loss_buf = bufs["loss"]
-            idx_t = self.sampler.stream if hasattr(
-                self.sampler, "stream") else self.sampler.buf
-            stride = self.sampler.S if hasattr(
-                self.sampler, "S") else self.B
             ext.fc_block(
                 bufs["acts"][0], self.theta, self.Y_all, idx_t,
                 stride, self._last_off, self.grad, bufs["dzs"][0],
@@ -663,8 +669,9 @@ class StackedEngine:
                 M, fc1.in_dim, fc1.out_dim, fc2.out_dim, loss_scale,
                 True,
             )
-            ext.conv_pool_bwd(
-                bufs["dzs"][0], bufs["idxs"][0], xb, self.grad,
+            ext.conv_pool_bwd_idx(
+                bufs["dzs"][0], bufs["idxs"][0], self.X_all, idx_t,
+                stride, self._last_off, self.grad,
                 conv.w_off, conv.b_off, M, conv.out_dim,
                 conv.kernel_size, conv.in_dim,
             )
