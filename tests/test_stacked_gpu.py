@@ -293,9 +293,13 @@ def test_fused_mnist_step_matches_layered(dtype, monkeypatch):
             eng.run_fused_mnist(off=off)
             eng.reduce_fused_grad()
         else:
+            # force the LAYERED chain (next_batch skips the gather on
+            # the fused-fc path, whose grads are covered elsewhere)
+            monkeypatch.setenv("NDTA_FC_BLOCK", "0")
             xb, yb = eng.next_batch()
             eng.forward(xb, train_skip_logp=True)
             eng.backward(xb, yb)
+            monkeypatch.delenv("NDTA_FC_BLOCK")
         return eng.grad.clone()
 
     g_layered = grads(False)
