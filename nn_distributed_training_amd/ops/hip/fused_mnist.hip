@@ -441,12 +441,35 @@ __global__ __launch_bounds__(256) void fc_block_k(
   }
   __syncthreads();
 
-  // P2: fc2 logits + row log-softmax + NLL dZ2
-  for (int t = tid; t < RT * C; t += 256) {
-    const int m = t / C, c = t % C;
-    T z = w2s[C * H + c];
-    for (int h = 0; h < H; ++h) z += y1s[m * 65 + h] * w2s[c * H + h];
-    dz2s[m * 17 + c] = z;
+  // P2: fc2 logits + row log-softmax + NLL dZ2 (H==64 unrolls the
+  // dot loop 8-wide so the LDS loads batch instead of stalling one
+  // lgkm wait per iteration)
+  if (H == 64) {
+    for (int t = tid; t < RT * C; t += 256) {
+      const int m = t / C, c = t % C;
+      T z = w2s[C * 64 + c];  // b2
+      T part[8] = {};
+#pragma unroll
+      for (int hq = 0; hq < 8; ++hq) {
+#pragma unroll
+        for (int hh = 0; hh < 8; ++hh) {
+          const int h = hq * 8 + hh;
+          part[hh] += y1s[m * 65 + h] * w2s[c * 64 + h];
+        }
+      }
+#pragma unroll
+      for (int hh = 0; hh < 8; ++hh) z += part[hh];
+      dz2s[m * 17 + c] = z;
+    }
+  } else {
+    for (int t = tid; t < RT * C; t += 256) {
+      const int m = t / C, c = t % C;
+      T z = w2s[C * H + c];
+      for (int h = 0; h < H; ++h) {
+        z += y1s[m * 65 + h] * w2s[c * H + h];
+      }
+      dz2s[m * 17 + c] = z;
+    }
   }
   __syncthreads();
   if (tid < RT) {
