@@ -345,3 +345,16 @@ def test_dinno_ppo_hip_round_matches_torch(monkeypatch):
         torch.testing.assert_close(
             o2.duals[i], o1.duals[i], rtol=2e-4, atol=2e-5
         )
+
+
+def test_eval_policy_rollout_animation(tmp_path):
+    """--animate renders a rollout GIF (parity with the reference's
+    live renderer in RL/dist_rl/eval_policy.py)."""
+    from nn_distributed_training_amd.rl.eval_policy import main
+
+    out = tmp_path / "roll.gif"
+    main([
+        "examples/rl_trained/ppo_actors_tag_cadmm_0.pth",
+        "--episodes", "1", "--animate", str(out),
+    ])
+    assert out.exists() and out.stat().st_size > 10000
