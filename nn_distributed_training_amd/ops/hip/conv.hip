@@ -122,20 +122,36 @@ __global__ void conv_pool_bwd_k(
     // exact-K fast path: no per-element zero-skip branch and no
     // runtime breaks inside the unrolled GLOBAL-load chain — both
     // forced per-element branch + vmcnt waits around the 25 img loads
-    // (trap 4c); unconditional taps pipeline (g = 0 contributes 0)
-    for (int t = threadIdx.x; t < work; t += blockDim.x) {
-      const int b = b0 + t / (P * P);
-      const int py = (t / P) % P;
-      const int px = t % P;
-      const long lb = (long)l * B + b;
-      const long o = lb * npool + f * P * P + py * P + px;
-      const T g = dY[o];
-      const int d = idx[o];
+    // (trap 4c); unconditional taps pipeline (g = 0 contributes 0).
+    // The (g, pool-argmax) pair for iteration t+1 prefetches while
+    // iteration t's taps run — idx[o] -> img-address is a 2-hop
+    // dependency that otherwise heads every iteration (loads use a
+    // CLAMPED index so the prefetch needs no bounds branch).
+    auto hdr = [&](long t, T& g, int& d, long& b) {
+      const long tc = t < work ? t : (work > 0 ? work - 1 : 0);
+      b = b0 + tc / (P * P);
+      const long o = ((long)l * B + b) * npool + f * P * P
+                     + ((tc / P) % P) * P + tc % P;
+      g = dY[o];
+      d = idx[o];
+    };
+    long t = threadIdx.x;
+    T g = T(0);
+    int d = 0;
+    long b = 0;
+    if (work > 0) hdr(t, g, d, b);
+    for (; t < work; t += blockDim.x) {
+      T gn;
+      int dn;
+      long bn;
+      hdr(t + blockDim.x, gn, dn, bn);
+      const int py = (int)((t / P) % P);
+      const int px = (int)(t % P);
       const int cy = 2 * py + (d >> 1);
       const int cx = 2 * px + (d & 1);
       const long src_row =
           src_idx ? l * maxlen + src_idx[l * idx_stride + idx_off + b]
-                  : lb;
+                  : (long)l * B + b;
       const T* img = X + src_row * IMG * IMG;
       db += g;
 #pragma unroll
@@ -146,6 +162,9 @@ __global__ void conv_pool_bwd_k(
           dw[ky * KMAX + kx] += g * row[kx];
         }
       }
+      g = gn;
+      d = dn;
+      b = bn;
     }
   } else {
     for (int t = threadIdx.x; t < work; t += blockDim.x) {
