@@ -821,7 +821,7 @@ void fc_block(torch::Tensor x0, torch::Tensor theta,
   const long mtiles = (M + fmnist::FC_RT - 1) / fmnist::FC_RT;
   DISPATCH_FT(x0, {
     hipLaunchKernelGGL(fmnist::fc_block_k<scalar_t>,
-        dim3(mtiles, 1, L), dim3(256), 0, cur_stream(),
+        dim3(mtiles, 1, L), dim3(512), 0, cur_stream(),
         x0.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
         Y_all.data_ptr<long>(), idx.data_ptr<long>(), idx_stride,
         idx_off, maxlen, grad.data_ptr<scalar_t>(),

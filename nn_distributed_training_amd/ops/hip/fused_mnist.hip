@@ -347,7 +347,7 @@ __global__ __launch_bounds__(256) void mnist_train_step_k(
 // (caller zeroes the whole grad stack).
 constexpr int FC_RT = 16;
 template <typename T>
-__global__ __launch_bounds__(256) void fc_block_k(
+__global__ __launch_bounds__(512) void fc_block_k(
     const T* __restrict__ x0,        // [L*M, I] conv/pool output
     const T* __restrict__ theta,     // [L, n]
     const long* __restrict__ Y_all,  // [L, maxlen] targets
@@ -362,75 +362,118 @@ __global__ __launch_bounds__(256) void fc_block_k(
     T* __restrict__ loss,            // nullable [L]
     long n, long w1_off, long b1_off, long w2_off, long b2_off,
     int M, int I, int H, int C, T loss_scale, int mask_dx0) {
+  // v2: 512 threads = 4 CONSUMER waves (MFMA) + 4 PRODUCER waves
+  // (HBM -> double-buffered LDS). W1's 221 KB/block stream was the
+  // serial chain of the 256-thread version (stage, barrier, MFMA,
+  // barrier x14): wave specialization keeps the stream entirely off
+  // the compute path.
   using MF = gmfma::mfma_t<T>;
   using acc_t = typename MF::acc_t;
+  typedef T vec2 __attribute__((ext_vector_type(2)));
   constexpr int RT = FC_RT;
 
-  // single LDS arena (a second __shared__ object de-pipelines glds
-  // paths elsewhere in this TU; keep the convention here too)
-  __shared__ T lds[32 * (RT + 1) + 32 * 65 + 2 * RT * 65 + RT * 17 +
-                   16 * 64 + 16 + RT];
-  T* As = lds;                          // [32][RT+1] x0 k-image
-  T* Bs = As + 32 * (RT + 1);           // [32][65]   W1 k-image
-  T* y1s = Bs + 32 * 65;                // [RT][65]
-  T* dz1s = y1s + RT * 65;              // [RT][65]
-  T* dz2s = dz1s + RT * 65;             // [RT][17]
-  T* w2s = dz2s + RT * 17;              // [C*H + C]
-  T* rowval = w2s + 16 * 64 + 16;       // [RT] scratch
+  // single LDS arena (trap 4a)
+  __shared__ T lds[2 * 32 * (RT + 1) + 2 * 32 * 65 + 2 * RT * 65 +
+                   RT * 17 + 16 * 64 + 16];
+  T* As = lds;                             // [2][32][RT+1]
+  T* Bs = As + 2 * 32 * (RT + 1);          // [2][32][65]
+  T* y1s = Bs + 2 * 32 * 65;               // [RT][65]
+  T* dz1s = y1s + RT * 65;                 // [RT][65]
+  T* dz2s = dz1s + RT * 65;                // [RT][17]
+  T* w2s = dz2s + RT * 17;                 // [C*H + C]
 
   const long l = blockIdx.z;
   const int m0 = blockIdx.x * RT;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
-  const int wid = tid >> 6;
+  const int wid = tid >> 6;                // 0-3 consume, 4-7 produce
+  const bool producer = wid >= 4;
   const T* th = theta + l * n;
   const T* W1 = th + w1_off;
   const T* b1 = th + b1_off;
   const int lo = lane & 15;
   const int lk = lane >> 4;
-
-  // P0: W2 + b2 into LDS
-  for (int t = tid; t < C * H; t += 256) w2s[t] = th[w2_off + t];
-  for (int t = tid; t < C; t += 256) w2s[C * H + t] = th[b2_off + t];
-
-  // P1: fc1 forward on MFMA — each wave owns one 16-wide o-fragment.
-  // T14 write-after-barrier register pipeline + vec2 loads: stage s+1
-  // streams from HBM while stage s runs its MFMAs (W1's 221 KB/block
-  // stream was the serial chain that left the first builds at
-  // 97 -> 67 us); guard-free loads on full tiles/stages (trap 4c).
-  typedef T vec2 __attribute__((ext_vector_type(2)));
-  acc_t a1 = {};
-  const int o0w = wid * 16;
   const bool fullm = (m0 + RT) <= M;
-  const int am = tid / 16, akp = tid % 16;   // A pair: row am, kpair
-  vec2 rA, rB[4];
   const int nst = (I + 31) / 32;
 
-#define FC_LOAD_STAGE(k0v)                                               {                                                                        const int k0_ = (k0v);                                                 const bool fullk_ = (k0_ + 32) <= I;                                   if (fullm && fullk_ && H == 64) { rA = *reinterpret_cast<const vec2*>(&x0[(long)(l * (long)M + m0 + am) * I + k0_ + 2 * akp]); _Pragma("unroll") for (int q = 0; q < 4; ++q) { rB[q] = *reinterpret_cast<const vec2*>(&W1[(long)(am + q * 16) * I + k0_ + 2 * akp]); } } else {                                                                 rA = vec2{0, 0};                                                       const int k_ = k0_ + 2 * akp;                                          if (m0 + am < M) {                                                       if (k_ + 1 < I) {                                                        rA = *reinterpret_cast<const vec2*>(                                       &x0[(long)(l * (long)M + m0 + am) * I + k_]);                    } else if (k_ < I) {                                                     rA.x = x0[(long)(l * (long)M + m0 + am) * I + k_];                   }                                                                    }                                                                      _Pragma("unroll") for (int q = 0; q < 4; ++q) {                          const int o = am + q * 16;                                             rB[q] = vec2{0, 0};                                                    if (o < H) {                                                             if (k_ + 1 < I) {                                                        rB[q] = *reinterpret_cast<const vec2*>(                                    &W1[(long)o * I + k_]);                                          } else if (k_ < I) {                                                     rB[q].x = W1[(long)o * I + k_];                                      }                                                                    }                                                                    }                                                                    }                                                                    }
+  // P0: W2 + b2 into LDS
+  for (int t = tid; t < C * H; t += 512) w2s[t] = th[w2_off + t];
+  for (int t = tid; t < C; t += 512) w2s[C * H + t] = th[b2_off + t];
 
-  FC_LOAD_STAGE(0)
+  // producer staging: 256 threads cover the [32 k][RT m] x0 image
+  // (1 vec2 pair each) and the [32 k][64 o] W1 image (4 pairs)
+  const int ptid = tid - 256;
+  const int am = producer ? ptid / 16 : 0;
+  const int akp = producer ? ptid % 16 : 0;
+  const auto stage = [&](int buf, int k0) {
+    T* Ab = As + buf * 32 * (RT + 1);
+    T* Bb = Bs + buf * 32 * 65;
+    if (fullm && (k0 + 32) <= I && H == 64) {  // guard-free (trap 4c)
+      const vec2 va = *reinterpret_cast<const vec2*>(
+          &x0[(long)(l * (long)M + m0 + am) * I + k0 + 2 * akp]);
+      Ab[(2 * akp) * (RT + 1) + am] = va.x;
+      Ab[(2 * akp + 1) * (RT + 1) + am] = va.y;
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const vec2 vb = *reinterpret_cast<const vec2*>(
+            &W1[(long)(am + q * 16) * I + k0 + 2 * akp]);
+        Bb[(2 * akp) * 65 + am + q * 16] = vb.x;
+        Bb[(2 * akp + 1) * 65 + am + q * 16] = vb.y;
+      }
+    } else {
+      const int k_ = k0 + 2 * akp;
+      vec2 va = vec2{0, 0};
+      if (m0 + am < M) {
+        if (k_ + 1 < I) {
+          va = *reinterpret_cast<const vec2*>(
+              &x0[(long)(l * (long)M + m0 + am) * I + k_]);
+        } else if (k_ < I) {
+          va.x = x0[(long)(l * (long)M + m0 + am) * I + k_];
+        }
+      }
+      Ab[(2 * akp) * (RT + 1) + am] = va.x;
+      Ab[(2 * akp + 1) * (RT + 1) + am] = va.y;
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        const int o = am + q * 16;
+        vec2 vb = vec2{0, 0};
+        if (o < H) {
+          if (k_ + 1 < I) {
+            vb = *reinterpret_cast<const vec2*>(&W1[(long)o * I + k_]);
+          } else if (k_ < I) {
+            vb.x = W1[(long)o * I + k_];
+          }
+        }
+        Bb[(2 * akp) * 65 + o] = vb.x;
+        Bb[(2 * akp + 1) * 65 + o] = vb.y;
+      }
+    }
+  };
+
+  // P1: fc1 forward on MFMA (consumer waves: one 16-wide o-fragment
+  // each) while producers fill the next stage's buffers
+  acc_t a1 = {};
+  const int o0w = wid * 16;  // consumers: 0..48
+  if (producer) stage(0, 0);
+  __syncthreads();
   for (int st = 0; st < nst; ++st) {
-    __syncthreads();
-    As[(2 * akp) * (RT + 1) + am] = rA.x;
-    As[(2 * akp + 1) * (RT + 1) + am] = rA.y;
+    if (producer) {
+      if (st + 1 < nst) stage((st + 1) & 1, (st + 1) * 32);
+    } else {
+      const T* Ab = As + (st & 1) * 32 * (RT + 1);
+      const T* Bb = Bs + (st & 1) * 32 * 65;
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      Bs[(2 * akp) * 65 + am + q * 16] = rB[q].x;
-      Bs[(2 * akp + 1) * 65 + am + q * 16] = rB[q].y;
+      for (int kk = 0; kk < 32; kk += 4) {
+        const int ka = kk + lk;
+        const T a = Ab[ka * (RT + 1) + lo];
+        const T b = Bb[ka * 65 + o0w + lo];
+        a1 = MF::mma(a, b, a1);
+      }
     }
-    if (st + 1 < nst) FC_LOAD_STAGE((st + 1) * 32)
     __syncthreads();
-#pragma unroll
-    for (int kk = 0; kk < 32; kk += 4) {
-      const int ka = kk + lk;
-      const T a = As[ka * (RT + 1) + lo];
-      const T b = Bs[ka * 65 + o0w + lo];
-      a1 = MF::mma(a, b, a1);
-    }
   }
-#undef FC_LOAD_STAGE
   // y1 = relu(z1 + b1) into LDS (never stored to HBM)
-  if (o0w < H) {
+  if (!producer && o0w < H) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int m = MF::acc_row(lane, r);
@@ -442,10 +485,9 @@ __global__ __launch_bounds__(256) void fc_block_k(
   __syncthreads();
 
   // P2: fc2 logits + row log-softmax + NLL dZ2 (H==64 unrolls the
-  // dot loop 8-wide so the LDS loads batch instead of stalling one
-  // lgkm wait per iteration)
+  // dot loop 8-wide so the LDS loads batch)
   if (H == 64) {
-    for (int t = tid; t < RT * C; t += 256) {
+    for (int t = tid; t < RT * C; t += 512) {
       const int m = t / C, c = t % C;
       T z = w2s[C * 64 + c];  // b2
       T part[8] = {};
@@ -462,7 +504,7 @@ __global__ __launch_bounds__(256) void fc_block_k(
       dz2s[m * 17 + c] = z;
     }
   } else {
-    for (int t = tid; t < RT * C; t += 256) {
+    for (int t = tid; t < RT * C; t += 512) {
       const int m = t / C, c = t % C;
       T z = w2s[C * H + c];
       for (int h = 0; h < H; ++h) {
@@ -475,16 +517,15 @@ __global__ __launch_bounds__(256) void fc_block_k(
   if (tid < RT) {
     const int m = tid;
     if (m0 + m < M) {
-      const long row = l * (long)M + m0 + m;
       const long tgt =
           Y_all[l * maxlen + idx[l * idx_stride + idx_off + m0 + m]];
       T mx = dz2s[m * 17];
       for (int c = 1; c < C; ++c) {
         mx = dz2s[m * 17 + c] > mx ? dz2s[m * 17 + c] : mx;
       }
-      T s = T(0);
-      for (int c = 0; c < C; ++c) s += ::exp(dz2s[m * 17 + c] - mx);
-      const T lse = mx + ::log(s);
+      T sum = T(0);
+      for (int c = 0; c < C; ++c) sum += ::exp(dz2s[m * 17 + c] - mx);
+      const T lse = mx + ::log(sum);
       const T w = loss_scale / T(M);
       if (loss != nullptr) {
         atomicAdd(&loss[l], -(dz2s[m * 17 + (int)tgt] - lse) / T(M));
@@ -495,15 +536,15 @@ __global__ __launch_bounds__(256) void fc_block_k(
                                                             : T(0)))
             * w;
       }
-      (void)row;
     } else {
       for (int c = 0; c < C; ++c) dz2s[m * 17 + c] = T(0);
     }
   }
   __syncthreads();
 
-  // P3: dz1 = (dz2 @ W2) * relu'(y1); fc2 dW/db; fc1 db
-  for (int t = tid; t < RT * H; t += 256) {
+  // P3: dz1 = (dz2 @ W2) * relu'(y1) -> LDS + global (fc1 dW/db run
+  // on the store-path dw kernel); fc2 dW/db
+  for (int t = tid; t < RT * H; t += 512) {
     const int m = t / H, h = t % H;
     T s = T(0);
     for (int c = 0; c < C; ++c) {
@@ -516,13 +557,12 @@ __global__ __launch_bounds__(256) void fc_block_k(
     }
   }
   if (H < 64) {  // zero the pad columns the dX0 MFMA sweeps over
-    for (int t = tid; t < RT * (64 - H); t += 256) {
+    for (int t = tid; t < RT * (64 - H); t += 512) {
       const int m = t / (64 - H), h = H + t % (64 - H);
       dz1s[m * 65 + h] = T(0);
     }
   }
-  __syncthreads();
-  for (int t = tid; t < C * H; t += 256) {
+  for (int t = tid; t < C * H; t += 512) {
     const int c = t / H, h = t % H;
     T s = T(0);
     for (int m = 0; m < RT; ++m) {
@@ -530,24 +570,22 @@ __global__ __launch_bounds__(256) void fc_block_k(
     }
     atomicAdd(&grad[l * n + w2_off + c * H + h], s);
   }
-  if (tid < C) {
+  if (tid >= 32 && tid < 32 + C) {
+    const int c = tid - 32;
     T s = T(0);
-    for (int m = 0; m < RT; ++m) s += dz2s[m * 17 + tid];
-    atomicAdd(&grad[l * n + b2_off + tid], s);
+    for (int m = 0; m < RT; ++m) s += dz2s[m * 17 + c];
+    atomicAdd(&grad[l * n + b2_off + c], s);
   }
   __syncthreads();
 
-  // P4: per 64-wide i-chunk — fc1 dW (K = RT, x0 read direct from
-  // global, coalesced) and dX0 (K = H, W1 read direct, coalesced).
-  // Full chunks take the GUARD-FREE loops (trap 4c, see P1); only the
-  // ragged last chunk / last m-tile pay the guarded versions.
-  // dX0 carries the conv layer's relu' mask (the layered path fused it
-  // in linear_bwd_dx; conv_pool_bwd expects dZ pre-masked).
-  // dX0[m, i]: A[m][k=o] = dz1 (LDS), B[k=o][i] = W1 (global,
-  // coalesced). W1's 16-value column block for chunk c+1 prefetches
-  // into registers while chunk c's MFMAs issue (dz1s pad columns are
-  // zeroed above, so the full 64-deep sweep is guard-free).
+  // P4: dX0 = dz1 @ W1 with the conv relu' mask. A[m][k=o] from LDS,
+  // B[k=o][i] = W1 direct from global (coalesced); the 64-wide
+  // i-chunks split between the two wave HALVES (half h takes chunks
+  // h, h+2, ...), each half prefetching its next chunk's W1 column
+  // block while the current chunk's MFMAs issue.
   {
+    const int half = wid >> 2;   // 0 or 1
+    const int wq = wid & 3;      // wave within the half
     T w1v[16], w1n[16];
     const auto ldw1 = [&](int ixv, T* dst) {
       if (H == 64 && ixv + 16 <= I) {  // guard-free (trap 4c)
@@ -564,31 +602,33 @@ __global__ __launch_bounds__(256) void fc_block_k(
         }
       }
     };
-    ldw1(wid * 16, w1v);
-    for (int i0 = 0; i0 < I; i0 += 64) {
-      const int ix = i0 + wid * 16;
-      if (i0 + 64 < I) ldw1(i0 + 64 + wid * 16, w1n);
-      acc_t ax = {};
+    const int i0_first = half * 64;
+    if (i0_first < I) {
+      ldw1(i0_first + wq * 16, w1v);
+      for (int i0 = i0_first; i0 < I; i0 += 128) {
+        const int ix = i0 + wq * 16;
+        if (i0 + 128 < I) ldw1(i0 + 128 + wq * 16, w1n);
+        acc_t ax = {};
 #pragma unroll
-      for (int t = 0; t < 16; ++t) {
-        const T a = dz1s[lo * 65 + 4 * t + lk];
-        ax = MF::mma(a, w1v[t], ax);
-      }
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int m = m0 + MF::acc_row(lane, r);
-        const int i = ix + lo;
-        if (m < M && i < I) {
-          const long off = (long)(l * (long)M + m) * I + i;
-          // conv relu' mask: x0 IS the conv block's relu output
-          dx0[off] = (!mask_dx0 || x0[off] > T(0)) ? ax[r] : T(0);
+        for (int t = 0; t < 16; ++t) {
+          const T a = dz1s[lo * 65 + 4 * t + lk];
+          ax = MF::mma(a, w1v[t], ax);
         }
-      }
 #pragma unroll
-      for (int t = 0; t < 16; ++t) w1v[t] = w1n[t];
+        for (int r = 0; r < 4; ++r) {
+          const int m = m0 + MF::acc_row(lane, r);
+          const int i = ix + lo;
+          if (m < M && i < I) {
+            const long off = (long)(l * (long)M + m) * I + i;
+            // conv relu' mask: x0 IS the conv block's relu output
+            dx0[off] = (!mask_dx0 || x0[off] > T(0)) ? ax[r] : T(0);
+          }
+        }
+#pragma unroll
+        for (int t = 0; t < 16; ++t) w1v[t] = w1n[t];
+      }
     }
   }
-  (void)rowval;
 }
 
 }  // namespace fmnist
