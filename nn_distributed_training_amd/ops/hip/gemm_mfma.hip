@@ -771,9 +771,12 @@ __global__ __launch_bounds__(512) void mfma_dw_direct_k(
       // the first modulo-slot version ran 1.7x SLOWER). The main loop
       // processes D steps per iteration so each slot index is the
       // unrolled d; depth covers the ~900-cycle HBM latency against
-      // each step's MFMA-issue cover (NIMAX=1 steps cover only
-      // 2x64 cycles -> deeper).
-      constexpr int D = (NIMAX == 1) ? 8 : 2;
+      // each step's MFMA-issue cover. f64 MFMA issues at 64 cyc/SIMD,
+      // f32 at 32: NIMAX=1 steps cover 2 issues, NIMAX>=4 cover 8 —
+      // the cheaper the step, the deeper the pipeline.
+      constexpr int D =
+          (NIMAX == 1) ? (sizeof(T) == 4 ? 16 : 8)
+                       : (sizeof(T) == 4 ? 4 : 2);
       int step = 0;
       if (nfull >= D) {
         T a0p[D], a1p[D], bp[D][NIMAX];

@@ -369,3 +369,27 @@ def test_stacked_fourier_forward_matches_module():
         cur = out
     ref = model(x)
     torch.testing.assert_close(cur, ref, rtol=1e-10, atol=1e-10)
+
+
+@requires_gpu
+@pytest.mark.parametrize("alg", ["dinno", "dsgt"])
+def test_chain_dispatch_matches_python_loops(alg, monkeypatch):
+    """The C++ fwd/bwd chain dispatchers launch exactly the kernels
+    the python per-op loops launch: bitwise-identical parameters after
+    3 rounds (NDTA_PY_CHAIN A/B)."""
+    torch.set_default_dtype(torch.float64)
+    conf = _conf(copy.deepcopy(ALG_CONFS[alg]))
+
+    def run(py_chain):
+        monkeypatch.setenv("NDTA_PY_CHAIN", "1" if py_chain else "0")
+        monkeypatch.setenv("NDTA_FC_BLOCK", "0")  # exercise the chains
+        pr = _build_problem(conf)
+        pr.stacked = StackedEngine(pr)
+        assert pr.stacked._use_chain == (not py_chain)
+        opt = build_optimizer(pr, pr.device, conf["optimizer_config"])
+        opt.train()
+        return pr.stacked.theta.clone()
+
+    a = run(False)
+    b = run(True)
+    torch.testing.assert_close(a, b, rtol=0, atol=0)
