@@ -392,4 +392,6 @@ def test_chain_dispatch_matches_python_loops(alg, monkeypatch):
 
     a = run(False)
     b = run(True)
-    torch.testing.assert_close(a, b, rtol=0, atol=0)
+    # conv dW accumulates atomically -> run-to-run reassociation at
+    # ~1e-13; anything above 1e-10 would be a real dispatch difference
+    torch.testing.assert_close(a, b, rtol=1e-10, atol=1e-10)
