@@ -18,35 +18,28 @@ namespace conv {
 // dataset (X = X_all [L, maxlen, IMG*IMG], row = src_idx[l*idx_stride
 // + idx_off + b]) — the separate gather_batch launch and the xb
 // buffer round-trip disappear from the fused-fc train path.
-// TI images per block: at one image/block the kernel was
-// launch/latency-padded (~170 MACs per thread; 12 us/call at the
-// MNIST shape) — four images amortize the stage + dispatch.
 template <typename T, int KMAX>
 __global__ void conv_pool_fwd_k(
     const T* __restrict__ X, const T* __restrict__ theta,
     T* __restrict__ Y, unsigned char* __restrict__ idx,
     long n, long w_off, long b_off, int B, int F, int K, int IMG,
     const long* __restrict__ src_idx, long idx_stride, long idx_off,
-    long maxlen, int TI) {
+    long maxlen) {
   extern __shared__ __align__(16) unsigned char smem_raw[];
-  T* img = reinterpret_cast<T*>(smem_raw);          // [TI][IMG*IMG]
-  T* wgt = img + (long)TI * IMG * IMG;              // [F*K*K + F]
+  T* img = reinterpret_cast<T*>(smem_raw);             // [IMG*IMG]
+  T* wgt = img + IMG * IMG;                            // [F*K*K + F]
 
-  const int tiles = (B + TI - 1) / TI;
-  const long l = blockIdx.x / tiles;
-  const int b0 = (int)(blockIdx.x % tiles) * TI;
-  const int bcnt = min(TI, B - b0);
+  const long lb = blockIdx.x;           // image index in [0, L*B)
+  const long l = lb / B;
   const int conv_out = IMG - (K - 1);
   const int P = conv_out / 2;
 
-  // stage images + this node's conv weights/bias
-  for (int t = threadIdx.x; t < bcnt * IMG * IMG; t += blockDim.x) {
-    const int ti = t / (IMG * IMG);
-    const int b = b0 + ti;
-    const long src_row =
-        src_idx ? l * maxlen + src_idx[l * idx_stride + idx_off + b]
-                : l * (long)B + b;
-    img[t] = X[src_row * IMG * IMG + (t - ti * IMG * IMG)];
+  // stage image + this node's conv weights/bias
+  const long src_row =
+      src_idx ? l * maxlen + src_idx[l * idx_stride + idx_off + lb % B]
+              : lb;
+  for (int t = threadIdx.x; t < IMG * IMG; t += blockDim.x) {
+    img[t] = X[src_row * IMG * IMG + t];
   }
   const T* Wg = theta + l * n + w_off;
   const T* bg = theta + l * n + b_off;
@@ -59,11 +52,7 @@ __global__ void conv_pool_fwd_k(
   __syncthreads();
 
   const int npool = F * P * P;
-  for (int u = threadIdx.x; u < bcnt * npool; u += blockDim.x) {
-    const int ti = u / npool;
-    const int t = u - ti * npool;
-    const long lb = l * (long)B + b0 + ti;
-    const T* imgb = img + (long)ti * IMG * IMG;
+  for (int t = threadIdx.x; t < npool; t += blockDim.x) {
     const int f = t / (P * P);
     const int py = (t / P) % P;
     const int px = t % P;
@@ -80,7 +69,7 @@ __global__ void conv_pool_fwd_k(
 #pragma unroll
       for (int ky = 0; ky < KMAX; ++ky) {
         if (ky >= K) break;
-        const T* row = imgb + (cy + ky) * IMG + cx;
+        const T* row = img + (cy + ky) * IMG + cx;
         const T* wr = wf + ky * K;
 #pragma unroll
         for (int kx = 0; kx < KMAX; ++kx) {

@@ -551,25 +551,23 @@ void conv_pool_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
   CHECK_DEV(X); CHECK_DEV(theta); CHECK_DEV(Y);
   const long L = theta.size(0), n = theta.size(1);
   TORCH_CHECK(K <= 7, "conv_pool_fwd supports kernel size <= 7");
-  const long TI = std::min<long>(4, B);
-  const long tiles = (B + TI - 1) / TI;
   DISPATCH_FT(X, {
     const size_t shmem =
-        (TI * IMG * IMG + F * K * K + F) * sizeof(scalar_t);
+        (IMG * IMG + F * K * K + F) * sizeof(scalar_t);
     if (K <= 5) {
       hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 5>),
-          dim3(L * tiles), dim3(256), shmem, cur_stream(),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
           X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
           n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
-          (const long*)nullptr, 0L, 0L, 0L, (int)TI);
+          (const long*)nullptr, 0L, 0L, 0L);
     } else {
       hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 7>),
-          dim3(L * tiles), dim3(256), shmem, cur_stream(),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
           X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
           n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
-          (const long*)nullptr, 0L, 0L, 0L, (int)TI);
+          (const long*)nullptr, 0L, 0L, 0L);
     }
   });
   HIP_CHECK_LAST();
@@ -754,27 +752,23 @@ void conv_pool_fwd_idx(torch::Tensor X_all, torch::Tensor src_idx,
   const long L = theta.size(0), n = theta.size(1);
   const long maxlen = X_all.size(1);
   TORCH_CHECK(K <= 7, "conv_pool_fwd supports kernel size <= 7");
-  const long TI = std::min<long>(4, B);
-  const long tiles = (B + TI - 1) / TI;
   DISPATCH_FT(X_all, {
     const size_t shmem =
-        (TI * IMG * IMG + F * K * K + F) * sizeof(scalar_t);
+        (IMG * IMG + F * K * K + F) * sizeof(scalar_t);
     if (K <= 5) {
       hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 5>),
-          dim3(L * tiles), dim3(256), shmem, cur_stream(),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
           X_all.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
           n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
-          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen,
-          (int)TI);
+          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen);
     } else {
       hipLaunchKernelGGL((conv::conv_pool_fwd_k<scalar_t, 7>),
-          dim3(L * tiles), dim3(256), shmem, cur_stream(),
+          dim3(L * B), dim3(256), shmem, cur_stream(),
           X_all.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), idx.data_ptr<unsigned char>(),
           n, w_off, b_off, (int)B, (int)F, (int)K, (int)IMG,
-          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen,
-          (int)TI);
+          src_idx.data_ptr<long>(), idx_stride, idx_off, maxlen);
     }
   });
   HIP_CHECK_LAST();
