@@ -589,3 +589,51 @@ def test_consensus_cdist_matches_torch(ext, N, n):
     assert torch.all(D.diagonal().abs() < 1e-12)
     torch.testing.assert_close(D, Dref, rtol=1e-9, atol=5e-7)
     torch.testing.assert_close(Dm, Dmref, rtol=1e-9, atol=5e-7)
+
+
+@requires_gpu
+def test_linear_kernels_randomized_shapes(ext):
+    """Seeded random-shape sweep across ALL the linear dispatch
+    branches (VALU / MFMA fwd, dx new+old, dw direct/chunked/skinny):
+    fwd+dw+db+dx vs autograd at every shape. Guards the many
+    shape-routing conditions against edge regressions."""
+    import random as _random
+
+    rng = _random.Random(0xC0FFEE)
+    torch.manual_seed(11)
+    dev = _dev()
+    for trial in range(14):
+        L = rng.choice([1, 2, 3])
+        M = rng.choice([17, 40, 64, 150, 300, 1024, 2600])
+        I = rng.choice([2, 3, 16, 17, 48, 64, 70, 128, 256, 333])
+        O = rng.choice([1, 2, 5, 10, 17, 40, 64, 128, 200])
+        n = I * O + O
+        X = torch.randn(L * M, I, dtype=torch.float64, device=dev)
+        theta = torch.randn(L, n, dtype=torch.float64, device=dev)
+        dY = torch.randn(L * M, O, dtype=torch.float64, device=dev)
+        Y = torch.empty(L * M, O, dtype=torch.float64, device=dev)
+        ext.linear_fwd(X, theta, Y, None, 0, I * O, M, I, O, 1, 1.0)
+        dZ = torch.empty_like(dY)
+        ext.act_grad(dY, Y, None, dZ, 1, 1.0)
+        dX = torch.empty_like(X)
+        ext.linear_bwd_dx(dZ, theta, dX, None, None, 0, 1.0, 0, M, I,
+                          O, None, 0, 0, 0)
+        g = torch.zeros_like(theta)
+        ext.linear_bwd_dw(dZ, X, g, 0, I * O, M, I, O)
+
+        tol = dict(rtol=1e-8, atol=1e-8)
+        for l in range(L):
+            W = theta[l, : I * O].reshape(O, I).detach() \
+                .requires_grad_()
+            b = theta[l, I * O:].detach().requires_grad_()
+            xl = X[l * M:(l + 1) * M].detach().requires_grad_()
+            y = torch.relu(xl @ W.T + b)
+            torch.testing.assert_close(Y[l * M:(l + 1) * M], y, **tol)
+            y.backward(dY[l * M:(l + 1) * M])
+            torch.testing.assert_close(
+                dX[l * M:(l + 1) * M], xl.grad, **tol
+            ), (trial, L, M, I, O)
+            torch.testing.assert_close(
+                g[l, : I * O].reshape(O, I), W.grad, **tol
+            )
+            torch.testing.assert_close(g[l, I * O:], b.grad, **tol)
