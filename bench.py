@@ -88,7 +88,11 @@ def main():
         faulthandler.dump_traceback_later(600, repeat=True)
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
     if torch.cuda.is_available():
-        device = torch.device("cuda", local_rank)
+        # one rank per GPU normally; the mod keeps oversubscribed
+        # smoke runs (2 ranks on a 1-GPU box) working like comm.py
+        device = torch.device(
+            "cuda", local_rank % torch.cuda.device_count()
+        )
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
