@@ -335,6 +335,24 @@ void linear_fwd(torch::Tensor X, torch::Tensor theta, torch::Tensor Y,
           X.data_ptr<scalar_t>(), theta.data_ptr<scalar_t>(),
           Y.data_ptr<scalar_t>(), zp, n, w_off, b_off,
           (int)M, (int)I, (int)O, (int)act, (scalar_t)scale);
+    } else if (I <= 4 && M >= 1024 && O >= 64 && O <= 1024 &&
+               O % 64 == 0) {
+      // register-resident encode kernel: one thread per output
+      // column (blockDim == O), W row + bias in registers, scalar
+      // X-row loads, compile-time-I guard-free dot
+      const long blocks = std::min<long>(M, 8192);
+      dim3 grid(blocks, 1, L);
+      auto launch_enc = [&](auto ik_tag) {
+        constexpr int IK = decltype(ik_tag)::value;
+        hipLaunchKernelGGL((gemm::encode_fwd_k<scalar_t, IK>), grid,
+            dim3((int)O), 0, cur_stream(), X.data_ptr<scalar_t>(),
+            theta.data_ptr<scalar_t>(), Y.data_ptr<scalar_t>(), zp, n,
+            w_off, b_off, (int)M, (int)O, (int)act, (scalar_t)scale);
+      };
+      if (I == 1) launch_enc(std::integral_constant<int, 1>{});
+      else if (I == 2) launch_enc(std::integral_constant<int, 2>{});
+      else if (I == 3) launch_enc(std::integral_constant<int, 3>{});
+      else launch_enc(std::integral_constant<int, 4>{});
     } else if (I <= 4 && M >= 1024) {
       const long total = (long)M * O;
       const long blocks =

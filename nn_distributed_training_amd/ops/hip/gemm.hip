@@ -96,6 +96,39 @@ __global__ void linear_fwd_smallk_k(
   }
 }
 
+// Small-K forward, register-resident variant (the FourierNet encode's
+// hot shape: I = 2, O = 256, M = 160k). One thread per output column
+// with its W row + bias held in REGISTERS for the whole kernel; the
+// X row address is block-uniform each iteration so the loads are
+// scalar; I is a compile-time template so the dot is guard-free
+// (trap 4c: the generic kernel's `i < I` test inside the unrolled
+// loop serializes its loads). Requires O == blockDim.x.
+template <typename T, int IK>
+__global__ void encode_fwd_k(
+    const T* __restrict__ X, const T* __restrict__ theta,
+    T* __restrict__ Y, T* __restrict__ Z,
+    long n, long w_off, long b_off, int M, int O,
+    int act, T scale) {
+  const long l = blockIdx.z;
+  const int o = threadIdx.x;
+  const T* W = theta + l * n + w_off;
+  T w[IK];
+#pragma unroll
+  for (int i = 0; i < IK; ++i) w[i] = W[(long)o * IK + i];
+  const T bo = theta[l * n + b_off + o];
+  const T* Xl = X + l * (long)M * IK;
+  T* Yl = Y + l * (long)M * O;
+  T* Zl = Z ? Z + l * (long)M * O : nullptr;
+  for (long m = blockIdx.x; m < M; m += gridDim.x) {
+    T acc = bo;
+#pragma unroll
+    for (int i = 0; i < IK; ++i) acc += Xl[m * IK + i] * w[i];
+    const long t = m * O + o;
+    if (Zl) Zl[t] = acc;
+    Yl[t] = act_fwd(act, acc, scale);
+  }
+}
+
 // dZ = dY * act'(z, y) — fused activation backward, one pass.
 template <typename T>
 __global__ void act_grad_k(
