@@ -78,6 +78,41 @@ def test_linear_fwd_sin_relu(ext, dtype, M):
         ref = torch.relu(torch.sin(scale * z))
         torch.testing.assert_close(Y[l * M : (l + 1) * M], ref,
                                    **TOL[dtype])
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float64, torch.float32])
+@pytest.mark.parametrize(
+    "I,O",  # O%64==0 -> register-resident encode_fwd_k; else smallk
+    [(2, 256), (3, 64), (2, 100), (4, 128)],
+)
+def test_linear_fwd_encode_routes(ext, dtype, I, O):
+    """Small-K forward at shapes hitting the register-resident encode
+    kernel (O a multiple of 64) and the LDS smallk fallback (ragged
+    O), with and without the Z stash."""
+    torch.manual_seed(2)
+    L, M = 3, 2048
+    n = I * O + O
+    dev = _dev()
+    scale = 0.05
+    X = torch.randn(L * M, I, dtype=dtype, device=dev)
+    theta = torch.randn(L, n, dtype=dtype, device=dev)
+    for with_z in (False, True):
+        Y = torch.empty(L * M, O, dtype=dtype, device=dev)
+        Z = torch.empty_like(Y) if with_z else None
+        ext.linear_fwd(X, theta, Y, Z, 0, I * O, M, I, O, 2, scale)
+        for l in range(L):
+            W = theta[l, : I * O].reshape(O, I)
+            b = theta[l, I * O :]
+            z = X[l * M : (l + 1) * M] @ W.T + b
+            ref = torch.relu(torch.sin(scale * z))
+            torch.testing.assert_close(
+                Y[l * M : (l + 1) * M], ref, **TOL[dtype]
+            )
+            if with_z:
+                torch.testing.assert_close(
+                    Z[l * M : (l + 1) * M], z, **TOL[dtype]
+                )
         torch.testing.assert_close(Z[l * M : (l + 1) * M], z,
                                    **TOL[dtype])
 
