@@ -78,6 +78,8 @@ def test_linear_fwd_sin_relu(ext, dtype, M):
         ref = torch.relu(torch.sin(scale * z))
         torch.testing.assert_close(Y[l * M : (l + 1) * M], ref,
                                    **TOL[dtype])
+        torch.testing.assert_close(Z[l * M : (l + 1) * M], z,
+                                   **TOL[dtype])
 
 
 @requires_gpu
@@ -113,8 +115,6 @@ def test_linear_fwd_encode_routes(ext, dtype, I, O):
                 torch.testing.assert_close(
                     Z[l * M : (l + 1) * M], z, **TOL[dtype]
                 )
-        torch.testing.assert_close(Z[l * M : (l + 1) * M], z,
-                                   **TOL[dtype])
 
 
 @requires_gpu
